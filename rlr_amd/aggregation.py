@@ -1,0 +1,254 @@
+"""Aggregation server (reference src/aggregation.py).
+
+The server consumes a stacked (S, n_params) fp64 update matrix in sampled
+order — on multi-GPU runs this is the RCCL all-gather result, identical on
+every rank, and every rank applies the aggregate redundantly (deterministic,
+zero extra traffic; SURVEY.md §2c call-site 3).  The dict-based reference
+API (aggregation.py:19) is accepted too.
+
+Rules (each a HIP kernel on the GPU path, ops/csrc/aggregation.hip):
+  * RLR sign-vote (aggregation.py:48-54): sm = |sum_k sign(U_k)| per coord;
+    lr = +server_lr where sm >= theta else -server_lr.  (The reference's
+    in-place two-pass masking is only correct because -server_lr < theta
+    for positive server_lr; this implementation is the explicit select and
+    asserts that precondition — SURVEY.md §7 quirks.)
+  * FedAvg (aggregation.py:57-64): sum_k n_k U_k / sum_k n_k, fp64, summed
+    in sampled order for world-size-invariant bitwise results.
+  * coordinate median (aggregation.py:66-69): torch.median column semantics
+    (lower of the two middle values for even K).
+  * sign (aggregation.py:71-75): sign(sum_k sign(U_k)).
+  * optional N(0, noise*clip) gaussian noise (aggregation.py:34-35) from a
+    per-round derived stream.
+  * fused apply (aggregation.py:38-40): theta <- float32(theta + lr * agg).
+"""
+
+import torch
+from torch.nn import functional as F
+
+from .ops import ext, force_eager
+from .utils.rng import torch_gen
+
+
+def _gpu(t):
+    return t.is_cuda and not force_eager()
+
+
+class Aggregation:
+    def __init__(self, agent_data_sizes, n_params, poisoned_val_loader,
+                 args, writer=None):
+        self.agent_data_sizes = agent_data_sizes
+        self.args = args
+        self.writer = writer
+        self.server_lr = args.server_lr
+        self.n_params = n_params
+        self.poisoned_val_loader = poisoned_val_loader
+        self.cum_net_mov = 0.0
+
+    # ------------------------------------------------------------- entry
+
+    def aggregate_updates(self, global_model, agent_updates, cur_round,
+                          agent_ids=None):
+        """agent_updates: dict {agent_id: fp64 vec} (reference API) or a
+        stacked (S, n) fp64 tensor with agent_ids giving the sampled order."""
+        if isinstance(agent_updates, dict):
+            agent_ids = list(agent_updates.keys())
+            stacked = torch.stack([agent_updates[i] for i in agent_ids])
+        else:
+            stacked = agent_updates
+            assert agent_ids is not None
+
+        lr_vector = None
+        if self.args.robustLR_threshold > 0:
+            lr_vector = self.compute_robustLR(stacked)
+
+        if self.args.aggr == 'avg':
+            agg = self.agg_avg(stacked, agent_ids)
+        elif self.args.aggr == 'comed':
+            agg = self.agg_comed(stacked)
+        elif self.args.aggr == 'sign':
+            agg = self.agg_sign(stacked)
+        else:
+            raise ValueError(self.args.aggr)
+
+        if self.args.noise > 0:
+            agg = agg + self._noise(cur_round, agg.device, agg.dtype)
+
+        self._apply(global_model, lr_vector, agg)
+
+    # ------------------------------------------------------------- rules
+
+    def compute_robustLR(self, stacked: torch.Tensor) -> torch.Tensor:
+        assert self.server_lr > 0 and self.args.robustLR_threshold > 0
+        if _gpu(stacked):
+            return ext().rlr_vote(stacked, float(self.args.robustLR_threshold),
+                                  float(self.server_lr))
+        sm = torch.abs(torch.sign(stacked).sum(dim=0))
+        return torch.where(sm >= self.args.robustLR_threshold,
+                           torch.full_like(sm, self.server_lr),
+                           torch.full_like(sm, -self.server_lr))
+
+    def _weights(self, agent_ids, device):
+        w = torch.tensor([float(self.agent_data_sizes[i]) for i in agent_ids],
+                         dtype=torch.float64, device=device)
+        return w
+
+    def agg_avg(self, stacked, agent_ids):
+        w = self._weights(agent_ids, stacked.device)
+        if _gpu(stacked):
+            return ext().agg_avg(stacked, w)
+        # sequential accumulation in sampled order (ws-invariant)
+        out = torch.zeros(stacked.shape[1], dtype=torch.float64,
+                          device=stacked.device)
+        for k in range(stacked.shape[0]):
+            out += w[k] * stacked[k]
+        return out / w.sum()
+
+    def agg_comed(self, stacked):
+        if _gpu(stacked):
+            return ext().agg_comed(stacked)
+        return torch.median(stacked, dim=0).values
+
+    def agg_sign(self, stacked):
+        if _gpu(stacked):
+            return ext().agg_sign(stacked)
+        return torch.sign(torch.sign(stacked).sum(dim=0))
+
+    def _noise(self, cur_round, device, dtype):
+        g = torch_gen(self.args.seed, 'noise', cur_round)
+        n = torch.normal(mean=0.0, std=self.args.noise * self.args.clip,
+                         size=(self.n_params,), generator=g)
+        return n.to(device=device, dtype=dtype)
+
+    def _apply(self, global_model, lr_vector, agg):
+        """theta <- float32(theta + lr * agg) (reference aggregation.py:38-40)."""
+        p = global_model.flat_params
+        if _gpu(p):
+            ext().apply_update(p, agg,
+                               lr_vector if lr_vector is not None
+                               else torch.empty(0, device=p.device),
+                               float(self.server_lr))
+            return
+        if lr_vector is None:
+            new = p.double() + self.server_lr * agg
+        else:
+            new = p.double() + lr_vector.double() * agg
+        p.copy_(new.float())
+
+    def aggregate_buffers(self, global_model, buffer_deltas, agent_ids):
+        """FedAvg-BN: data-weighted mean of BatchNorm running-stat deltas
+        (build extension — the reference has no BN)."""
+        if global_model.n_buffers == 0 or not buffer_deltas:
+            return
+        w = self._weights(agent_ids, buffer_deltas.device
+                          if isinstance(buffer_deltas, torch.Tensor)
+                          else buffer_deltas[0].device)
+        if isinstance(buffer_deltas, list):
+            buffer_deltas = torch.stack(buffer_deltas)
+        mean = (buffer_deltas * (w / w.sum()).unsqueeze(1)).sum(dim=0)
+        global_model.flat_buffers.add_(mean.to(global_model.flat_buffers.dtype))
+
+    # ------------------------------------------------- server-side extras
+
+    def clip_updates(self, agent_updates_dict):
+        """Server-side L2 projection (reference aggregation.py:77-81;
+        disconnected from the main path there and here — the client does it
+        per batch, agent.py:54-60)."""
+        for update in agent_updates_dict.values():
+            l2 = torch.norm(update, p=2)
+            update.div_(max(1, l2 / self.args.clip))
+
+    def plot_norms(self, agent_updates_dict, cur_round, norm=2):
+        """Honest vs corrupt update-norm scalars (reference
+        aggregation.py:83-100)."""
+        honest, corrupt = [], []
+        for key, upd in agent_updates_dict.items():
+            (corrupt if key < self.args.num_corrupt else honest).append(upd)
+        if honest and self.writer:
+            avg = sum(torch.norm(u, p=norm) for u in honest) / len(honest)
+            self.writer.add_scalar(f'Norms/Avg_Honest_L{norm}', avg, cur_round)
+        if corrupt and self.writer:
+            avg = sum(torch.norm(u, p=norm) for u in corrupt) / len(corrupt)
+            self.writer.add_scalar(f'Norms/Avg_Corrupt_L{norm}', avg, cur_round)
+
+    # -------------------------------------------- Fisher diagnostics
+
+    def comp_diag_fisher(self, model_params, poisoned_eval, adv=True):
+        """Diagonal FIM over the poisoned val data (reference
+        aggregation.py:102-129; device handling fixed — the reference
+        instantiates the probe model on CPU even for GPU loaders,
+        SURVEY.md §7 quirks).  `poisoned_eval` is (X, Y) tensors.
+        Mirrors the reference's quirk of gathering LOGITS (not log-probs)
+        for the objective (aggregation.py:123)."""
+        from . import models as M
+        from .flatmodel import FlatParamModel
+        X, Y = poisoned_eval
+        model = M.get_model(self.args.data,
+                            getattr(self.args, 'model', None))
+        fm = FlatParamModel(model, X.device)
+        fm.load_vector(model_params.float())
+        fm.eval()
+        precision = torch.zeros_like(fm.flat_params)
+        bs = self.args.bs
+        n_total = X.shape[0]
+        for lo in range(0, n_total, bs):
+            inputs = X[lo:lo + bs]
+            labels = Y[lo:lo + bs].view(-1, 1)
+            if not adv:
+                labels = torch.full_like(labels, self.args.base_class)
+            fm.zero_grad()
+            outputs = fm(inputs)
+            F.log_softmax(outputs, dim=1)  # parity: computed, unused (ref quirk)
+            target_logits = outputs.gather(1, labels)
+            target_logits.sum().backward()
+            precision += (fm.flat_grads ** 2) / n_total
+        return precision.detach()
+
+    def plot_sign_agreement(self, robustLR, cur_global_params,
+                            new_global_params, cur_round):
+        """Sign-agreement diagnostics (reference aggregation.py:132-191):
+        intersect top-Fisher coords with RLR-maximized/minimized coords and
+        log 9 L2 scalars."""
+        import numpy as np
+        update = (new_global_params - cur_global_params)
+
+        fisher_adv = self.comp_diag_fisher(cur_global_params,
+                                           self.poisoned_val_loader)
+        fisher_hon = self.comp_diag_fisher(cur_global_params,
+                                           self.poisoned_val_loader, adv=False)
+        _, adv_idxs = fisher_adv.sort()
+        _, hon_idxs = fisher_hon.sort()
+        n_idxs = self.args.top_frac
+        adv_top = adv_idxs[-n_idxs:].cpu().numpy()
+        hon_top = hon_idxs[-n_idxs:].cpu().numpy()
+
+        min_idxs = (robustLR == -self.server_lr).nonzero().cpu().numpy()
+        max_idxs = (robustLR == self.server_lr).nonzero().cpu().numpy()
+
+        max_adv = np.intersect1d(adv_top, max_idxs)
+        max_hon = np.intersect1d(hon_top, max_idxs)
+        min_adv = np.intersect1d(adv_top, min_idxs)
+        min_hon = np.intersect1d(hon_top, min_idxs)
+
+        max_adv_only = np.setdiff1d(max_adv, max_hon)
+        max_hon_only = np.setdiff1d(max_hon, max_adv)
+        min_adv_only = np.setdiff1d(min_adv, min_hon)
+        min_hon_only = np.setdiff1d(min_hon, min_adv)
+
+        l2 = lambda idxs: torch.norm(update[idxs]).item()
+        max_adv_l2, max_hon_l2 = l2(max_adv_only), l2(max_hon_only)
+        min_adv_l2, min_hon_l2 = l2(min_adv_only), l2(min_hon_only)
+
+        if self.writer:
+            self.writer.add_scalar('Sign/Hon_Maxim_L2', max_hon_l2, cur_round)
+            self.writer.add_scalar('Sign/Adv_Maxim_L2', max_adv_l2, cur_round)
+            self.writer.add_scalar('Sign/Adv_Minim_L2', min_adv_l2, cur_round)
+            self.writer.add_scalar('Sign/Hon_Minim_L2', min_hon_l2, cur_round)
+        net_adv = max_adv_l2 - min_adv_l2
+        net_hon = max_hon_l2 - min_hon_l2
+        if self.writer:
+            self.writer.add_scalar('Sign/Adv_Net_L2', net_adv, cur_round)
+            self.writer.add_scalar('Sign/Hon_Net_L2', net_hon, cur_round)
+        self.cum_net_mov += (net_hon - net_adv)
+        if self.writer:
+            self.writer.add_scalar('Sign/Model_Net_L2_Cumulative',
+                                   self.cum_net_mov, cur_round)

@@ -1,0 +1,203 @@
+"""Dataset registry (reference: src/utils.py:95-124).
+
+Synthetic-first: the north-star benchmark measures on synthetic data /
+random-init weights (BASELINE.json), and this container has no network, so
+`get_datasets` defaults to deterministic synthetic tensors *shaped exactly
+like* the real datasets (fmnist 60k x 28x28 uint8, cifar10 50k x 32x32x3
+uint8 HWC, fed-emnist 3383 non-IID user shards of normalized floats).
+Synthetic images are class-prototype + noise so that both the classification
+task and the trojan backdoor are actually learnable (the defense-efficacy
+test depends on this).  When torchvision data is present on disk the real
+datasets load through the same ArrayDataset container.
+
+Raw storage conventions mirror the reference so the poison layer
+(utils.py:160-284 semantics) operates pre-normalization on uint8 for
+fmnist/cifar10 and on normalized floats for fedemnist.
+"""
+
+import os
+
+import numpy as np
+import torch
+
+from ..utils.rng import np_rng
+
+NORM_STATS = {
+    # mean/std per channel (reference utils.py:100-120)
+    'fmnist': ((0.2860,), (0.3530,)),
+    'cifar10': ((0.4914, 0.4822, 0.4465), (0.2023, 0.1994, 0.2010)),
+}
+
+
+class ArrayDataset:
+    """Torchvision-like container: raw uint8 `.data` (N,H,W) or (N,H,W,C),
+    LongTensor `.targets`, normalization applied at read time — but also
+    batch/whole-tensor materialization for the GPU-resident path."""
+
+    def __init__(self, data: torch.Tensor, targets: torch.Tensor, kind: str):
+        assert kind in NORM_STATS
+        self.data = data            # uint8, HWC (cifar) or HW (fmnist)
+        self.targets = targets.long()
+        self.kind = kind
+        mean, std = NORM_STATS[kind]
+        self._mean = torch.tensor(mean).view(1, -1, 1, 1)
+        self._std = torch.tensor(std).view(1, -1, 1, 1)
+
+    def __len__(self):
+        return self.data.shape[0]
+
+    def normalize(self, raw_u8: torch.Tensor) -> torch.Tensor:
+        """uint8 (B,H,W) or (B,H,W,C) -> normalized float32 NCHW."""
+        x = raw_u8.float().div_(255.0)
+        if x.dim() == 3:            # (B,H,W) -> (B,1,H,W)
+            x = x.unsqueeze(1)
+        else:                       # (B,H,W,C) -> (B,C,H,W)
+            x = x.permute(0, 3, 1, 2).contiguous()
+        mean = self._mean.to(x.device)
+        std = self._std.to(x.device)
+        return x.sub_(mean).div_(std)
+
+    def __getitem__(self, idx):
+        x = self.normalize(self.data[idx:idx + 1])[0]
+        return x, int(self.targets[idx])
+
+
+class TensorDataset:
+    """Fed-EMNIST-style container: already-normalized float `.inputs`
+    (N,1,H,W) + `.targets` (reference stores these in per-user .pt files)."""
+
+    def __init__(self, inputs: torch.Tensor, targets: torch.Tensor):
+        self.inputs = inputs
+        self.targets = targets.long()
+
+    def __len__(self):
+        return self.inputs.shape[0]
+
+    def __getitem__(self, idx):
+        return self.inputs[idx], int(self.targets[idx])
+
+
+class H5Dataset(TensorDataset):
+    """API-parity port of the reference's H5Dataset (utils.py:11-36):
+    wraps a {client_id: {'label','pixels'}} dict, NCHW-reshaped."""
+
+    def __init__(self, dataset, client_id):
+        targets = torch.LongTensor(dataset[client_id]['label'])
+        inputs = torch.Tensor(dataset[client_id]['pixels'])
+        s = inputs.shape
+        super().__init__(inputs.view(s[0], 1, s[1], s[2]), targets)
+
+    def classes(self):
+        return torch.unique(self.targets)
+
+    def __add__(self, other):
+        self.targets = torch.cat((self.targets, other.targets), 0)
+        self.inputs = torch.cat((self.inputs, other.inputs), 0)
+        return self
+
+    def to(self, device):
+        self.targets = self.targets.to(device)
+        self.inputs = self.inputs.to(device)
+
+
+# ---------------------------------------------------------------- synthetic
+
+def _synth_images(rng, n_per_class, n_classes, shape, noise=40.0):
+    """class prototype (fixed random uint8 image) + gaussian noise."""
+    protos = rng.integers(0, 256, size=(n_classes,) + shape)
+    imgs = np.empty((n_per_class * n_classes,) + shape, dtype=np.uint8)
+    targets = np.empty(n_per_class * n_classes, dtype=np.int64)
+    for c in range(n_classes):
+        lo = c * n_per_class
+        noise_block = rng.normal(0.0, noise, size=(n_per_class,) + shape)
+        imgs[lo:lo + n_per_class] = np.clip(
+            protos[c][None] + noise_block, 0, 255).astype(np.uint8)
+        targets[lo:lo + n_per_class] = c
+    # deterministic interleave so classes are mixed
+    perm = rng.permutation(len(targets))
+    return imgs[perm], targets[perm]
+
+
+def _synthetic_pair(kind, seed, train_n, val_n, n_classes=10):
+    shape = (28, 28) if kind == 'fmnist' else (32, 32, 3)
+    rng = np_rng(seed, 'data', 0 if kind == 'fmnist' else 1)
+    tr_img, tr_t = _synth_images(rng, train_n // n_classes, n_classes, shape)
+    va_img, va_t = _synth_images(rng, val_n // n_classes, n_classes, shape)
+    train = ArrayDataset(torch.from_numpy(tr_img), torch.from_numpy(tr_t), kind)
+    val = ArrayDataset(torch.from_numpy(va_img), torch.from_numpy(va_t), kind)
+    return train, val
+
+
+def _synthetic_fedemnist(seed, num_users, samples_per_user, val_n, n_classes=10):
+    """Non-IID writers: each user draws from 2-4 classes with its own style
+    offset; inputs are already-normalized floats like the reference's
+    pre-built .pt shards (utils.py:105-109)."""
+    mean, std = NORM_STATS['fmnist']
+    rng = np_rng(seed, 'data', 2)
+    protos = rng.normal(0.0, 1.0, size=(n_classes, 28, 28))
+    users = []
+    for u in range(num_users):
+        urng = np_rng(seed, 'data', 3, u)
+        k = int(urng.integers(2, 5))
+        classes = urng.choice(n_classes, k, replace=False)
+        t = urng.choice(classes, samples_per_user)
+        style = urng.normal(0.0, 0.3, size=(28, 28))
+        x = protos[t] + style[None] + urng.normal(0.0, 0.5, size=(len(t), 28, 28))
+        users.append(TensorDataset(
+            torch.from_numpy(x.astype(np.float32)).view(-1, 1, 28, 28),
+            torch.from_numpy(t.astype(np.int64))))
+    vrng = np_rng(seed, 'data', 4)
+    vt = vrng.integers(0, n_classes, size=val_n)
+    vx = protos[vt] + vrng.normal(0.0, 0.5, size=(val_n, 28, 28))
+    val = TensorDataset(torch.from_numpy(vx.astype(np.float32)).view(-1, 1, 28, 28),
+                        torch.from_numpy(vt.astype(np.int64)))
+    return users, val
+
+
+# ----------------------------------------------------------------- registry
+
+DEFAULT_SIZES = {
+    'fmnist': (60000, 10000),
+    'cifar10': (50000, 10000),
+    'fedemnist': (3383, 161),  # users, samples/user (reference scale ~341 avg)
+}
+
+
+def get_datasets(data, args=None, train_n=None, val_n=None, data_dir='../data'):
+    """Returns (train, val) — for fedemnist, (list-of-user TensorDatasets, val).
+
+    Synthetic unless real torchvision data is on disk and args.synthetic is
+    False (reference downloads at utils.py:100-122; this environment has no
+    network, so synthetic is the operational default)."""
+    seed = getattr(args, 'seed', 42) if args is not None else 42
+    synthetic = True if args is None else bool(getattr(args, 'synthetic', True))
+
+    if data in ('fmnist', 'cifar10'):
+        if not synthetic and _try_real_available(data, data_dir):
+            return _load_real(data, data_dir)
+        tn, vn = DEFAULT_SIZES[data]
+        return _synthetic_pair(data, seed, train_n or tn, val_n or vn)
+    elif data == 'fedemnist':
+        users, spu = DEFAULT_SIZES['fedemnist']
+        if args is not None:
+            users = getattr(args, 'num_agents', users) or users
+        return _synthetic_fedemnist(seed, train_n or users, spu, val_n or 10000)
+    raise ValueError(f"unknown dataset {data}")
+
+
+def _try_real_available(data, data_dir):
+    sub = {'fmnist': 'FashionMNIST', 'cifar10': 'cifar-10-batches-py'}[data]
+    return os.path.isdir(os.path.join(data_dir, sub))
+
+
+def _load_real(data, data_dir):
+    from torchvision import datasets as tvd
+    if data == 'fmnist':
+        tr = tvd.FashionMNIST(data_dir, train=True, download=False)
+        va = tvd.FashionMNIST(data_dir, train=False, download=False)
+        return (ArrayDataset(tr.data, tr.targets, 'fmnist'),
+                ArrayDataset(va.data, va.targets, 'fmnist'))
+    tr = tvd.CIFAR10(data_dir, train=True, download=False)
+    va = tvd.CIFAR10(data_dir, train=False, download=False)
+    return (ArrayDataset(torch.from_numpy(tr.data), torch.tensor(tr.targets), 'cifar10'),
+            ArrayDataset(torch.from_numpy(va.data), torch.tensor(va.targets), 'cifar10'))

@@ -1,0 +1,75 @@
+"""The reference model zoo (reference src/models.py:11-58), rebuilt on the
+rlr_amd op layer.  Parameter shapes, registration order and default init are
+identical to the reference (nn.Conv2d / nn.Linear containers), so the flat
+parameter vector is layout-compatible: CNN_MNIST = 1,199,882 params,
+CNN_CIFAR = 537,610 params (asserted in tests/test_models.py).
+
+forward() composes the fused conv+relu / linear+relu HIP ops; dropout uses
+the deterministic philox stream (ops.functional.DropoutCtx)."""
+
+import torch.nn as nn
+
+from ..ops import functional as Fo
+
+
+class _OpsModel(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.rng = Fo.DropoutCtx()
+
+    def set_dropout_seed(self, seed: int):
+        self.rng.reset(seed)
+
+
+class CNN_MNIST(_OpsModel):
+    """28x28x1: conv(1->32,3x3)+relu, conv(32->64,3x3)+relu, maxpool2,
+    flatten 9216, dropout .5, fc 9216->128+relu, dropout, fc 128->10
+    (reference models.py:11-31)."""
+
+    def __init__(self):
+        super().__init__()
+        self.conv1 = nn.Conv2d(1, 32, kernel_size=(3, 3))
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=(3, 3))
+        self.fc1 = nn.Linear(9216, 128)
+        self.fc2 = nn.Linear(128, 10)
+        self.p_drop = 0.5
+
+    def forward(self, x):
+        x = Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True)
+        x = Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True)
+        x = Fo.max_pool2d_2x2(x)
+        x = x.reshape(x.shape[0], -1)
+        x = Fo.dropout(x, self.p_drop, self.training, self.rng)
+        x = Fo.linear(x, self.fc1.weight, self.fc1.bias, relu=True)
+        x = Fo.dropout(x, self.p_drop, self.training, self.rng)
+        x = Fo.linear(x, self.fc2.weight, self.fc2.bias)
+        return x
+
+
+class CNN_CIFAR(_OpsModel):
+    """32x32x3: 3 x [conv3x3 (3->64->128->256)+relu+maxpool2], flatten 1024,
+    dropout-interleaved FCs 1024->128->256->10 (reference models.py:33-58;
+    the reference's `64*4*4` flatten literal equals 256*2*2 = 1024)."""
+
+    def __init__(self):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 64, 3)
+        self.conv2 = nn.Conv2d(64, 128, 3)
+        self.conv3 = nn.Conv2d(128, 256, 3)
+        self.fc1 = nn.Linear(1024, 128)
+        self.fc2 = nn.Linear(128, 256)
+        self.fc3 = nn.Linear(256, 10)
+        self.p_drop = 0.5
+
+    def forward(self, x):
+        x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True))
+        x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True))
+        x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv3.weight, self.conv3.bias, relu=True))
+        x = x.reshape(x.shape[0], -1)
+        x = Fo.dropout(x, self.p_drop, self.training, self.rng)
+        x = Fo.linear(x, self.fc1.weight, self.fc1.bias, relu=True)
+        x = Fo.dropout(x, self.p_drop, self.training, self.rng)
+        x = Fo.linear(x, self.fc2.weight, self.fc2.bias, relu=True)
+        x = Fo.dropout(x, self.p_drop, self.training, self.rng)
+        x = Fo.linear(x, self.fc3.weight, self.fc3.bias)
+        return x

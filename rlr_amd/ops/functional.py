@@ -1,0 +1,266 @@
+"""Autograd-visible ops.
+
+Every op has two implementations:
+  * CUDA(HIP) tensors -> a hand-written CDNA4 kernel from ops/csrc via a
+    torch.autograd.Function (required; loud failure if the extension is
+    missing — see ops.__init__).
+  * CPU tensors -> the plain torch.nn.functional op (GPU-less CI only).
+
+The HIP forward kernels fuse the activation where the reference composes
+relu(conv(...)) / relu(linear(...)) (reference models.py:22-28,47-56): the
+backward recovers the relu mask from the *post*-activation output (y>0).
+
+Dropout parity note: the reference uses nn.Dropout2d(p=0.5) on flattened
+2-D activations (models.py:17-19,40-44), where each "channel" is a single
+scalar — i.e. plain elementwise Bernoulli dropout.  This build implements
+elementwise dropout with a counter-based philox stream so masks are
+reproducible for any world size (SURVEY.md §7 hard part 3).
+"""
+
+import torch
+import torch.nn.functional as F
+
+from . import ext, force_eager
+
+
+def _gpu(x: torch.Tensor) -> bool:
+    return x.is_cuda and not force_eager()
+
+
+class DropoutCtx:
+    """Deterministic dropout stream: (seed64, call counter)."""
+
+    def __init__(self):
+        self.seed = 0
+        self.counter = 0
+
+    def reset(self, seed: int):
+        self.seed = int(seed)
+        self.counter = 0
+
+    def next_offset(self) -> int:
+        c = self.counter
+        self.counter += 1
+        return c
+
+
+# ----------------------------------------------------------------- conv2d
+
+class _ConvReLU(torch.autograd.Function):
+    """y = conv2d(x, w, stride, pad) + b, optionally relu-fused.
+    The reference only uses no-pad stride-1 3x3 (models.py:14-15,36-38);
+    stride/pad generality serves the build's ResNet18 extension."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride, pad, relu):
+        y = ext().conv2d_fwd(x, w, b, stride, pad, relu)
+        ctx.save_for_backward(x, w, y)
+        ctx.cfg = (stride, pad, relu, b is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        stride, pad, relu, has_b = ctx.cfg
+        if relu:
+            dy = ext().relu_bwd(y, dy)
+        dx, dw, db = ext().conv2d_bwd(x, w, dy, stride, pad, has_b)
+        return dx, dw, db, None, None, None
+
+
+def conv2d(x, w, b, stride=1, pad=0, relu=False):
+    if _gpu(x):
+        return _ConvReLU.apply(x, w, b, stride, pad, relu)
+    y = F.conv2d(x, w, b, stride=stride, padding=pad)
+    return F.relu(y) if relu else y
+
+
+# ----------------------------------------------------------------- linear
+
+class _LinearReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, relu):
+        y = ext().linear_fwd(x, w, b, relu)
+        ctx.save_for_backward(x, w, y)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        if ctx.relu:
+            dy = ext().relu_bwd(y, dy)
+        dx, dw, db = ext().linear_bwd(x, w, dy)
+        return dx, dw, db, None
+
+
+def linear(x, w, b, relu=False):
+    if _gpu(x):
+        return _LinearReLU.apply(x, w, b, relu)
+    y = F.linear(x, w, b)
+    return F.relu(y) if relu else y
+
+
+# ------------------------------------------------------------------- relu
+
+class _ReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = ext().relu_fwd(x)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        return ext().relu_bwd(y, dy)
+
+
+def relu(x):
+    if _gpu(x):
+        return _ReLU.apply(x)
+    return F.relu(x)
+
+
+# ------------------------------------------------------------- max_pool2d
+
+class _MaxPool2x2(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y, idx = ext().maxpool2x2_fwd(x)
+        ctx.save_for_backward(idx)
+        ctx.in_shape = x.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        return ext().maxpool2x2_bwd(dy.contiguous(), idx, list(ctx.in_shape))
+
+
+def max_pool2d_2x2(x):
+    """kernel 2, stride 2 — the reference's only pooling
+    (models.py:16,39)."""
+    if _gpu(x):
+        return _MaxPool2x2.apply(x)
+    return F.max_pool2d(x, 2, 2)
+
+
+# ---------------------------------------------------------------- dropout
+
+class _Dropout(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed, offset):
+        y, mask = ext().dropout_fwd(x, p, seed, offset)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        return ext().dropout_bwd(dy.contiguous(), mask, ctx.p), None, None, None
+
+
+def dropout(x, p, training, rng: DropoutCtx):
+    if not training or p == 0:
+        return x
+    if _gpu(x):
+        return _Dropout.apply(x, p, rng.seed, rng.next_offset())
+    from ..utils.rng import derive_seed
+    g = torch.Generator(device='cpu')
+    g.manual_seed(derive_seed(rng.seed, 'dropout', rng.next_offset()))
+    mask = (torch.rand(x.shape, generator=g) >= p).to(x.dtype)
+    return x * mask / (1.0 - p)
+
+
+# ---------------------------------------------------------- cross entropy
+
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        loss, softmax = ext().cross_entropy_fwd(logits, labels)
+        ctx.save_for_backward(softmax, labels)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        softmax, labels = ctx.saved_tensors
+        return ext().cross_entropy_bwd(softmax, labels, dloss), None
+
+
+def cross_entropy(logits, labels):
+    """Mean-reduced CE (the reference's criterion, federated.py:61)."""
+    if _gpu(logits):
+        return _CrossEntropy.apply(logits, labels)
+    return F.cross_entropy(logits, labels)
+
+
+# ------------------------------------------------------------- batch norm
+
+class _BatchNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, running_mean, running_var, momentum, eps,
+                training):
+        y, save_mean, save_rstd = ext().batchnorm_fwd(
+            x, w, b, running_mean, running_var, momentum, eps, training)
+        ctx.save_for_backward(x, w, save_mean, save_rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, save_mean, save_rstd = ctx.saved_tensors
+        dx, dw, db = ext().batchnorm_bwd(x, w, save_mean, save_rstd,
+                                         dy.contiguous())
+        return dx, dw, db, None, None, None, None, None
+
+
+def batch_norm(x, w, b, running_mean, running_var, momentum, eps, training):
+    if _gpu(x):
+        return _BatchNorm.apply(x, w, b, running_mean, running_var, momentum,
+                                eps, training)
+    return F.batch_norm(x, running_mean, running_var, w, b, training,
+                        momentum, eps)
+
+
+# -------------------------------------------------------- global avg pool
+
+class _GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_shape = x.shape
+        return ext().gap_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return ext().gap_bwd(dy.contiguous(), list(ctx.in_shape))
+
+
+def global_avg_pool(x):
+    """(N,C,H,W) -> (N,C) mean over HxW (build's ResNet18 head)."""
+    if _gpu(x):
+        return _GlobalAvgPool.apply(x)
+    return x.mean(dim=(2, 3))
+
+
+# -------------------------------------------------------------- add+relu
+
+class _AddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        y = ext().add_relu_fwd(a, b)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        d = ext().relu_bwd(y, dy)
+        return d, d
+
+
+def add_relu(a, b):
+    """relu(a + b) — the residual-join fusion for ResNet blocks."""
+    if _gpu(a):
+        return _AddReLU.apply(a, b)
+    return F.relu(a + b)
