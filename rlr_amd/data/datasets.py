@@ -102,16 +102,23 @@ class H5Dataset(TensorDataset):
 
 # ---------------------------------------------------------------- synthetic
 
-def _synth_images(rng, n_per_class, n_classes, shape, noise=40.0):
-    """class prototype (fixed random uint8 image) + gaussian noise."""
-    protos = rng.integers(0, 256, size=(n_classes,) + shape)
+def _synth_images(rng, protos, n_per_class, n_classes, shape, noise=60.0):
+    """Class prototype + per-sample brightness/contrast jitter + pixel
+    noise.  The jitter matters: it gives honest agents genuinely diverse
+    samples, so their per-coordinate gradient signs decorrelate the way
+    real FMNIST's do — with pure prototype+noise data honest gradients
+    align perfectly and the RLR vote semantics invert (the defense-efficacy
+    test pins the intended dynamics)."""
     imgs = np.empty((n_per_class * n_classes,) + shape, dtype=np.uint8)
     targets = np.empty(n_per_class * n_classes, dtype=np.int64)
+    bshape = (n_per_class,) + (1,) * len(shape)
     for c in range(n_classes):
         lo = c * n_per_class
-        noise_block = rng.normal(0.0, noise, size=(n_per_class,) + shape)
-        imgs[lo:lo + n_per_class] = np.clip(
-            protos[c][None] + noise_block, 0, 255).astype(np.uint8)
+        brightness = rng.normal(0.0, 40.0, size=bshape)
+        contrast = rng.uniform(0.6, 1.4, size=bshape)
+        x = ((protos[c][None] - 128.0) * contrast + 128.0 + brightness
+             + rng.normal(0.0, noise, size=(n_per_class,) + shape))
+        imgs[lo:lo + n_per_class] = np.clip(x, 0, 255).astype(np.uint8)
         targets[lo:lo + n_per_class] = c
     # deterministic interleave so classes are mixed
     perm = rng.permutation(len(targets))
@@ -121,8 +128,14 @@ def _synth_images(rng, n_per_class, n_classes, shape, noise=40.0):
 def _synthetic_pair(kind, seed, train_n, val_n, n_classes=10):
     shape = (28, 28) if kind == 'fmnist' else (32, 32, 3)
     rng = np_rng(seed, 'data', 0 if kind == 'fmnist' else 1)
-    tr_img, tr_t = _synth_images(rng, train_n // n_classes, n_classes, shape)
-    va_img, va_t = _synth_images(rng, val_n // n_classes, n_classes, shape)
+    # prototypes = shared base image + per-class deltas, shared by train AND
+    # val: classes share most structure (like real image sets) but are
+    # separably labeled
+    base = rng.integers(60, 196, size=shape).astype(np.float64)
+    deltas = rng.normal(0.0, 35.0, size=(n_classes,) + shape)
+    protos = np.clip(base[None] + deltas, 0, 255)
+    tr_img, tr_t = _synth_images(rng, protos, train_n // n_classes, n_classes, shape)
+    va_img, va_t = _synth_images(rng, protos, val_n // n_classes, n_classes, shape)
     train = ArrayDataset(torch.from_numpy(tr_img), torch.from_numpy(tr_t), kind)
     val = ArrayDataset(torch.from_numpy(va_img), torch.from_numpy(va_t), kind)
     return train, val
