@@ -59,7 +59,7 @@ class Agent:
             return
         ds = self.train_dataset
         if isinstance(ds, DatasetSplit):
-            idx = torch.as_tensor(ds.idxs)
+            idx = torch.as_tensor(ds.idxs, device=ds.dataset.data.device)
             raw = ds.dataset.data[idx]
             self._X = ds.dataset.normalize(raw).to(self.device, non_blocking=True)
             self._Y = ds.dataset.targets[idx].to(self.device, non_blocking=True)

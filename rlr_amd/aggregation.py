@@ -57,6 +57,21 @@ class Aggregation:
             stacked = agent_updates
             assert agent_ids is not None
 
+        if _gpu(stacked) and self.args.aggr == 'avg':
+            # headline path: ONE fused kernel does sign-vote + weighted avg
+            # + noise + fp32 apply in a single pass over the K x n matrix
+            from .utils.rng import derive_seed
+            w = self._weights(agent_ids, stacked.device)
+            noise_std = (self.args.noise * self.args.clip
+                         if self.args.noise > 0 else 0.0)
+            seed = derive_seed(self.args.seed, 'noise', cur_round)
+            ext().fused_avg_rlr_apply(
+                stacked, w, global_model.flat_params,
+                self.args.robustLR_threshold > 0,
+                float(self.args.robustLR_threshold), float(self.server_lr),
+                float(noise_std), seed, 0, False)
+            return
+
         lr_vector = None
         if self.args.robustLR_threshold > 0:
             lr_vector = self.compute_robustLR(stacked)

@@ -47,7 +47,14 @@ class ArrayDataset:
         return self.data.shape[0]
 
     def normalize(self, raw_u8: torch.Tensor) -> torch.Tensor:
-        """uint8 (B,H,W) or (B,H,W,C) -> normalized float32 NCHW."""
+        """uint8 (B,H,W) or (B,H,W,C) -> normalized float32 NCHW.
+        On CUDA(HIP) this is one fused HIP kernel (poison.hip
+        normalize_u8): layout change + /255 + mean/std in a single pass."""
+        if raw_u8.is_cuda:
+            from ..ops import ext
+            mean = self._mean.reshape(-1).to(raw_u8.device)
+            std = self._std.reshape(-1).to(raw_u8.device)
+            return ext().normalize_u8(raw_u8.contiguous(), mean, std)
         x = raw_u8.float().div_(255.0)
         if x.dim() == 3:            # (B,H,W) -> (B,1,H,W)
             x = x.unsqueeze(1)

@@ -20,8 +20,9 @@ class DatasetSplit:
     def __init__(self, dataset, idxs):
         self.dataset = dataset
         self.idxs = list(idxs)
-        self.targets = torch.tensor(
-            [int(dataset.targets[i]) for i in self.idxs])
+        sel = torch.as_tensor(self.idxs, dtype=torch.long,
+                              device=dataset.targets.device)
+        self.targets = dataset.targets[sel]
 
     def classes(self):
         return torch.unique(self.targets)

@@ -155,6 +155,33 @@ def apply_pattern_(raw: torch.Tensor, spec: PatternSpec):
         raise ValueError(spec.mode)
 
 
+def apply_pattern_indexed_(storage: torch.Tensor, idx_t: torch.Tensor,
+                           spec: PatternSpec):
+    """Apply a pattern to storage[idx] in place.  On CUDA(HIP) tensors this
+    is the on-device trojan kernel (ops/csrc/poison.hip) — raw data stays in
+    HBM; the CPU path gathers/patches/scatters with torch ops."""
+    if storage.is_cuda:
+        from ..ops import ext
+        if spec.mode == 'set':
+            coords = torch.as_tensor(spec.coords, dtype=torch.int32,
+                                     device=storage.device).contiguous()
+            if storage.dtype == torch.uint8:
+                ext().poison_set_u8(storage, idx_t, coords, int(spec.value))
+            else:
+                ext().poison_set_f32(storage, idx_t, coords,
+                                     float(spec.value))
+        elif spec.mode == 'add_wrap_u8':
+            mask = torch.as_tensor(spec.mask, device=storage.device)
+            ext().poison_addwrap_u8(storage, idx_t, mask)
+        else:  # sub_float
+            mask = torch.as_tensor(spec.mask, device=storage.device)
+            ext().poison_subf(storage, idx_t, mask)
+        return
+    batch = storage[idx_t]
+    apply_pattern_(batch, spec)
+    storage[idx_t] = batch
+
+
 def poison_dataset(dataset, args, data_idxs=None, poison_all=False,
                    agent_idx=-1):
     """Reference utils.py:160-178 semantics with a deterministic,
@@ -179,8 +206,6 @@ def poison_dataset(dataset, args, data_idxs=None, poison_all=False,
     spec = pattern_spec(args.data, args.pattern_type, agent_idx)
     storage = dataset.inputs if hasattr(dataset, 'inputs') else dataset.data
     idx_t = torch.as_tensor(poison_idxs, device=storage.device)
-    batch = storage[idx_t]
-    apply_pattern_(batch, spec)
-    storage[idx_t] = batch
-    dataset.targets[idx_t] = args.target_class
+    apply_pattern_indexed_(storage, idx_t, spec)
+    dataset.targets[idx_t.to(dataset.targets.device)] = args.target_class
     return poison_idxs.tolist()

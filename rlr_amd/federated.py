@@ -54,6 +54,14 @@ def build_world(args):
         X_pv, Y_pv = materialize_eval_set(pv, idxs=idxs, device=device)
     else:
         train_dataset, val_dataset = get_datasets(args.data, args)
+        if device.type == 'cuda':
+            # raw uint8 storage moves to HBM BEFORE poisoning: trojan
+            # injection runs on-device and poisoned batches never leave HBM
+            # (BASELINE.json; ops/csrc/poison.hip)
+            train_dataset.data = train_dataset.data.to(device)
+            train_dataset.targets = train_dataset.targets.to(device)
+            val_dataset.data = val_dataset.data.to(device)
+            val_dataset.targets = val_dataset.targets.to(device)
         user_groups = distribute_data(train_dataset, args)
         agents, agent_data_sizes = [], {}
         for _id in range(args.num_agents):

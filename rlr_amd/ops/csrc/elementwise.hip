@@ -1,0 +1,346 @@
+// Elementwise / pooling / dropout / CE-loss kernels (gfx950).
+// Covers SURVEY.md §2b K3 (relu), K4 (maxpool2x2), K5 (dropout, philox),
+// K7 (cross-entropy), K18 (eval reductions) plus the build's add_relu and
+// global-avg-pool for ResNet18.
+// All memory-bound: float4-vectorized grid-stride loops at the HBM roofline
+// (cdna_hip_programming.md Appendix B, Guideline 13).
+#include "common.h"
+
+// ---------------------------------------------------------------- relu
+
+__global__ void relu_fwd_k(const float* __restrict__ x, float* __restrict__ y,
+                           long n) {
+  long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long i = i4; i < n4; i += stride) {
+    float4 v = ((const float4*)x)[i];
+    v.x = fmaxf(v.x, 0.f); v.y = fmaxf(v.y, 0.f);
+    v.z = fmaxf(v.z, 0.f); v.w = fmaxf(v.w, 0.f);
+    ((float4*)y)[i] = v;
+  }
+  for (long i = n4 * 4 + i4; i < n; i += stride) y[i] = fmaxf(x[i], 0.f);
+}
+
+// dx = dy * (y > 0): the mask comes from the POST-activation output, which
+// is what the fused conv/linear+relu forwards save.
+__global__ void relu_bwd_k(const float* __restrict__ y,
+                           const float* __restrict__ dy,
+                           float* __restrict__ dx, long n) {
+  long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long i = i4; i < n4; i += stride) {
+    float4 a = ((const float4*)y)[i];
+    float4 g = ((const float4*)dy)[i];
+    g.x = a.x > 0.f ? g.x : 0.f; g.y = a.y > 0.f ? g.y : 0.f;
+    g.z = a.z > 0.f ? g.z : 0.f; g.w = a.w > 0.f ? g.w : 0.f;
+    ((float4*)dx)[i] = g;
+  }
+  for (long i = n4 * 4 + i4; i < n; i += stride)
+    dx[i] = y[i] > 0.f ? dy[i] : 0.f;
+}
+
+__global__ void add_relu_k(const float* __restrict__ a,
+                           const float* __restrict__ b,
+                           float* __restrict__ y, long n) {
+  long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long i = i4; i < n4; i += stride) {
+    float4 va = ((const float4*)a)[i];
+    float4 vb = ((const float4*)b)[i];
+    va.x = fmaxf(va.x + vb.x, 0.f); va.y = fmaxf(va.y + vb.y, 0.f);
+    va.z = fmaxf(va.z + vb.z, 0.f); va.w = fmaxf(va.w + vb.w, 0.f);
+    ((float4*)y)[i] = va;
+  }
+  for (long i = n4 * 4 + i4; i < n; i += stride)
+    y[i] = fmaxf(a[i] + b[i], 0.f);
+}
+
+extern "C" {
+void launch_relu_fwd(const float* x, float* y, long n, void* s) {
+  relu_fwd_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(x, y, n);
+}
+void launch_relu_bwd(const float* y, const float* dy, float* dx, long n,
+                     void* s) {
+  relu_bwd_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(y, dy, dx, n);
+}
+void launch_add_relu(const float* a, const float* b, float* y, long n,
+                     void* s) {
+  add_relu_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(a, b, y, n);
+}
+}
+
+// ------------------------------------------------------------ maxpool 2x2
+
+// One thread per output element; idx stores the 2x2 argmax (0..3) for the
+// gather-style backward (no zero-init, write-once).
+__global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
+                                 float* __restrict__ y,
+                                 uint8_t* __restrict__ idx,
+                                 int NC, int H, int W, int OH, int OW) {
+  long n_out = (long)NC * OH * OW;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += stride) {
+    int ow = i % OW;
+    int oh = (i / OW) % OH;
+    long nc = i / ((long)OW * OH);
+    const float* p = x + (nc * H + 2 * oh) * W + 2 * ow;
+    float v00 = p[0], v01 = p[1], v10 = p[W], v11 = p[W + 1];
+    float m = v00; uint8_t a = 0;
+    if (v01 > m) { m = v01; a = 1; }
+    if (v10 > m) { m = v10; a = 2; }
+    if (v11 > m) { m = v11; a = 3; }
+    y[i] = m;
+    idx[i] = a;
+  }
+}
+
+// Gather form: one thread per INPUT element; contributes dy iff it was the
+// argmax of its (unique) window.  Odd H/W tails (never pooled) get 0.
+__global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
+                                 const uint8_t* __restrict__ idx,
+                                 float* __restrict__ dx,
+                                 int NC, int H, int W, int OH, int OW) {
+  long n_in = (long)NC * H * W;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_in;
+       i += stride) {
+    int iw = i % W;
+    int ih = (i / W) % H;
+    long nc = i / ((long)W * H);
+    int oh = ih >> 1, ow = iw >> 1;
+    float g = 0.f;
+    if (oh < OH && ow < OW) {
+      long o = (nc * OH + oh) * OW + ow;
+      uint8_t a = ((ih & 1) << 1) | (iw & 1);
+      if (idx[o] == a) g = dy[o];
+    }
+    dx[i] = g;
+  }
+}
+
+extern "C" {
+void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, int NC,
+                           int H, int W, int OH, int OW, void* s) {
+  long n = (long)NC * OH * OW;
+  maxpool2x2_fwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, idx, NC,
+                                                               H, W, OH, OW);
+}
+void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
+                           int NC, int H, int W, int OH, int OW, void* s) {
+  long n = (long)NC * H * W;
+  maxpool2x2_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx,
+                                                               NC, H, W, OH,
+                                                               OW);
+}
+}
+
+// ---------------------------------------------------------------- dropout
+
+// Elementwise Bernoulli(1-p) * 1/(1-p); 4 elements per philox draw.
+__global__ void dropout_fwd_k(const float* __restrict__ x,
+                              float* __restrict__ y,
+                              uint8_t* __restrict__ mask, long n, float p,
+                              uint64_t seed, uint64_t offset) {
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = (n + 3) / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    Philox4 r = philox4(seed, offset, (uint32_t)i);
+    uint32_t rv[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long k = i * 4 + j;
+      if (k < n) {
+        bool keep = u32_to_uniform(rv[j]) >= p;
+        mask[k] = keep;
+        y[k] = keep ? x[k] * scale : 0.f;
+      }
+    }
+  }
+}
+
+__global__ void dropout_bwd_k(const float* __restrict__ dy,
+                              const uint8_t* __restrict__ mask,
+                              float* __restrict__ dx, long n, float p) {
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dx[i] = mask[i] ? dy[i] * scale : 0.f;
+}
+
+extern "C" {
+void launch_dropout_fwd(const float* x, float* y, uint8_t* mask, long n,
+                        float p, uint64_t seed, uint64_t offset, void* s) {
+  dropout_fwd_k<<<grid_for((n + 3) / 4), kBlock, 0, (hipStream_t)s>>>(
+      x, y, mask, n, p, seed, offset);
+}
+void launch_dropout_bwd(const float* dy, const uint8_t* mask, float* dx,
+                        long n, float p, void* s) {
+  dropout_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, mask, dx, n,
+                                                            p);
+}
+}
+
+// ---------------------------------------------------------- global avgpool
+
+__global__ void gap_fwd_k(const float* __restrict__ x, float* __restrict__ y,
+                          int NC, int HW) {
+  // one wave per (n,c) plane
+  int nc = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
+  if (nc >= NC) return;
+  const float* p = x + (long)nc * HW;
+  float acc = 0.f;
+  for (int i = lane; i < HW; i += kWave) acc += p[i];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) y[nc] = acc / HW;
+}
+
+__global__ void gap_bwd_k(const float* __restrict__ dy,
+                          float* __restrict__ dx, int NC, int HW) {
+  long n = (long)NC * HW;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dx[i] = dy[i / HW] / HW;
+}
+
+extern "C" {
+void launch_gap_fwd(const float* x, float* y, int NC, int HW, void* s) {
+  int wpb = kBlock / kWave;
+  gap_fwd_k<<<(NC + wpb - 1) / wpb, kBlock, 0, (hipStream_t)s>>>(x, y, NC, HW);
+}
+void launch_gap_bwd(const float* dy, float* dx, int NC, int HW, void* s) {
+  gap_bwd_k<<<grid_for((long)NC * HW), kBlock, 0, (hipStream_t)s>>>(dy, dx,
+                                                                    NC, HW);
+}
+}
+
+// ------------------------------------------------------------ cross entropy
+
+// Mean-reduced CE over (B, C<=64) logits.  One wave per row: lane c holds
+// logit c; wave shuffle reduce for max and sum.  Saves softmax for bwd.
+__global__ void ce_fwd_k(const float* __restrict__ logits,
+                         const long* __restrict__ labels,
+                         float* __restrict__ softmax,
+                         float* __restrict__ row_loss, int B, int C) {
+  int row = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
+  if (row >= B) return;
+  float v = (lane < C) ? logits[(long)row * C + lane] : -3.4e38f;
+  float m = v;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    m = fmaxf(m, __shfl_xor(m, off, kWave));
+  float e = (lane < C) ? __expf(v - m) : 0.f;
+  float sum = e;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    sum += __shfl_xor(sum, off, kWave);
+  float sm = e / sum;
+  if (lane < C) softmax[(long)row * C + lane] = sm;
+  if (lane == 0) {
+    long t = labels[row];
+    float lt = logits[(long)row * C + t];
+    row_loss[row] = logf(sum) + m - lt;
+  }
+}
+
+// single-block deterministic mean over row losses
+__global__ void ce_reduce_k(const float* __restrict__ row_loss,
+                            float* __restrict__ out, int B) {
+  __shared__ float sh[kBlock];
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < B; i += blockDim.x) acc += row_loss[i];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = kBlock / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[0] = sh[0] / B;
+}
+
+// dlogits = dloss * (softmax - onehot) / B
+__global__ void ce_bwd_k(const float* __restrict__ softmax,
+                         const long* __restrict__ labels,
+                         const float* __restrict__ dloss,
+                         float* __restrict__ dlogits, int B, int C) {
+  long n = (long)B * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  float g = dloss[0];
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long row = i / C;
+    float v = softmax[i] - (labels[row] == c ? 1.f : 0.f);
+    dlogits[i] = g * v / B;
+  }
+}
+
+extern "C" {
+void launch_ce_fwd(const float* logits, const long* labels, float* softmax,
+                   float* row_loss, float* loss_out, int B, int C, void* s) {
+  int wpb = kBlock / kWave;
+  ce_fwd_k<<<(B + wpb - 1) / wpb, kBlock, 0, (hipStream_t)s>>>(
+      logits, labels, softmax, row_loss, B, C);
+  ce_reduce_k<<<1, kBlock, 0, (hipStream_t)s>>>(row_loss, loss_out, B);
+}
+void launch_ce_bwd(const float* softmax, const long* labels,
+                   const float* dloss, float* dlogits, int B, int C,
+                   void* s) {
+  ce_bwd_k<<<grid_for((long)B * C), kBlock, 0, (hipStream_t)s>>>(
+      softmax, labels, dloss, dlogits, B, C);
+}
+}
+
+// ------------------------------------------------------------- eval fused
+
+// Fused eval reductions: per-row CE loss + argmax + confusion scatter.
+// Confusion counts are integer-valued float atomicAdds (exact, order-free);
+// the loss sum is reduced single-block for determinism.
+__global__ void eval_update_k(const float* __restrict__ logits,
+                              const long* __restrict__ labels,
+                              float* __restrict__ conf,
+                              double* __restrict__ loss_sum, int B, int C,
+                              int num_classes) {
+  __shared__ double sh[kBlock];
+  double acc = 0.0;
+  for (int row = threadIdx.x; row < B; row += blockDim.x) {
+    const float* l = logits + (long)row * C;
+    float m = l[0];
+    int arg = 0;
+    for (int c = 1; c < C; ++c)
+      if (l[c] > m) { m = l[c]; arg = c; }
+    float sum = 0.f;
+    for (int c = 0; c < C; ++c) sum += __expf(l[c] - m);
+    long t = labels[row];
+    acc += (double)(logf(sum) + m - l[t]);
+    atomicAdd(&conf[t * num_classes + arg], 1.0f);
+  }
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = kBlock / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) loss_sum[0] += sh[0];
+}
+
+extern "C" {
+void launch_eval_update(const float* logits, const long* labels, float* conf,
+                        double* loss_sum, int B, int C, int num_classes,
+                        void* s) {
+  eval_update_k<<<1, kBlock, 0, (hipStream_t)s>>>(logits, labels, conf,
+                                                  loss_sum, B, C,
+                                                  num_classes);
+}
+}

@@ -1,0 +1,260 @@
+// MFMA fp32 GEMM for gfx950 (SURVEY.md §2b K6 — Linear fwd/bwd; also the
+// inner engine shape for the conv kernels).
+//
+// C[M,N] = A[M,K] x B[K,N], all row-major, exact fp32 numerics via
+// v_mfma_f32_16x16x4_f32 (f32-in/f32-acc MFMA at the 157 TF f32 vector
+// rate — cdna_hip_programming.md §3 "FP32-input MFMA"; there is no
+// xf32/TF32 on gfx950 and this is bitwise an fmaf chain).
+//
+// Structure: 256-thread block = 4 waves (2x2), block tile BM=128 x BN=64,
+// K-step BK=32, double-buffered LDS:
+//   A_lds[2][BM][BK+2]  — +2 pad makes the b32 fragment read (lane groups
+//                         {m0..m0+15} x k, bank = (2m+k) mod 32) conflict-
+//                         free across the two 16-lane halves of a group
+//   B_lds[2][BK][BN+16] — +16 pad shifts consecutive k rows by 16 banks
+// Deterministic split-K for small-M*N / large-K shapes (fc1: M=256, N=128,
+// K=9216 — 4 output tiles would leave 252 CUs idle): gridDim.z K-chunks
+// write fp32 partial slabs, a fixed-order reduce kernel (fused bias+relu)
+// combines them.  No atomics anywhere: bitwise run-to-run reproducible.
+#include "common.h"
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int BM = 128, BN = 64, BK = 32;
+constexpr int LDA_S = BK + 2;   // A_lds row stride (floats)
+constexpr int LDB_S = BN + 16;  // B_lds row stride (floats)
+
+// Staging guards handle M/N/K edges by zero-fill; VEC selects float4 global
+// loads (requires 16B-aligned rows: ld % 4 == 0).
+template <bool VEC>
+__global__ __launch_bounds__(256)
+void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
+                float* __restrict__ C, const float* __restrict__ bias,
+                int M, int N, int K, int lda, int ldb, int ldc,
+                long k_per_chunk, int relu, int direct_out) {
+  __shared__ float A_lds[2][BM * LDA_S];
+  __shared__ float B_lds[2][BK * LDB_S];
+
+  const int m_blk = blockIdx.x * BM;
+  const int n_blk = blockIdx.y * BN;
+  const long k_lo = (long)blockIdx.z * k_per_chunk;
+  const long k_hi = min((long)K, k_lo + k_per_chunk);
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;       // 2x2 wave grid
+  const int l15 = lane & 15, l4 = lane >> 4;     // fragment coords
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging coordinates (per thread, fixed across k-tiles)
+  const int am = t >> 3;            // 0..31 (+32 per round, 4 rounds)
+  const int ak = (t & 7) * 4;       // 0,4,..,28
+  const int bk = t >> 4;            // 0..15 (+16 per round, 2 rounds)
+  const int bn = (t & 15) * 4;      // 0,4,..,60
+
+  auto stage = [&](int buf, long k0) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int m = am + j * 32;
+      long gm = m_blk + m;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (gm < M) {
+        long base = gm * (long)lda + k0 + ak;
+        if (VEC && k0 + ak + 3 < k_hi) {
+          const float4 q = *(const float4*)(A + base);
+          v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+        } else {
+          if (k0 + ak + 0 < k_hi) v0 = A[base + 0];
+          if (k0 + ak + 1 < k_hi) v1 = A[base + 1];
+          if (k0 + ak + 2 < k_hi) v2 = A[base + 2];
+          if (k0 + ak + 3 < k_hi) v3 = A[base + 3];
+        }
+      }
+      float* dst = &A_lds[buf][m * LDA_S + ak];
+      ((float2*)dst)[0] = {v0, v1};
+      ((float2*)dst)[1] = {v2, v3};
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int kk = bk + j * 16;
+      long gk = k0 + kk;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (gk < k_hi) {
+        long base = gk * (long)ldb + n_blk + bn;
+        if (VEC && n_blk + bn + 3 < N) {
+          const float4 q = *(const float4*)(B + base);
+          v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+        } else {
+          if (n_blk + bn + 0 < N) v0 = B[base + 0];
+          if (n_blk + bn + 1 < N) v1 = B[base + 1];
+          if (n_blk + bn + 2 < N) v2 = B[base + 2];
+          if (n_blk + bn + 3 < N) v3 = B[base + 3];
+        }
+      }
+      *(float4*)&B_lds[buf][kk * LDB_S + bn] = {v0, v1, v2, v3};
+    }
+  };
+
+  stage(0, k_lo);
+  __syncthreads();
+
+  int buf = 0;
+  for (long k0 = k_lo; k0 < k_hi; k0 += BK) {
+    if (k0 + BK < k_hi) stage(buf ^ 1, k0 + BK);
+    const float* Abuf = A_lds[buf];
+    const float* Bbuf = B_lds[buf];
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      float a_frag[4], b_frag[2];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a_frag[mi] = Abuf[(wr * 64 + mi * 16 + l15) * LDA_S + kk * 4 + l4];
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        b_frag[ni] = Bbuf[(kk * 4 + l4) * LDB_S + wc * 32 + ni * 16 + l15];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // epilogue: C/D mapping col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      int col = n_blk + wc * 32 + ni * 16 + l15;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m_blk + wr * 64 + mi * 16 + l4 * 4 + r;
+        if (row >= M) continue;
+        float v = acc[mi][ni][r];
+        if (direct_out) {
+          if (bias) v += bias[col];
+          if (relu) v = fmaxf(v, 0.f);
+          C[(long)row * ldc + col] = v;
+        } else {
+          // split-K partial slab: [z][M][N] dense
+          C[((long)blockIdx.z * M + row) * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+// fixed-order split-K reduce + bias + relu
+__global__ void splitk_reduce_k(const float* __restrict__ ws,
+                                float* __restrict__ C,
+                                const float* __restrict__ bias, int M, int N,
+                                int ldc, int SK, int relu) {
+  long n_out = (long)M * N;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += stride) {
+    float acc = 0.f;
+    for (int z = 0; z < SK; ++z) acc += ws[(long)z * n_out + i];
+    int col = i % N;
+    if (bias) acc += bias[col];
+    if (relu) acc = fmaxf(acc, 0.f);
+    C[(i / N) * (long)ldc + col] = acc;
+  }
+}
+
+// 32x32 LDS-tiled transpose: out[N,M] = in[M,N]^T
+__global__ void transpose_f32_k(const float* __restrict__ in,
+                                float* __restrict__ out, int M, int N) {
+  __shared__ float tile[32][33];
+  int bx = blockIdx.x * 32, by = blockIdx.y * 32;
+  int x = bx + (threadIdx.x & 31);
+  int y0 = by + (threadIdx.x >> 5) * 4;
+  for (int j = 0; j < 4; ++j) {
+    int y = y0 + j;
+    if (y < M && x < N)
+      tile[threadIdx.x & 31][(threadIdx.x >> 5) * 4 + j] =
+          in[(long)y * N + x];
+  }
+  __syncthreads();
+  int ox = by + (threadIdx.x & 31);
+  int oy0 = bx + (threadIdx.x >> 5) * 4;
+  for (int j = 0; j < 4; ++j) {
+    int oy = oy0 + j;
+    if (oy < N && ox < M)
+      out[(long)oy * M + ox] = tile[(threadIdx.x >> 5) * 4 + j]
+                                   [threadIdx.x & 31];
+  }
+}
+
+// column sum: db[n] = sum_m dY[m][n] (bias gradient)
+__global__ void colsum_k(const float* __restrict__ dY,
+                         float* __restrict__ db, int M, int N) {
+  for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int m = 0; m < M; ++m) acc += dY[(long)m * N + n];
+    db[n] = acc;
+  }
+}
+
+extern "C" {
+
+// Heuristic split-K: fill the chip (256 CUs) when the tile grid is small.
+int gemm_f32_splitk(int M, int N, int K) {
+  long tiles = ((M + BM - 1) / BM) * (long)((N + BN - 1) / BN);
+  if (tiles >= 192 || K <= 2 * BK) return 1;
+  long want = (256 + tiles - 1) / tiles;
+  long max_chunks = (K + BK - 1) / BK;
+  long sk = want < max_chunks ? want : max_chunks;
+  return (int)(sk < 1 ? 1 : (sk > 128 ? 128 : sk));
+}
+
+// ws: null unless SK>1, then SK*M*N floats.
+void launch_gemm_f32(const float* A, const float* B, float* C,
+                     const float* bias, float* ws, int M, int N, int K,
+                     int lda, int ldb, int ldc, int SK, int relu, void* s) {
+  hipStream_t st = (hipStream_t)s;
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, SK);
+  long k_per_chunk = SK == 1 ? (long)K
+                             : ((((long)K + SK - 1) / SK + BK - 1) / BK) * BK;
+  bool vec = (lda % 4 == 0) && (ldb % 4 == 0);
+  float* out = SK == 1 ? C : ws;
+  if (vec)
+    gemm_f32_k<true><<<grid, 256, 0, st>>>(A, B, out, bias, M, N, K, lda,
+                                           ldb, ldc, k_per_chunk, relu,
+                                           SK == 1);
+  else
+    gemm_f32_k<false><<<grid, 256, 0, st>>>(A, B, out, bias, M, N, K, lda,
+                                            ldb, ldc, k_per_chunk, relu,
+                                            SK == 1);
+  if (SK > 1)
+    splitk_reduce_k<<<grid_for((long)M * N), kBlock, 0, st>>>(
+        ws, C, bias, M, N, ldc, SK, relu);
+}
+
+void launch_splitk_reduce(const float* ws, float* C, const float* bias,
+                          int M, int N, int ldc, int SK, int relu, void* s) {
+  splitk_reduce_k<<<grid_for((long)M * N), kBlock, 0, (hipStream_t)s>>>(
+      ws, C, bias, M, N, ldc, SK, relu);
+}
+
+void launch_transpose_f32(const float* in, float* out, int M, int N,
+                          void* s) {
+  dim3 grid((N + 31) / 32, (M + 31) / 32);
+  transpose_f32_k<<<grid, 256, 0, (hipStream_t)s>>>(in, out, M, N);
+}
+
+void launch_colsum(const float* dY, float* db, int M, int N, void* s) {
+  colsum_k<<<grid_for(N, kBlock, 64), kBlock, 0, (hipStream_t)s>>>(dY, db, M,
+                                                                   N);
+}
+}

@@ -64,7 +64,8 @@ class _ConvReLU(torch.autograd.Function):
         stride, pad, relu, has_b = ctx.cfg
         if relu:
             dy = ext().relu_bwd(y, dy)
-        dx, dw, db = ext().conv2d_bwd(x, w, dy, stride, pad, has_b)
+        dx, dw, db = ext().conv2d_bwd(x, w, dy, stride, pad, has_b,
+                                      ctx.needs_input_grad[0])
         return dx, dw, db, None, None, None
 
 

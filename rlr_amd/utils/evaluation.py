@@ -52,13 +52,13 @@ def materialize_eval_set(dataset, idxs=None, device='cpu'):
     if hasattr(dataset, 'inputs'):  # TensorDataset
         X, Y = dataset.inputs, dataset.targets
         if idxs is not None:
-            sel = torch.as_tensor(list(idxs))
+            sel = torch.as_tensor(list(idxs), device=X.device)
             X, Y = X[sel], Y[sel]
     else:
         if idxs is None:
             raw, Y = dataset.data, dataset.targets
         else:
-            sel = torch.as_tensor(list(idxs))
+            sel = torch.as_tensor(list(idxs), device=dataset.data.device)
             raw, Y = dataset.data[sel], dataset.targets[sel]
         X = dataset.normalize(raw)
     return X.to(device), Y.to(device)
