@@ -82,6 +82,10 @@ def args_parser(argv=None):
     parser.add_argument('--agents_per_stream', type=int, default=0,
                         help="train up to this many agents concurrently on separate "
                              "HIP streams per rank (0 = auto)")
+    parser.add_argument('--model', type=str, default=None,
+                        choices=[None, 'resnet18'],
+                        help="override the dataset->model mapping "
+                             "(build extension: ResNet18 w/ BatchNorm)")
     parser.add_argument('--no_tb', action='store_true', default=False,
                         help="disable the TensorBoard writer")
     parser.add_argument('--log_dir', type=str, default='logs',

