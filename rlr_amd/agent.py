@@ -89,11 +89,20 @@ class Agent:
 
         n = self._X.shape[0]
         bs = args.bs
+        use_graphs = (self.device.type == 'cuda'
+                      and getattr(args, 'hip_graphs', True)
+                      and not Fo.force_eager())
+        engine = gm.get_engine(args) if use_graphs else None
+        if engine is not None:
+            engine.begin_round(theta0)
         for ep in range(args.local_ep):
             perm = np_rng(args.seed, 'shuffle', self.id, rnd, ep).permutation(n)
             perm_t = torch.as_tensor(perm, device=self.device)
             for lo in range(0, n, bs):
                 sel = perm_t[lo:lo + bs]
+                if engine is not None:
+                    engine.step(self._X, self._Y, sel)
+                    continue
                 inputs, labels = self._X[sel], self._Y[sel]
                 gm.zero_grad()
                 outputs = gm(inputs)
@@ -104,6 +113,7 @@ class Agent:
                                            args.client_moment, 10.0)
                 if args.clip > 0:
                     flat_ops.pgd_project_(gm.flat_params, theta0, args.clip)
+                gm.model.rng.advance_step()
 
         update = flat_ops.delta64(gm.flat_params, theta0_64)
         # restore the global replica for the next agent on this rank

@@ -84,3 +84,10 @@ class FlatParamModel:
 
     def set_dropout_seed(self, seed):
         self.model.set_dropout_seed(seed)
+
+    def get_engine(self, args):
+        """Cached hipGraph TrainEngine (engine.py) for this model."""
+        if not hasattr(self, '_engine') or self._engine is None:
+            from .engine import TrainEngine
+            self._engine = TrainEngine(self, args)
+        return self._engine

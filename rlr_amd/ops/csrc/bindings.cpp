@@ -25,6 +25,8 @@ void launch_maxpool2x2_bwd(const float*, const uint8_t*, float*, int, int,
                            int, int, int, void*);
 void launch_dropout_fwd(const float*, float*, uint8_t*, long, float,
                         uint64_t, uint64_t, void*);
+void launch_dropout_fwd_dev(const float*, float*, uint8_t*, long, float,
+                            const unsigned long long*, int, void*);
 void launch_dropout_bwd(const float*, const uint8_t*, float*, long, float,
                         void*);
 void launch_gap_fwd(const float*, float*, int, int, void*);
@@ -155,6 +157,22 @@ std::tuple<torch::Tensor, torch::Tensor> dropout_fwd(torch::Tensor x,
   launch_dropout_fwd(x.data_ptr<float>(), y.data_ptr<float>(),
                      mask.data_ptr<uint8_t>(), x.numel(), (float)p,
                      (uint64_t)seed, (uint64_t)offset, stream_of(x));
+  return {y, mask};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> dropout_fwd_dev(torch::Tensor x,
+                                                         double p,
+                                                         torch::Tensor state,
+                                                         int64_t site) {
+  CHK_CUDA(x);
+  CHK(state.scalar_type() == torch::kUInt64 ||
+      state.scalar_type() == torch::kInt64);
+  auto y = torch::empty_like(x);
+  auto mask = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
+  launch_dropout_fwd_dev(
+      x.data_ptr<float>(), y.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+      x.numel(), (float)p,
+      (const unsigned long long*)state.data_ptr(), (int)site, stream_of(x));
   return {y, mask};
 }
 
@@ -547,6 +565,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
   m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_fwd_dev", &dropout_fwd_dev);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("gap_fwd", &gap_fwd);
   m.def("gap_bwd", &gap_bwd);

@@ -164,6 +164,36 @@ __global__ void dropout_fwd_k(const float* __restrict__ x,
   }
 }
 
+// Device-state variant for hipGraph capture: seed/base-offset live in a
+// 2 x u64 device buffer bumped between replays, so masks advance across
+// graph replays without re-capture.
+__global__ void dropout_fwd_dev_k(const float* __restrict__ x,
+                                  float* __restrict__ y,
+                                  uint8_t* __restrict__ mask, long n,
+                                  float p,
+                                  const unsigned long long* __restrict__
+                                      state, int site) {
+  uint64_t seed = state[0];
+  uint64_t offset = state[1] + site;
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = (n + 3) / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    Philox4 r = philox4(seed, offset, (uint32_t)i);
+    uint32_t rv[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long k = i * 4 + j;
+      if (k < n) {
+        bool keep = u32_to_uniform(rv[j]) >= p;
+        mask[k] = keep;
+        y[k] = keep ? x[k] * scale : 0.f;
+      }
+    }
+  }
+}
+
 __global__ void dropout_bwd_k(const float* __restrict__ dy,
                               const uint8_t* __restrict__ mask,
                               float* __restrict__ dx, long n, float p) {
@@ -179,6 +209,12 @@ void launch_dropout_fwd(const float* x, float* y, uint8_t* mask, long n,
                         float p, uint64_t seed, uint64_t offset, void* s) {
   dropout_fwd_k<<<grid_for((n + 3) / 4), kBlock, 0, (hipStream_t)s>>>(
       x, y, mask, n, p, seed, offset);
+}
+void launch_dropout_fwd_dev(const float* x, float* y, uint8_t* mask, long n,
+                            float p, const unsigned long long* state,
+                            int site, void* s) {
+  dropout_fwd_dev_k<<<grid_for((n + 3) / 4), kBlock, 0, (hipStream_t)s>>>(
+      x, y, mask, n, p, state, site);
 }
 void launch_dropout_bwd(const float* dy, const uint8_t* mask, float* dx,
                         long n, float p, void* s) {

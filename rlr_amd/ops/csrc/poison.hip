@@ -63,7 +63,8 @@ __global__ void poison_subf_k(float* __restrict__ data,
        t += stride) {
     int px = t % HW;
     long b = idxs[t / HW];
-    data[b * HW + px] -= mask[px] * (1.0f / 255.0f);
+    data[b * HW + px] -= mask[px] / 255.0f;  // exact div: matches the CPU
+                                             // path bit-for-bit
   }
 }
 

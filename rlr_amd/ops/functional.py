@@ -28,20 +28,54 @@ def _gpu(x: torch.Tensor) -> bool:
 
 
 class DropoutCtx:
-    """Deterministic dropout stream: (seed64, call counter)."""
+    """Deterministic dropout stream.
+
+    CPU path: (seed64, monotonically increasing call counter).
+    GPU path: a 2 x u64 DEVICE buffer [seed, base_offset] + a per-forward
+    site index; the kernel draws philox(seed, base+site, idx).  Keeping the
+    state in device memory lets a hipGraph-captured training step advance
+    its masks between replays (engine.py bumps base_offset by SITE_STRIDE
+    after every step) without re-capture."""
+
+    SITE_STRIDE = 16  # max dropout call sites per forward
 
     def __init__(self):
         self.seed = 0
-        self.counter = 0
+        self.counter = 0        # CPU stream
+        self.site = 0           # GPU per-forward site index
+        self._state = None      # device u64[2]
 
     def reset(self, seed: int):
         self.seed = int(seed)
         self.counter = 0
+        self.site = 0
+        if self._state is not None:
+            self._state.copy_(torch.tensor([self.seed, 0],
+                                           dtype=torch.int64))
 
     def next_offset(self) -> int:
         c = self.counter
         self.counter += 1
         return c
+
+    # ---- GPU/graph path ----
+    def gpu_state(self, device) -> torch.Tensor:
+        if self._state is None or self._state.device != torch.device(device):
+            self._state = torch.tensor([self.seed, 0], dtype=torch.int64,
+                                       device=device)
+        return self._state
+
+    def next_site(self) -> int:
+        s = self.site
+        self.site += 1
+        assert self.site <= self.SITE_STRIDE
+        return s
+
+    def advance_step(self):
+        """Call once after each training step (outside any graph)."""
+        self.site = 0
+        if self._state is not None:
+            self._state[1] += self.SITE_STRIDE
 
 
 # ----------------------------------------------------------------- conv2d
@@ -151,8 +185,8 @@ def max_pool2d_2x2(x):
 
 class _Dropout(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, p, seed, offset):
-        y, mask = ext().dropout_fwd(x, p, seed, offset)
+    def forward(ctx, x, p, state, site):
+        y, mask = ext().dropout_fwd_dev(x, p, state, site)
         ctx.save_for_backward(mask)
         ctx.p = p
         return y
@@ -167,7 +201,7 @@ def dropout(x, p, training, rng: DropoutCtx):
     if not training or p == 0:
         return x
     if _gpu(x):
-        return _Dropout.apply(x, p, rng.seed, rng.next_offset())
+        return _Dropout.apply(x, p, rng.gpu_state(x.device), rng.next_site())
     from ..utils.rng import derive_seed
     g = torch.Generator(device='cpu')
     g.manual_seed(derive_seed(rng.seed, 'dropout', rng.next_offset()))

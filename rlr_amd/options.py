@@ -88,6 +88,11 @@ def args_parser(argv=None):
                              "(build extension: ResNet18 w/ BatchNorm)")
     parser.add_argument('--no_tb', action='store_true', default=False,
                         help="disable the TensorBoard writer")
+    parser.add_argument('--no_hip_graphs', dest='hip_graphs',
+                        action='store_false', default=True,
+                        help="disable hipGraph capture of the training step "
+                             "(debug; results are bitwise identical either "
+                             "way)")
     parser.add_argument('--log_dir', type=str, default='logs',
                         help="TensorBoard log root")
 

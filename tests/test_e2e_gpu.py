@@ -106,9 +106,9 @@ def test_defense_semantics_gpu():
     poisons = {}
     for thr in (0, 5):
         args = _args(num_agents=10, num_corrupt=2, poison_frac=1.0,
-                     rounds=10, snap=5, local_ep=2, bs=128,
+                     rounds=15, snap=3, local_ep=2, bs=128,
                      pattern_type='square', robustLR_threshold=thr)
         h = run(args)
         poisons[thr] = h['poison_acc']
-    assert max(poisons[0][-1:]) > 0.4, poisons
-    assert poisons[5][-1] < 0.2, poisons
+    assert max(poisons[0][-3:]) > 0.4, poisons
+    assert max(poisons[5][-2:]) < 0.2, poisons
