@@ -51,6 +51,10 @@ class TrainEngine:
 
     def _build(self, bs, x_shape, device):
         gm = self.gm
+        # graph capture must not race other streams' submissions (several
+        # replicas train concurrently on their own streams): drain the
+        # device first; capture itself is thread_local-error-mode
+        torch.cuda.synchronize(device)
         static_x = torch.zeros((bs,) + tuple(x_shape), device=device)
         static_y = torch.zeros(bs, dtype=torch.long, device=device)
 
@@ -58,6 +62,7 @@ class TrainEngine:
         # and restore so graph building is invisible to training
         snap_p = gm.flat_params.clone()
         snap_m = gm.momentum.clone()
+        snap_b = gm.flat_buffers.clone() if gm.n_buffers else None
         rng = gm.model.rng
         snap_state = rng.gpu_state(device).clone()
         snap_site = rng.site
@@ -71,17 +76,17 @@ class TrainEngine:
         torch.cuda.current_stream(device).wait_stream(side)
 
         g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, pool=self.pool,
+                              capture_error_mode="thread_local"):
+            self._step_body(static_x, static_y)
         if self.pool is None:
-            with torch.cuda.graph(g):
-                self._step_body(static_x, static_y)
             self.pool = g.pool()
-        else:
-            with torch.cuda.graph(g, pool=self.pool):
-                self._step_body(static_x, static_y)
         rng.site = snap_site
 
         gm.flat_params.copy_(snap_p)
         gm.momentum.copy_(snap_m)
+        if snap_b is not None:
+            gm.flat_buffers.copy_(snap_b)
         rng.gpu_state(device).copy_(snap_state)
         torch.cuda.synchronize(device)
 

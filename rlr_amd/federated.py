@@ -87,6 +87,22 @@ def build_world(args):
                 X_val=X_val, Y_val=Y_val, X_pv=X_pv, Y_pv=Y_pv)
 
 
+def _replica_pool(world, args, gm):
+    """Model replicas on separate HIP streams: up to R sampled agents train
+    CONCURRENTLY per GPU (the reference trains agents strictly sequentially,
+    federated.py:68-72; the tiny CNN's kernels underfill 256 CUs, so
+    stream-level concurrency is the chip-filling lever)."""
+    if 'replicas' not in world:
+        R = args.agents_per_stream if args.agents_per_stream > 0 else 4
+        reps = []
+        for _ in range(max(1, R)):
+            m = get_model(args.data, getattr(args, 'model', None))
+            rep = FlatParamModel(m, gm.device)
+            reps.append((rep, torch.cuda.Stream(device=gm.device)))
+        world['replicas'] = reps
+    return world['replicas']
+
+
 def run_round(args, world, rnd, rank, world_size):
     """One FL round; returns the stacked update matrix's agent id list."""
     gm = world['global_model']
@@ -100,11 +116,33 @@ def run_round(args, world, rnd, rank, world_size):
                         device=gm.device)
     local_buf = (torch.zeros(chunk, gm.n_buffers, device=gm.device)
                  if gm.n_buffers else None)
-    for slot, agent_id in enumerate(mine):
-        update = agents[agent_id].local_train(gm, rnd=rnd)
-        local[slot].copy_(update)
-        if local_buf is not None:
-            local_buf[slot].copy_(agents[agent_id].buffer_delta)
+
+    use_streams = (gm.device.type == 'cuda' and len(mine) > 1
+                   and args.agents_per_stream != 1)
+    if use_streams:
+        reps = _replica_pool(world, args, gm)
+        R = len(reps)
+        main = torch.cuda.current_stream(gm.device)
+        for _, st in reps:
+            st.wait_stream(main)
+        for slot, agent_id in enumerate(mine):
+            rep, st = reps[slot % R]
+            with torch.cuda.stream(st):
+                rep.flat_params.copy_(gm.flat_params)
+                if rep.n_buffers:
+                    rep.flat_buffers.copy_(gm.flat_buffers)
+                update = agents[agent_id].local_train(rep, rnd=rnd)
+                local[slot].copy_(update)
+                if local_buf is not None:
+                    local_buf[slot].copy_(agents[agent_id].buffer_delta)
+        for _, st in reps:
+            main.wait_stream(st)
+    else:
+        for slot, agent_id in enumerate(mine):
+            update = agents[agent_id].local_train(gm, rnd=rnd)
+            local[slot].copy_(update)
+            if local_buf is not None:
+                local_buf[slot].copy_(agents[agent_id].buffer_delta)
 
     n_valid = []
     for r in range(world_size):
