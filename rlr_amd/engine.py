@@ -76,9 +76,17 @@ class TrainEngine:
         torch.cuda.current_stream(device).wait_stream(side)
 
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g, pool=self.pool,
-                              capture_error_mode="thread_local"):
-            self._step_body(static_x, static_y)
+        # a GC cycle during capture frees pre-capture tensors -> hipFree
+        # inside an active capture -> abort; collect first, then hold GC off
+        import gc
+        gc.collect()
+        gc.disable()
+        try:
+            with torch.cuda.graph(g, pool=self.pool,
+                                  capture_error_mode="thread_local"):
+                self._step_body(static_x, static_y)
+        finally:
+            gc.enable()
         if self.pool is None:
             self.pool = g.pool()
         rng.site = snap_site
