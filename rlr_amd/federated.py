@@ -117,6 +117,7 @@ def run_round(args, world, rnd, rank, world_size):
     local_buf = (torch.zeros(chunk, gm.n_buffers, device=gm.device)
                  if gm.n_buffers else None)
 
+    from .utils.tracing import trace_range
     use_streams = (gm.device.type == 'cuda' and len(mine) > 1
                    and args.agents_per_stream != 1)
     if use_streams:
@@ -148,9 +149,11 @@ def run_round(args, world, rnd, rank, world_size):
     for r in range(world_size):
         rlo, rhi, _ = pdist.chunk_bounds(len(sampled), world_size, r)
         n_valid.append(rhi - rlo)
-    stacked = pdist.all_gather_updates(local, n_valid, chunk)
-    world['aggregator'].aggregate_updates(gm, stacked, rnd,
-                                          agent_ids=sampled)
+    with trace_range('gather_updates'):
+        stacked = pdist.all_gather_updates(local, n_valid, chunk)
+    with trace_range('aggregate'):
+        world['aggregator'].aggregate_updates(gm, stacked, rnd,
+                                              agent_ids=sampled)
     if local_buf is not None:
         buf_stacked = pdist.all_gather_updates(local_buf, n_valid, chunk)
         world['aggregator'].aggregate_buffers(gm, buf_stacked, sampled)
@@ -219,6 +222,7 @@ def run(args, writer=None, progress=False):
                                       cum_poison_acc_mean / rnd, rnd)
                     writer.add_scalar('Perf/Rounds_Per_Sec', rps, rnd)
                 print(f'| Val_Loss/Val_Acc: {val_loss:.3f} / {val_acc:.3f} |')
+                print(f'| Val_Per_Class_Acc: {val_pc} ')
                 print(f'| Poison Loss/Poison Acc: '
                       f'{poison_loss:.3f} / {poison_acc:.3f} |')
                 if args.ckpt_dir:
