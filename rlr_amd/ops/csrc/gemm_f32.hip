@@ -195,15 +195,20 @@ __global__ void transpose_f32_k(const float* __restrict__ in,
   }
 }
 
-// column sum: db[n] = sum_m dY[m][n] (bias gradient)
+// column sum: db[n] = sum_m dY[m][n] (bias gradient).
+// One wave per column; lane l accumulates rows l, l+64, ... then a shuffle
+// tree (deterministic).  Consecutive waves in a block handle consecutive
+// columns, so each lane-step reads a coalesced-ish (64*N-strided) front.
 __global__ void colsum_k(const float* __restrict__ dY,
                          float* __restrict__ db, int M, int N) {
-  for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
-       n += gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int m = 0; m < M; ++m) acc += dY[(long)m * N + n];
-    db[n] = acc;
-  }
+  int n = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
+  if (n >= N) return;
+  float acc = 0.f;
+  for (int m = lane; m < M; m += kWave) acc += dY[(long)m * N + n];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) db[n] = acc;
 }
 
 extern "C" {
@@ -254,7 +259,7 @@ void launch_transpose_f32(const float* in, float* out, int M, int N,
 }
 
 void launch_colsum(const float* dY, float* db, int M, int N, void* s) {
-  colsum_k<<<grid_for(N, kBlock, 64), kBlock, 0, (hipStream_t)s>>>(dY, db, M,
-                                                                   N);
+  int wpb = kBlock / kWave;
+  colsum_k<<<(N + wpb - 1) / wpb, kBlock, 0, (hipStream_t)s>>>(dY, db, M, N);
 }
 }
