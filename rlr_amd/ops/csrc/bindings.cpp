@@ -71,7 +71,7 @@ int conv_bwd_weight_splitk(int, int, long);
 void launch_conv_bwd_weight(const float*, const float*, float*, float*, int,
                             int, int, int, int, int, int, int, int, int, int,
                             int, void*);
-void launch_conv_db(const float*, float*, int, int, int, void*);
+void launch_conv_db(const float*, float*, float*, int, int, int, void*);
 void launch_wperm_crs_ko(const float*, float*, int, int, int, void*);
 void launch_wperm_kors_c(const float*, float*, int, int, int, void*);
 // batchnorm.hip
@@ -456,8 +456,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   torch::Tensor db;
   if (has_b) {
     db = torch::empty({Kout}, dy.options());
-    launch_conv_db(dy.data_ptr<float>(), db.data_ptr<float>(), Nb, Kout,
-                   OH * OW, st);
+    auto parts = torch::empty({(long)Kout * 64}, dy.options());
+    launch_conv_db(dy.data_ptr<float>(), db.data_ptr<float>(),
+                   parts.data_ptr<float>(), Nb, Kout, OH * OW, st);
   } else {
     db = torch::Tensor();
   }

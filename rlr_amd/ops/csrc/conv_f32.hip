@@ -71,7 +71,7 @@ __device__ __forceinline__ void mfma_tile(const float* __restrict__ Abuf,
 
 // ------------------------------------------------------------------- fwd
 
-template <int RT>
+template <int RT, bool P0>
 __global__ __launch_bounds__(256)
 void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
                 const float* __restrict__ bias, float* __restrict__ y,
@@ -123,7 +123,7 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
             int c, r, s;
             crs_decomp<RT>(k, sh, c, r, s);
             int ih = oh0 + r, iw = ow0 + s;
-            if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
+            if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
               v[e] = xp[((long)c * sh.H + ih) * sh.W + iw];
           }
         }
@@ -304,7 +304,7 @@ void conv_bwd_data_k(const float* __restrict__ dy,
 // 64x64 tile (MI=NI=2): Kout rarely exceeds 64 per tile row and a 128-row
 // tile would idle half its MFMAs.  gridDim.z = split-K chunks over
 // m = (nb,oh,ow); partial slabs [z][Kout][C*R*S].
-template <int RT>
+template <int RT, bool P0>
 __global__ __launch_bounds__(256)
 void conv_bwd_weight_k(const float* __restrict__ dy,
                        const float* __restrict__ x, float* __restrict__ out,
@@ -392,7 +392,7 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
         for (int e = 0; e < 4; ++e) {
           if (bvalid[e]) {
             int ih = ih0 + br[e], iw = iw0 + bs[e];
-            if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
+            if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
               v[e] = xp[((long)bc[e] * sh.H + ih) * sh.W + iw];
           }
         }
@@ -431,14 +431,22 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
 }
 
 // ---- conv bias gradient: db[k] = sum over (nb,oh,ow) of dy ----
-// one block per k; coalesced strided loop; deterministic tree reduce.
-__global__ void conv_db_k(const float* __restrict__ dy,
-                          float* __restrict__ db, int Nb, int Kout,
-                          int OHW) {
+// Two-stage deterministic reduce: stage 1 fills partials[k][chunk] from a
+// (Kout x NCHUNK) grid (plenty of blocks for 256 CUs); stage 2 is one wave
+// per k over the NCHUNK partials.
+constexpr int kDbChunks = 64;
+
+__global__ void conv_db_stage1_k(const float* __restrict__ dy,
+                                 float* __restrict__ partials, int Nb,
+                                 int Kout, int OHW) {
   int k = blockIdx.x;
+  int chunk = blockIdx.y;
+  long total = (long)Nb * OHW;
+  long per = (total + kDbChunks - 1) / kDbChunks;
+  long lo = chunk * per, hi = min(total, lo + per);
   __shared__ float sh[kBlock];
   float acc = 0.f;
-  for (long i = threadIdx.x; i < (long)Nb * OHW; i += blockDim.x) {
+  for (long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
     long nb = i / OHW, px = i % OHW;
     acc += dy[(nb * Kout + k) * (long)OHW + px];
   }
@@ -448,7 +456,18 @@ __global__ void conv_db_k(const float* __restrict__ dy,
     if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
     __syncthreads();
   }
-  if (threadIdx.x == 0) db[k] = sh[0];
+  if (threadIdx.x == 0) partials[(long)k * kDbChunks + chunk] = sh[0];
+}
+
+__global__ void conv_db_stage2_k(const float* __restrict__ partials,
+                                 float* __restrict__ db, int Kout) {
+  int k = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
+  if (k >= Kout) return;
+  float acc = (lane < kDbChunks) ? partials[(long)k * kDbChunks + lane] : 0.f;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) db[k] = acc;
 }
 
 // permute w (Kout,C,R,S) -> dst layouts
@@ -493,12 +512,14 @@ void launch_conv_fwd(const float* x, const float* wt, const float* bias,
   long M = (long)Nb * OH * OW;
   dim3 grid((M + 127) / 128, (Kout + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
-  if (R == 3 && S == 3)
-    conv_fwd_k<3><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
-  else if (R == 1 && S == 1)
-    conv_fwd_k<1><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+  if (R == 3 && S == 3 && pad == 0)
+    conv_fwd_k<3, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+  else if (R == 3 && S == 3)
+    conv_fwd_k<3, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+  else if (R == 1 && S == 1 && pad == 0)
+    conv_fwd_k<1, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
   else
-    conv_fwd_k<0><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+    conv_fwd_k<0, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
 }
 
 void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
@@ -543,22 +564,30 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
   dim3 grid((Kout + 63) / 64, (Ncrs + BN - 1) / BN, SK);
   hipStream_t st = (hipStream_t)s;
   float* out = SK == 1 ? dw : ws;
-  if (R == 3 && S == 3)
-    conv_bwd_weight_k<3><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                               k_per_chunk, SK == 1);
-  else if (R == 1 && S == 1)
-    conv_bwd_weight_k<1><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                               k_per_chunk, SK == 1);
+  if (R == 3 && S == 3 && pad == 0)
+    conv_bwd_weight_k<3, true><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
+                                                     k_per_chunk, SK == 1);
+  else if (R == 3 && S == 3)
+    conv_bwd_weight_k<3, false><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
+                                                      k_per_chunk, SK == 1);
+  else if (R == 1 && S == 1 && pad == 0)
+    conv_bwd_weight_k<1, true><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
+                                                     k_per_chunk, SK == 1);
   else
-    conv_bwd_weight_k<0><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                               k_per_chunk, SK == 1);
+    conv_bwd_weight_k<0, false><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
+                                                      k_per_chunk, SK == 1);
   if (SK > 1)
     launch_splitk_reduce(ws, dw, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
 }
 
-void launch_conv_db(const float* dy, float* db, int Nb, int Kout, int OHW,
-                    void* s) {
-  conv_db_k<<<Kout, kBlock, 0, (hipStream_t)s>>>(dy, db, Nb, Kout, OHW);
+void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
+                    int Kout, int OHW, void* s) {
+  hipStream_t st = (hipStream_t)s;
+  conv_db_stage1_k<<<dim3(Kout, kDbChunks), kBlock, 0, st>>>(dy, partials,
+                                                             Nb, Kout, OHW);
+  int wpb = kBlock / kWave;
+  conv_db_stage2_k<<<(Kout + wpb - 1) / wpb, kBlock, 0, st>>>(partials, db,
+                                                              Kout);
 }
 
 void launch_wperm_crs_ko(const float* w, float* out, int Kout, int C, int RS,
