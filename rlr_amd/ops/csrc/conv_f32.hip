@@ -108,10 +108,15 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
     nbs[j] = gmc / ((long)sh.OW * sh.OH);
   }
 
-  auto stage = [&](int buf, int k0) {
+  // async-STAGE split (guide T14/G15): loads are issued to REGISTERS a
+  // full K-tile early and the LDS write happens after the barrier, so HBM
+  // latency hides under the previous tile's MFMAs.
+  float ra[BM / 32][4];
+  float4 rb[2];
+  auto stage_load = [&](int k0) {
 #pragma unroll
     for (int j = 0; j < BM / 32; ++j) {
-      float v[4] = {0.f, 0.f, 0.f, 0.f};
+      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
       if (mval[j]) {
         int oh0 = ohs[j] * sh.stride - sh.pad;
         int ow0 = ows[j] * sh.stride - sh.pad;
@@ -124,18 +129,14 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
             crs_decomp<RT>(k, sh, c, r, s);
             int ih = oh0 + r, iw = ow0 + s;
             if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
-              v[e] = xp[((long)c * sh.H + ih) * sh.W + iw];
+              ra[j][e] = xp[((long)c * sh.H + ih) * sh.W + iw];
           }
         }
       }
-      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
-      ((float2*)dst)[0] = {v[0], v[1]};
-      ((float2*)dst)[1] = {v[2], v[3]};
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int kk = bk + j * 16;
-      int gk = k0 + kk;
+      int gk = k0 + bk + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
         const float* src = wt + (long)gk * sh.Kout + n_blk + bn;
@@ -148,15 +149,31 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
           if (n_blk + bn + 3 < sh.Kout) q.w = src[3];
         }
       }
-      *(float4*)&B_lds[buf][kk * LDB_S + bn] = q;
+      rb[j] = q;
     }
   };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < BM / 32; ++j) {
+      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+      ((float2*)dst)[0] = {ra[j][0], ra[j][1]};
+      ((float2*)dst)[1] = {ra[j][2], ra[j][3]};
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+  };
 
-  stage(0, 0);
+  stage_load(0);
+  stage_write(0);
+  if (BK < Kdim) stage_load(BK);
   __syncthreads();
   int buf = 0;
   for (int k0 = 0; k0 < Kdim; k0 += BK) {
-    if (k0 + BK < Kdim) stage(buf ^ 1, k0 + BK);
+    if (k0 + BK < Kdim) {
+      stage_write(buf ^ 1);
+      if (k0 + 2 * BK < Kdim) stage_load(k0 + 2 * BK);
+    }
     mfma_tile<MI, NI>(A_lds[buf], B_lds[buf], wr, wc, l15, l4, acc);
     __syncthreads();
     buf ^= 1;
@@ -222,10 +239,12 @@ void conv_bwd_data_k(const float* __restrict__ dy,
     nbs[j] = gmc / ((long)sh.W * sh.H);
   }
 
-  auto stage = [&](int buf, int k0) {
+  float ra[BM / 32][4];
+  float4 rb[2];
+  auto stage_load = [&](int k0) {
 #pragma unroll
     for (int j = 0; j < BM / 32; ++j) {
-      float v[4] = {0.f, 0.f, 0.f, 0.f};
+      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
       if (mval[j]) {
         int ihp = ihs[j] + sh.pad, iwp = iws[j] + sh.pad;
         const float* dyp =
@@ -241,19 +260,15 @@ void conv_bwd_data_k(const float* __restrict__ dy,
                 own % stride == 0) {
               int oh = ohn / stride, ow = own / stride;
               if (oh < sh.OH && ow < sh.OW)
-                v[e] = dyp[((long)ko * sh.OH + oh) * sh.OW + ow];
+                ra[j][e] = dyp[((long)ko * sh.OH + oh) * sh.OW + ow];
             }
           }
         }
       }
-      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
-      ((float2*)dst)[0] = {v[0], v[1]};
-      ((float2*)dst)[1] = {v[2], v[3]};
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int kk = bk + j * 16;
-      int gk = k0 + kk;
+      int gk = k0 + bk + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
         const float* src = wp + (long)gk * sh.C + n_blk + bn;
@@ -266,15 +281,31 @@ void conv_bwd_data_k(const float* __restrict__ dy,
           if (n_blk + bn + 3 < sh.C) q.w = src[3];
         }
       }
-      *(float4*)&B_lds[buf][kk * LDB_S + bn] = q;
+      rb[j] = q;
     }
   };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < BM / 32; ++j) {
+      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+      ((float2*)dst)[0] = {ra[j][0], ra[j][1]};
+      ((float2*)dst)[1] = {ra[j][2], ra[j][3]};
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+  };
 
-  stage(0, 0);
+  stage_load(0);
+  stage_write(0);
+  if (BK < Kdim) stage_load(BK);
   __syncthreads();
   int buf = 0;
   for (int k0 = 0; k0 < Kdim; k0 += BK) {
-    if (k0 + BK < Kdim) stage(buf ^ 1, k0 + BK);
+    if (k0 + BK < Kdim) {
+      stage_write(buf ^ 1);
+      if (k0 + 2 * BK < Kdim) stage_load(k0 + 2 * BK);
+    }
     mfma_tile<MI, NI>(A_lds[buf], B_lds[buf], wr, wc, l15, l4, acc);
     __syncthreads();
     buf ^= 1;
@@ -342,13 +373,15 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     crs_decomp<RT>(bvalid[e] ? crs : 0, sh, bc[e], br[e], bs[e]);
   }
 
-  auto stage = [&](int buf, long k0) {
+  float ra[2][4];
+  float4 rb[2];
+  auto stage_load = [&](long k0) {
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       int ko = m_blk + am + j * 32;
-      float v[4] = {0.f, 0.f, 0.f, 0.f};
+      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
       if (ko < sh.Kout) {
-        // 4 consecutive m share (nb, oh) almost always; decompose per elem
+        // 4 consecutive m share (nb, oh) almost always; slow path on wrap
         long k = k0 + ak;
         int ow = k % sh.OW;
         int oh = (k / sh.OW) % sh.OH;
@@ -360,26 +393,22 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
           if (k + e < k_hi) {
             int owe = ow + e;
             if (owe < sh.OW)
-              v[e] = dyp[owe];
+              ra[j][e] = dyp[owe];
             else {
               long ke = k + e;
               int ow2 = ke % sh.OW;
               int oh2 = (ke / sh.OW) % sh.OH;
               int nb2 = ke / ((long)sh.OW * sh.OH);
-              v[e] = dy[(((long)nb2 * sh.Kout + ko) * sh.OH + oh2) * sh.OW +
-                        ow2];
+              ra[j][e] = dy[(((long)nb2 * sh.Kout + ko) * sh.OH + oh2) *
+                                sh.OW + ow2];
             }
           }
         }
       }
-      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
-      ((float2*)dst)[0] = {v[0], v[1]};
-      ((float2*)dst)[1] = {v[2], v[3]};
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int kk = bk + j * 16;
-      long k = k0 + kk;
+      long k = k0 + bk + j * 16;
       float v[4] = {0.f, 0.f, 0.f, 0.f};
       if (k < k_hi) {
         int ow = k % sh.OW;
@@ -397,15 +426,31 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
           }
         }
       }
-      *(float4*)&B_lds[buf][kk * LDB_S + bn] = {v[0], v[1], v[2], v[3]};
+      rb[j] = {v[0], v[1], v[2], v[3]};
     }
   };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+      ((float2*)dst)[0] = {ra[j][0], ra[j][1]};
+      ((float2*)dst)[1] = {ra[j][2], ra[j][3]};
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+  };
 
-  stage(0, k_lo);
+  stage_load(k_lo);
+  stage_write(0);
+  if (k_lo + BK < k_hi) stage_load(k_lo + BK);
   __syncthreads();
   int buf = 0;
   for (long k0 = k_lo; k0 < k_hi; k0 += BK) {
-    if (k0 + BK < k_hi) stage(buf ^ 1, k0 + BK);
+    if (k0 + BK < k_hi) {
+      stage_write(buf ^ 1);
+      if (k0 + 2 * BK < k_hi) stage_load(k0 + 2 * BK);
+    }
     mfma_tile<MI, NI>(A_lds[buf], B_lds[buf], wr, wc, l15, l4, acc);
     __syncthreads();
     buf ^= 1;

@@ -57,7 +57,10 @@ void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
   const int bk = t >> 4;            // 0..15 (+16 per round, 2 rounds)
   const int bn = (t & 15) * 4;      // 0,4,..,60
 
-  auto stage = [&](int buf, long k0) {
+  // async-STAGE split (guide T14/G15): global loads go to registers a
+  // K-tile early; the LDS write lands after the barrier, under the MFMAs.
+  float4 ra[4], rb[2];
+  auto stage_load = [&](long k0) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       int m = am + j * 32;
@@ -75,14 +78,11 @@ void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
           if (k0 + ak + 3 < k_hi) v3 = A[base + 3];
         }
       }
-      float* dst = &A_lds[buf][m * LDA_S + ak];
-      ((float2*)dst)[0] = {v0, v1};
-      ((float2*)dst)[1] = {v2, v3};
+      ra[j] = {v0, v1, v2, v3};
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int kk = bk + j * 16;
-      long gk = k0 + kk;
+      long gk = k0 + bk + j * 16;
       float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
       if (gk < k_hi) {
         long base = gk * (long)ldb + n_blk + bn;
@@ -96,16 +96,32 @@ void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
           if (n_blk + bn + 3 < N) v3 = B[base + 3];
         }
       }
-      *(float4*)&B_lds[buf][kk * LDB_S + bn] = {v0, v1, v2, v3};
+      rb[j] = {v0, v1, v2, v3};
     }
   };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+      ((float2*)dst)[0] = {ra[j].x, ra[j].y};
+      ((float2*)dst)[1] = {ra[j].z, ra[j].w};
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+  };
 
-  stage(0, k_lo);
+  stage_load(k_lo);
+  stage_write(0);
+  if (k_lo + BK < k_hi) stage_load(k_lo + BK);
   __syncthreads();
 
   int buf = 0;
   for (long k0 = k_lo; k0 < k_hi; k0 += BK) {
-    if (k0 + BK < k_hi) stage(buf ^ 1, k0 + BK);
+    if (k0 + BK < k_hi) {
+      stage_write(buf ^ 1);
+      if (k0 + 2 * BK < k_hi) stage_load(k0 + 2 * BK);
+    }
     const float* Abuf = A_lds[buf];
     const float* Bbuf = B_lds[buf];
 #pragma unroll
