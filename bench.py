@@ -31,6 +31,9 @@ def main():
     ap.add_argument('--warmup', type=int, default=5)
     ap.add_argument('--agents_per_gpu', type=int, default=10)
     ap.add_argument('--data', type=str, default='fmnist')
+    ap.add_argument('--model', type=str, default=None,
+                    help="e.g. resnet18 (BASELINE configs 3-4: CIFAR10 "
+                         "ResNet18); default: the dataset's reference CNN")
     a = ap.parse_args()
 
     from rlr_amd.federated import build_world, run_round
@@ -47,12 +50,13 @@ def main():
     device = f'cuda:{int(os.environ.get("LOCAL_RANK", 0))}' if use_cuda else 'cpu'
 
     num_agents = a.agents_per_gpu * n
-    D.DEFAULT_SIZES['fmnist'] = (6000 * num_agents, 10000)
+    samples_per_agent = 6000 if a.data == 'fmnist' else 1250
+    D.DEFAULT_SIZES[a.data] = (samples_per_agent * num_agents, 10000)
     args = default_args(
-        data=a.data, num_agents=num_agents, num_corrupt=1 * n,
-        poison_frac=0.5, robustLR_threshold=4, aggr='avg', local_ep=2,
-        bs=256, agent_frac=1.0, synthetic=True, no_tb=True, snap=10 ** 9,
-        device=device)
+        data=a.data, model=a.model, num_agents=num_agents,
+        num_corrupt=1 * n, poison_frac=0.5, robustLR_threshold=4,
+        aggr='avg', local_ep=2, bs=256, agent_frac=1.0, synthetic=True,
+        no_tb=True, snap=10 ** 9, device=device)
 
     world_state = build_world(args)
 
@@ -94,8 +98,10 @@ def main():
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": "CNN_MNIST",
-                "dataset": "fmnist (synthetic, 6000 samples/agent)",
+                "model": a.model or ("CNN_MNIST" if a.data != 'cifar10'
+                                     else "CNN_CIFAR"),
+                "dataset": f"{a.data} (synthetic, "
+                           f"{samples_per_agent} samples/agent)",
                 "num_agents": num_agents,
                 "agents_per_gpu": a.agents_per_gpu,
                 "num_corrupt": 1 * n,
