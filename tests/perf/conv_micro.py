@@ -2,11 +2,14 @@
 Usage: python tests/perf/conv_micro.py [fwd|bwd_data|bwd_weight|all] [iters]
 Runs the CNN_MNIST conv2 shape (the dominant kernel in the FL step)."""
 
+import os
 import sys
 
-import torch
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                '..', '..'))
+import torch  # noqa: E402
 
-from rlr_amd.ops import ext
+from rlr_amd.ops import ext  # noqa: E402
 
 
 def main():
