@@ -1,40 +1,45 @@
-// BatchNorm2d fwd/bwd (build extension for ResNet18 — the reference has no
-// BN anywhere, SURVEY.md §2b last row).  NCHW, per-channel statistics.
-// Deterministic: one block per channel, fixed-order tree reductions, no
-// atomics.
+// BatchNorm2d fwd/bwd — NHWC (channels_last), the build's ResNet18
+// extension (the reference has no BN, SURVEY.md §2b last row).
+// x is a [M = N*H*W][C] row-major matrix; per-channel statistics are
+// two-stage column sums (chunk partials -> combine), deterministic, no
+// atomics.  Elementwise passes use a stride-multiple-of-C grid so each
+// thread's channel is computed ONCE (launcher rounds the grid).
 #include "common.h"
 
-// ---- stage 1 (train fwd): per-channel sum and sum-of-squares ----
-__global__ void bn_stats_k(const float* __restrict__ x, float* __restrict__
-                           sums,  // [2][C]
-                           int Nb, int C, int HW) {
-  int c = blockIdx.x;
-  __shared__ float sh[2][kBlock];
+constexpr int kBnChunks = 64;
+
+// ---- stage 1: partials[chunk][2C] = (sum x, sum x^2) over a row chunk ----
+__global__ void bn_stats1_k(const float* __restrict__ x,
+                            float* __restrict__ partials, long M, int C) {
+  int chunk = blockIdx.x;
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  long per = (M + kBnChunks - 1) / kBnChunks;
+  long lo = (long)chunk * per, hi = min(M, lo + per);
   float s = 0.f, ss = 0.f;
-  for (long i = threadIdx.x; i < (long)Nb * HW; i += blockDim.x) {
-    long nb = i / HW, px = i % HW;
-    float v = x[(nb * C + c) * (long)HW + px];
+  for (long m = lo; m < hi; ++m) {
+    float v = x[m * C + c];
     s += v;
     ss += v * v;
   }
-  sh[0][threadIdx.x] = s;
-  sh[1][threadIdx.x] = ss;
-  __syncthreads();
-  for (int off = kBlock / 2; off > 0; off >>= 1) {
-    if (threadIdx.x < off) {
-      sh[0][threadIdx.x] += sh[0][threadIdx.x + off];
-      sh[1][threadIdx.x] += sh[1][threadIdx.x + off];
-    }
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    sums[c] = sh[0][0];
-    sums[C + c] = sh[1][0];
-  }
+  partials[((long)chunk * 2) * C + c] = s;
+  partials[((long)chunk * 2 + 1) * C + c] = ss;
 }
 
-// ---- stage 2: finalize mean/rstd, update running stats (torch semantics:
-// running_var uses the UNBIASED batch variance) ----
+__global__ void bn_stats2_k(const float* __restrict__ partials,
+                            float* __restrict__ sums, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, ss = 0.f;
+  for (int ch = 0; ch < kBnChunks; ++ch) {
+    s += partials[((long)ch * 2) * C + c];
+    ss += partials[((long)ch * 2 + 1) * C + c];
+  }
+  sums[c] = s;
+  sums[C + c] = ss;
+}
+
+// ---- finalize mean/rstd + running stats (torch: running_var unbiased) ----
 __global__ void bn_finalize_k(const float* __restrict__ sums,
                               float* __restrict__ save_mean,
                               float* __restrict__ save_rstd,
@@ -55,7 +60,6 @@ __global__ void bn_finalize_k(const float* __restrict__ sums,
   }
 }
 
-// eval path: mean/rstd from running stats
 __global__ void bn_eval_stats_k(const float* __restrict__ running_mean,
                                 const float* __restrict__ running_var,
                                 float* __restrict__ save_mean,
@@ -67,115 +71,145 @@ __global__ void bn_eval_stats_k(const float* __restrict__ running_mean,
   save_rstd[c] = rsqrtf(running_var[c] + eps);
 }
 
-// ---- stage 3: y = w * (x - mean) * rstd + b ----
+// ---- normalize: y = w[c]*(x-mean[c])*rstd[c] + b[c]; stride % C == 0 so
+// each thread's channel is fixed across its grid-stride loop ----
 __global__ void bn_norm_k(const float* __restrict__ x,
                           const float* __restrict__ w,
                           const float* __restrict__ b,
                           const float* __restrict__ mean,
                           const float* __restrict__ rstd,
-                          float* __restrict__ y, int Nb, int C, int HW) {
-  long n = (long)Nb * C * HW;
+                          float* __restrict__ y, long n, int C) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int c = (i / HW) % C;
-    y[i] = w[c] * (x[i] - mean[c]) * rstd[c] + b[c];
+  if (stride % C == 0) {  // launcher arranges this for power-of-two C
+    int c = (int)(i0 % C);
+    float wc = w[c] * rstd[c];
+    float bc = b[c] - mean[c] * wc;
+    for (long i = i0; i < n; i += stride) y[i] = x[i] * wc + bc;
+  } else {
+    for (long i = i0; i < n; i += stride) {
+      int c = (int)(i % C);
+      y[i] = w[c] * rstd[c] * (x[i] - mean[c]) + b[c];
+    }
   }
 }
 
 // ---- bwd stage 1: per-channel sum(dy), sum(dy * xhat) ----
-__global__ void bn_bwd_stats_k(const float* __restrict__ x,
-                               const float* __restrict__ dy,
-                               const float* __restrict__ mean,
-                               const float* __restrict__ rstd,
-                               float* __restrict__ out,  // [2][C]: db, dwdot
-                               int Nb, int C, int HW) {
-  int c = blockIdx.x;
-  __shared__ float sh[2][kBlock];
-  float s_dy = 0.f, s_dyx = 0.f;
-  float m = mean[c], rs = rstd[c];
-  for (long i = threadIdx.x; i < (long)Nb * HW; i += blockDim.x) {
-    long nb = i / HW, px = i % HW;
-    long idx = (nb * C + c) * (long)HW + px;
-    float g = dy[idx];
-    s_dy += g;
-    s_dyx += g * (x[idx] - m) * rs;
+__global__ void bn_bwd_stats1_k(const float* __restrict__ x,
+                                const float* __restrict__ dy,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ rstd,
+                                float* __restrict__ partials, long M,
+                                int C) {
+  int chunk = blockIdx.x;
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  long per = (M + kBnChunks - 1) / kBnChunks;
+  long lo = (long)chunk * per, hi = min(M, lo + per);
+  float m_ = mean[c], rs = rstd[c];
+  float sdy = 0.f, sdyx = 0.f;
+  for (long m = lo; m < hi; ++m) {
+    float g = dy[m * C + c];
+    sdy += g;
+    sdyx += g * (x[m * C + c] - m_) * rs;
   }
-  sh[0][threadIdx.x] = s_dy;
-  sh[1][threadIdx.x] = s_dyx;
-  __syncthreads();
-  for (int off = kBlock / 2; off > 0; off >>= 1) {
-    if (threadIdx.x < off) {
-      sh[0][threadIdx.x] += sh[0][threadIdx.x + off];
-      sh[1][threadIdx.x] += sh[1][threadIdx.x + off];
-    }
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) {
-    out[c] = sh[0][0];       // db
-    out[C + c] = sh[1][0];   // sum(dy * xhat) = dw
-  }
+  partials[((long)chunk * 2) * C + c] = sdy;
+  partials[((long)chunk * 2 + 1) * C + c] = sdyx;
 }
 
-// ---- bwd stage 2: dx = (w*rstd/Nc)*(Nc*dy - db - xhat*dwdot) ----
+// (bn_stats2_k combines these too: same [chunk][2C] layout)
+
+// ---- bwd: dx = w*rstd*(dy - (db + xhat*dwdot)/Nc) ----
 __global__ void bn_bwd_dx_k(const float* __restrict__ x,
                             const float* __restrict__ dy,
                             const float* __restrict__ w,
                             const float* __restrict__ mean,
                             const float* __restrict__ rstd,
-                            const float* __restrict__ stats, float*
-                            __restrict__ dx, int Nb, int C, int HW,
-                            int training) {
-  long n = (long)Nb * C * HW;
+                            const float* __restrict__ stats,
+                            float* __restrict__ dx, long n, int C,
+                            float inv_count) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  float inv = 1.f / ((float)Nb * HW);
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int c = (i / HW) % C;
-    float g = dy[i];
-    if (training) {
+  if (stride % C == 0) {
+    int c = (int)(i0 % C);
+    float wr = w[c] * rstd[c];
+    float m_ = mean[c], rs = rstd[c];
+    float db = stats[c], dwdot = stats[C + c];
+    for (long i = i0; i < n; i += stride) {
+      float xhat = (x[i] - m_) * rs;
+      dx[i] = wr * (dy[i] - inv_count * (db + xhat * dwdot));
+    }
+  } else {
+    for (long i = i0; i < n; i += stride) {
+      int c = (int)(i % C);
       float xhat = (x[i] - mean[c]) * rstd[c];
       dx[i] = w[c] * rstd[c] *
-              (g - inv * (stats[c] + xhat * stats[C + c]));
-    } else {
-      dx[i] = w[c] * rstd[c] * g;
+              (dy[i] - inv_count * (stats[c] + xhat * stats[C + c]));
     }
   }
+}
+
+static int bn_grid(long n, int C) {
+  // blocks such that gridDim*256 % C == 0 (C is a power of two <= 512 in
+  // practice; the rounding below also handles other C by rounding up to a
+  // multiple of C)
+  long want = (n + kBlock - 1) / kBlock;
+  if (want > kMaxBlocks) want = kMaxBlocks;
+  long mult = (C + kBlock - 1) / kBlock;  // blocks per C-span
+  long lcm = mult > 1 ? mult : 1;
+  if (256 % C != 0 && C % 256 == 0) lcm = C / 256;
+  if (lcm > 1) want = ((want + lcm - 1) / lcm) * lcm;
+  return (int)(want > 0 ? want : 1);
 }
 
 extern "C" {
 void launch_bn_fwd(const float* x, const float* w, const float* b,
                    float* running_mean, float* running_var, float* save_mean,
-                   float* save_rstd, float* y, float* scratch2C, int Nb,
+                   float* save_rstd, float* y, float* scratch, int Nb,
                    int C, int HW, float momentum, float eps, int training,
                    void* s) {
   hipStream_t st = (hipStream_t)s;
+  long M = (long)Nb * HW;
+  // scratch: kBnChunks*2*C partials + 2*C sums
+  float* partials = scratch;
+  float* sums = scratch + (long)kBnChunks * 2 * C;
   if (training) {
-    bn_stats_k<<<C, kBlock, 0, st>>>(x, scratch2C, Nb, C, HW);
+    dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
+    bn_stats1_k<<<g1, kBlock, 0, st>>>(x, partials, M, C);
+    bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials,
+                                                              sums, C);
     bn_finalize_k<<<(C + 255) / 256, 256, 0, st>>>(
-        scratch2C, save_mean, save_rstd, running_mean, running_var, C,
-        (long)Nb * HW, momentum, eps);
+        sums, save_mean, save_rstd, running_mean, running_var, C, M,
+        momentum, eps);
   } else {
     bn_eval_stats_k<<<(C + 255) / 256, 256, 0, st>>>(
         running_mean, running_var, save_mean, save_rstd, C, eps);
   }
-  bn_norm_k<<<grid_for((long)Nb * C * HW), kBlock, 0, st>>>(
-      x, w, b, save_mean, save_rstd, y, Nb, C, HW);
+  long n = M * C;
+  bn_norm_k<<<bn_grid(n, C), kBlock, 0, st>>>(x, w, b, save_mean, save_rstd,
+                                              y, n, C);
 }
 
 void launch_bn_bwd(const float* x, const float* dy, const float* w,
                    const float* save_mean, const float* save_rstd,
-                   float* stats2C, float* dx, float* dw, float* db, int Nb,
+                   float* scratch, float* dx, float* dw, float* db, int Nb,
                    int C, int HW, int training, void* s) {
   hipStream_t st = (hipStream_t)s;
-  bn_bwd_stats_k<<<C, kBlock, 0, st>>>(x, dy, save_mean, save_rstd, stats2C,
-                                       Nb, C, HW);
-  // db = stats[0:C], dw = stats[C:2C] — copied out by the binding
-  bn_bwd_dx_k<<<grid_for((long)Nb * C * HW), kBlock, 0, st>>>(
-      x, dy, w, save_mean, save_rstd, stats2C, dx, Nb, C, HW, training);
-  HIP_CHECK(hipMemcpyAsync(db, stats2C, C * sizeof(float),
+  long M = (long)Nb * HW;
+  float* partials = scratch;
+  float* stats = scratch + (long)kBnChunks * 2 * C;
+  dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
+  bn_bwd_stats1_k<<<g1, kBlock, 0, st>>>(x, dy, save_mean, save_rstd,
+                                         partials, M, C);
+  bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials, stats,
+                                                            C);
+  long n = M * C;
+  bn_bwd_dx_k<<<bn_grid(n, C), kBlock, 0, st>>>(
+      x, dy, w, save_mean, save_rstd, stats, dx, n, C,
+      training ? 1.f / (float)M : 0.f);
+  HIP_CHECK(hipMemcpyAsync(db, stats, C * sizeof(float),
                            hipMemcpyDeviceToDevice, st));
-  HIP_CHECK(hipMemcpyAsync(dw, stats2C + C, C * sizeof(float),
+  HIP_CHECK(hipMemcpyAsync(dw, stats + C, C * sizeof(float),
                            hipMemcpyDeviceToDevice, st));
 }
 }

@@ -7,11 +7,33 @@
 #include <cstdint>
 
 #define CHK(x) TORCH_CHECK(x, #x)
-#define CHK_CUDA(t) \
-  TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be cuda+contig")
+#define CHK_CUDA(t)                                                      \
+  TORCH_CHECK((t).is_cuda() &&                                           \
+                  ((t).is_contiguous() ||                                \
+                   (t).is_contiguous(torch::MemoryFormat::ChannelsLast)),\
+              #t " must be cuda + dense")
 
 static void* stream_of(const torch::Tensor& t) {
   return (void*)c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+// 4-D activations are channels_last (NHWC storage) on the HIP path.
+static torch::Tensor cl(torch::Tensor t) {
+  return t.contiguous(torch::MemoryFormat::ChannelsLast);
+}
+static torch::Tensor empty_cl(std::vector<int64_t> sizes,
+                              const torch::TensorOptions& opts) {
+  return torch::empty(sizes,
+                      opts.memory_format(torch::MemoryFormat::ChannelsLast));
+}
+// make t's storage order match ref's (flat elementwise kernels require it)
+static torch::Tensor match_layout(const torch::Tensor& ref,
+                                  torch::Tensor t) {
+  if (ref.dim() == 4 &&
+      ref.is_contiguous(torch::MemoryFormat::ChannelsLast) &&
+      !ref.is_contiguous())
+    return t.contiguous(torch::MemoryFormat::ChannelsLast);
+  return t.contiguous();
 }
 
 extern "C" {
@@ -19,18 +41,18 @@ extern "C" {
 void launch_relu_fwd(const float*, float*, long, void*);
 void launch_relu_bwd(const float*, const float*, float*, long, void*);
 void launch_add_relu(const float*, const float*, float*, long, void*);
-void launch_maxpool2x2_fwd(const float*, float*, uint8_t*, int, int, int,
-                           int, int, void*);
-void launch_maxpool2x2_bwd(const float*, const uint8_t*, float*, int, int,
+void launch_maxpool2x2_fwd(const float*, float*, uint8_t*, long, int, int,
                            int, int, int, void*);
+void launch_maxpool2x2_bwd(const float*, const uint8_t*, float*, long, int,
+                           int, int, int, int, void*);
 void launch_dropout_fwd(const float*, float*, uint8_t*, long, float,
                         uint64_t, uint64_t, void*);
 void launch_dropout_fwd_dev(const float*, float*, uint8_t*, long, float,
                             const unsigned long long*, int, void*);
 void launch_dropout_bwd(const float*, const uint8_t*, float*, long, float,
                         void*);
-void launch_gap_fwd(const float*, float*, int, int, void*);
-void launch_gap_bwd(const float*, float*, int, int, void*);
+void launch_gap_fwd(const float*, float*, long, int, int, void*);
+void launch_gap_bwd(const float*, float*, long, int, int, void*);
 void launch_ce_fwd(const float*, const long*, float*, float*, float*, int,
                    int, void*);
 void launch_ce_bwd(const float*, const long*, const float*, float*, int, int,
@@ -108,7 +130,7 @@ torch::Tensor relu_fwd(torch::Tensor x) {
 
 torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy) {
   CHK_CUDA(y);
-  dy = dy.contiguous();
+  dy = match_layout(y, dy);
   auto dx = torch::empty_like(y);
   launch_relu_bwd(y.data_ptr<float>(), dy.data_ptr<float>(),
                   dx.data_ptr<float>(), y.numel(), stream_of(y));
@@ -117,7 +139,7 @@ torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy) {
 
 torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
   CHK_CUDA(a);
-  b = b.contiguous();
+  b = match_layout(a, b);
   auto y = torch::empty_like(a);
   launch_add_relu(a.data_ptr<float>(), b.data_ptr<float>(),
                   y.data_ptr<float>(), a.numel(), stream_of(a));
@@ -125,26 +147,27 @@ torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
 }
 
 std::tuple<torch::Tensor, torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
-  CHK_CUDA(x);
-  CHK(x.dim() == 4);
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  x = cl(x);
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int OH = H / 2, OW = W / 2;
-  auto y = torch::empty({Nb, C, OH, OW}, x.options());
-  auto idx = torch::empty({Nb, C, OH, OW}, x.options().dtype(torch::kUInt8));
+  auto y = empty_cl({Nb, C, OH, OW}, x.options());
+  auto idx = empty_cl({Nb, C, OH, OW}, x.options().dtype(torch::kUInt8));
   launch_maxpool2x2_fwd(x.data_ptr<float>(), y.data_ptr<float>(),
-                        idx.data_ptr<uint8_t>(), Nb * C, H, W, OH, OW,
+                        idx.data_ptr<uint8_t>(), Nb, H, W, OH, OW, C,
                         stream_of(x));
   return {y, idx};
 }
 
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
                              std::vector<int64_t> in_shape) {
-  CHK_CUDA(dy);
+  TORCH_CHECK(dy.is_cuda());
+  dy = cl(dy);
   int Nb = in_shape[0], C = in_shape[1], H = in_shape[2], W = in_shape[3];
   int OH = dy.size(2), OW = dy.size(3);
-  auto dx = torch::empty({Nb, C, H, W}, dy.options());
+  auto dx = empty_cl({Nb, C, H, W}, dy.options());
   launch_maxpool2x2_bwd(dy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
-                        dx.data_ptr<float>(), Nb * C, H, W, OH, OW,
+                        dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
                         stream_of(dy));
   return dx;
 }
@@ -187,19 +210,21 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
 }
 
 torch::Tensor gap_fwd(torch::Tensor x) {
-  CHK_CUDA(x);
+  TORCH_CHECK(x.is_cuda());
+  x = cl(x);
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = torch::empty({Nb, C}, x.options());
-  launch_gap_fwd(x.data_ptr<float>(), y.data_ptr<float>(), Nb * C, HW,
+  launch_gap_fwd(x.data_ptr<float>(), y.data_ptr<float>(), Nb, HW, C,
                  stream_of(x));
   return y;
 }
 
 torch::Tensor gap_bwd(torch::Tensor dy, std::vector<int64_t> in_shape) {
-  CHK_CUDA(dy);
+  TORCH_CHECK(dy.is_cuda());
+  dy = dy.contiguous();
   int Nb = in_shape[0], C = in_shape[1], HW = in_shape[2] * in_shape[3];
-  auto dx = torch::empty({Nb, C, in_shape[2], in_shape[3]}, dy.options());
-  launch_gap_bwd(dy.data_ptr<float>(), dx.data_ptr<float>(), Nb * C, HW,
+  auto dx = empty_cl({Nb, C, in_shape[2], in_shape[3]}, dy.options());
+  launch_gap_bwd(dy.data_ptr<float>(), dx.data_ptr<float>(), Nb, HW, C,
                  stream_of(dy));
   return dx;
 }
@@ -399,8 +424,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> b, int64_t stride,
                          int64_t pad, bool relu) {
-  CHK_CUDA(x);
-  CHK_CUDA(w);
+  TORCH_CHECK(x.is_cuda() && w.is_cuda());
+  x = cl(x);
+  w = w.contiguous();
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
   int OH = (H + 2 * pad - R) / stride + 1;
@@ -408,7 +434,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
   auto wt = torch::empty({(long)C * R * S, Kout}, w.options());
   launch_wperm_crs_ko(w.data_ptr<float>(), wt.data_ptr<float>(), Kout, C,
                       R * S, stream_of(x));
-  auto y = torch::empty({Nb, Kout, OH, OW}, x.options());
+  auto y = empty_cl({Nb, Kout, OH, OW}, x.options());
   launch_conv_fwd(x.data_ptr<float>(), wt.data_ptr<float>(),
                   b ? b->data_ptr<float>() : nullptr, y.data_ptr<float>(),
                   Nb, C, H, W, Kout, R, S, OH, OW, (int)stride, (int)pad,
@@ -419,8 +445,10 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, int64_t stride,
     int64_t pad, bool has_b, bool need_dx) {
-  CHK_CUDA(x);
-  dy = dy.contiguous();
+  TORCH_CHECK(x.is_cuda());
+  x = cl(x);
+  w = w.contiguous();
+  dy = cl(dy);
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
   int OH = dy.size(2), OW = dy.size(3);
@@ -431,7 +459,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     auto wp = torch::empty({(long)Kout * R * S, C}, w.options());
     launch_wperm_kors_c(w.data_ptr<float>(), wp.data_ptr<float>(), Kout, C,
                         R * S, st);
-    dx = torch::empty_like(x);
+    dx = empty_cl({Nb, C, H, W}, x.options());
     launch_conv_bwd_data(dy.data_ptr<float>(), wp.data_ptr<float>(),
                          dx.data_ptr<float>(), Nb, C, H, W, Kout, R, S, OH,
                          OW, (int)stride, (int)pad, st);
@@ -443,15 +471,12 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   long Kdim = (long)Nb * OH * OW;
   int SK = conv_bwd_weight_splitk(Kout, Ncrs, Kdim);
   auto dw = torch::empty_like(w);
-  torch::Tensor ws;
-  float* wsp = nullptr;
-  if (SK > 1) {
-    ws = torch::empty({(long)SK * Kout * Ncrs}, w.options());
-    wsp = ws.data_ptr<float>();
-  }
+  // ws: SK slabs + one rsc-ordered temp (also needed when SK == 1)
+  auto ws = torch::empty({((long)SK + 1) * Kout * Ncrs}, w.options());
   launch_conv_bwd_weight(dy.data_ptr<float>(), x.data_ptr<float>(),
-                         dw.data_ptr<float>(), wsp, SK, Nb, C, H, W, Kout, R,
-                         S, OH, OW, (int)stride, (int)pad, st);
+                         dw.data_ptr<float>(), ws.data_ptr<float>(), SK, Nb,
+                         C, H, W, Kout, R, S, OH, OW, (int)stride, (int)pad,
+                         st);
 
   torch::Tensor db;
   if (has_b) {
@@ -471,12 +496,14 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b,
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
     double eps, bool training) {
-  CHK_CUDA(x);
+  TORCH_CHECK(x.is_cuda());
+  x = cl(x);
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
-  auto y = torch::empty_like(x);
+  auto y = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
+                    x.options());
   auto save_mean = torch::empty({C}, x.options());
   auto save_rstd = torch::empty({C}, x.options());
-  auto scratch = torch::empty({2 * C}, x.options());
+  auto scratch = torch::empty({(64 * 2 + 2) * (long)C}, x.options());
   launch_bn_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
                 b.data_ptr<float>(), running_mean.data_ptr<float>(),
                 running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
@@ -489,15 +516,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor save_mean,
     torch::Tensor save_rstd, torch::Tensor dy) {
-  CHK_CUDA(x);
+  TORCH_CHECK(x.is_cuda());
+  x = cl(x);
+  dy = cl(dy);
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
-  auto dx = torch::empty_like(x);
+  auto dx = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
+                     x.options());
   auto dw = torch::empty({C}, x.options());
   auto db = torch::empty({C}, x.options());
-  auto stats = torch::empty({2 * C}, x.options());
+  auto scratch = torch::empty({(64 * 2 + 2) * (long)C}, x.options());
   launch_bn_bwd(x.data_ptr<float>(), dy.data_ptr<float>(),
                 w.data_ptr<float>(), save_mean.data_ptr<float>(),
-                save_rstd.data_ptr<float>(), stats.data_ptr<float>(),
+                save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
                 dx.data_ptr<float>(), dw.data_ptr<float>(),
                 db.data_ptr<float>(), Nb, C, HW, 1, stream_of(x));
   return {dx, dw, db};
@@ -545,12 +575,12 @@ void poison_subf(torch::Tensor data, torch::Tensor idxs,
 
 torch::Tensor normalize_u8(torch::Tensor raw, torch::Tensor mean,
                            torch::Tensor stdv) {
-  CHK_CUDA(raw);
+  TORCH_CHECK(raw.is_cuda() && raw.is_contiguous());
   long B = raw.size(0);
   int H = raw.size(1), W = raw.size(2);
   int C = raw.dim() == 4 ? raw.size(3) : 1;
-  auto out = torch::empty({B, C, H, W},
-                          raw.options().dtype(torch::kFloat32));
+  // raw is HWC; output storage is NHWC = channels_last of (B,C,H,W)
+  auto out = empty_cl({B, C, H, W}, raw.options().dtype(torch::kFloat32));
   launch_normalize_u8(raw.data_ptr<uint8_t>(), out.data_ptr<float>(), B, H,
                       W, C, mean.data_ptr<float>(), stdv.data_ptr<float>(),
                       stream_of(raw));

@@ -1,20 +1,23 @@
-// Implicit-GEMM fp32 convolution on MFMA.
-// No im2col buffer: the patch matrix is gathered directly into LDS tiles
-// and fed to v_mfma_f32_16x16x4_f32 with the same bank padding as
-// gemm_f32.hip.
+// Implicit-GEMM fp32 convolution on MFMA — NHWC.
 //
-//   fwd:        GEMM M=Nb*OH*OW, N=K_out, Kdim=C*R*S (tile 128x64)
-//   bwd-data:   GEMM M=Nb*H*W,  N=C,     Kdim=K_out*R*S (tile 128x64),
-//               fractional-stride validity masks
-//   bwd-weight: GEMM M=K_out,   N=C*R*S, Kdim=Nb*OH*OW (tile 64x64 — Kout
-//               is small, a 128-row tile would waste half its M rows),
-//               deterministic split-K slabs + fixed-order reduce
+// Activations are channels_last (N,H,W,C storage): the K-dimension of the
+// implicit GEMM is ordered so its fastest index is the CONTIGUOUS channel
+// axis, which turns the patch gathers into float4 loads with ONE address
+// computation per 4 elements and makes the output stores coalesced.  The
+// first NCHW version of these kernels measured 13.7 VALU instructions per
+// MFMA (profiles/r01_bench_kernel_stats.md) — the gather address math, not
+// the matrix pipe, was the limiter.
 //
-// The (c,r,s) decomposition in the staging gathers is templated on the
-// kernel size (R_T in {1,3}; 0 = generic) and bwd-data on the stride
-// (S_T in {1,2}; 0 = generic): constant divisors compile to multiply-shift
-// instead of ~20-cycle integer division per staged element — the round-1
-// profile showed these gathers dominating (profiles/r01_bench_kernel_stats).
+//   fwd:        M=Nb*OH*OW, N=K_out, Kdim=(r,s,c) — C%32==0 makes a 32-wide
+//               K-tile sit inside one (r,s) tap: the tap decomposition is
+//               computed once per tile, not per element
+//   bwd-data:   M=Nb*H*W, N=C, Kdim=(r,s,ko), KO%32==0 fast path
+//   bwd-weight: M=K_out (64x64 tile), N=(r,s,c), Kdim=Nb*OH*OW, split-K
+//               slabs + fixed-order reduce; output permuted (r,s,c)->(c,r,s)
+//               to match the torch (KO,C,R,S) weight layout
+//
+// Weight transforms (tiny, per call): fwd wt[(r,s,c)][ko], bwd-data
+// wp[(r,s,ko)][c].  All reductions fixed-order; no atomics.
 #include "common.h"
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -28,22 +31,8 @@ struct ConvShape {
   int Nb, C, H, W, Kout, R, S, OH, OW, stride, pad;
 };
 
-template <int RT>
-__device__ __forceinline__ void crs_decomp(int k, const ConvShape& sh,
-                                           int& c, int& r, int& s) {
-  if (RT > 0) {
-    s = k % RT;
-    r = (k / RT) % RT;
-    c = k / (RT * RT);
-  } else {
-    s = k % sh.S;
-    r = (k / sh.S) % sh.R;
-    c = k / (sh.S * sh.R);
-  }
-}
-
-// Fragment compute over one staged K-tile.  Wave grid is 2x2; each wave
-// owns an (MI*16 x NI*16) output subtile.
+// Fragment compute over one staged K-tile; wave grid 2x2, each wave owns
+// an (MI*16 x NI*16) subtile.
 template <int MI, int NI>
 __device__ __forceinline__ void mfma_tile(const float* __restrict__ Abuf,
                                           const float* __restrict__ Bbuf,
@@ -71,7 +60,8 @@ __device__ __forceinline__ void mfma_tile(const float* __restrict__ Abuf,
 
 // ------------------------------------------------------------------- fwd
 
-template <int RT, bool P0>
+// V4: C % 32 == 0 (vectorized channel gather).  P0: pad == 0 (no bounds).
+template <bool P0, bool V4>
 __global__ __launch_bounds__(256)
 void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
                 const float* __restrict__ bias, float* __restrict__ y,
@@ -95,41 +85,58 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
   const int am = t >> 3, ak = (t & 7) * 4;
   const int bk = t >> 4, bn = (t & 15) * 4;
 
-  // per-thread m -> (nb, oh, ow) decomposition, once per staging row
-  int ows[BM / 32], ohs[BM / 32], nbs[BM / 32];
+  // per-thread m -> (nb, oh, ow), once
+  int ow0[BM / 32], oh0[BM / 32];
+  long base[BM / 32];  // nb * H*W*C
   bool mval[BM / 32];
 #pragma unroll
   for (int j = 0; j < BM / 32; ++j) {
     long gm = m_blk + am + j * 32;
     mval[j] = gm < M;
     long gmc = mval[j] ? gm : 0;
-    ows[j] = gmc % sh.OW;
-    ohs[j] = (gmc / sh.OW) % sh.OH;
-    nbs[j] = gmc / ((long)sh.OW * sh.OH);
+    int ow = gmc % sh.OW;
+    int oh = (gmc / sh.OW) % sh.OH;
+    long nb = gmc / ((long)sh.OW * sh.OH);
+    ow0[j] = ow * sh.stride - sh.pad;
+    oh0[j] = oh * sh.stride - sh.pad;
+    base[j] = nb * (long)sh.H * sh.W * sh.C;
   }
 
-  // async-STAGE split (guide T14/G15): loads are issued to REGISTERS a
-  // full K-tile early and the LDS write happens after the barrier, so HBM
-  // latency hides under the previous tile's MFMAs.
   float ra[BM / 32][4];
   float4 rb[2];
   auto stage_load = [&](int k0) {
+    if (V4) {
+      // one tap per 32-wide K-tile: decompose once per tile
+      int rs = k0 / sh.C;
+      int r = rs / sh.S, s = rs % sh.S;
+      int c0 = k0 - rs * sh.C + ak;
 #pragma unroll
-    for (int j = 0; j < BM / 32; ++j) {
-      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
-      if (mval[j]) {
-        int oh0 = ohs[j] * sh.stride - sh.pad;
-        int ow0 = ows[j] * sh.stride - sh.pad;
-        const float* xp = x + ((long)nbs[j] * sh.C) * sh.H * sh.W;
+      for (int j = 0; j < BM / 32; ++j) {
+        int ih = oh0[j] + r, iw = ow0[j] + s;
+        float4 q = {0.f, 0.f, 0.f, 0.f};
+        if (mval[j] &&
+            (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
+          q = *(const float4*)(x + base[j] +
+                               ((long)ih * sh.W + iw) * sh.C + c0);
+        ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      }
+    } else {
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          int k = k0 + ak + e;
-          if (k < Kdim) {
-            int c, r, s;
-            crs_decomp<RT>(k, sh, c, r, s);
-            int ih = oh0 + r, iw = ow0 + s;
-            if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
-              ra[j][e] = xp[((long)c * sh.H + ih) * sh.W + iw];
+      for (int j = 0; j < BM / 32; ++j) {
+        ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
+        if (mval[j]) {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            int k = k0 + ak + e;
+            if (k < Kdim) {
+              int c = k % sh.C;
+              int rs = k / sh.C;
+              int r = rs / sh.S, s = rs % sh.S;
+              int ih = oh0[j] + r, iw = ow0[j] + s;
+              if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
+                ra[j][e] =
+                    x[base[j] + ((long)ih * sh.W + iw) * sh.C + c];
+            }
           }
         }
       }
@@ -179,6 +186,7 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
     buf ^= 1;
   }
 
+  // NHWC epilogue: ko = col is the contiguous axis -> coalesced stores
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -189,23 +197,21 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
       for (int r = 0; r < 4; ++r) {
         long m = m_blk + wr * 64 + mi * 16 + l4 * 4 + r;
         if (m >= M) continue;
-        int ow = m % sh.OW;
-        int oh = (m / sh.OW) % sh.OH;
-        int nb = m / ((long)sh.OW * sh.OH);
         float v = acc[mi][ni][r];
         if (bias) v += bias[ko];
         if (relu) v = fmaxf(v, 0.f);
-        y[(((long)nb * sh.Kout + ko) * sh.OH + oh) * sh.OW + ow] = v;
+        y[m * sh.Kout + ko] = v;
       }
     }
 }
 
 // -------------------------------------------------------------- bwd-data
 
-template <int RT, int ST>
+// Kdim order (r,s,ko); V4: KO % 32 == 0.  ST: compile-time stride.
+template <int ST, bool V4>
 __global__ __launch_bounds__(256)
 void conv_bwd_data_k(const float* __restrict__ dy,
-                     const float* __restrict__ wp,  // [(ko,r,s)][C]
+                     const float* __restrict__ wp,  // [(r,s,ko)][C]
                      float* __restrict__ dx, ConvShape sh, int Kdim) {
   constexpr int BM = 128, MI = 4, NI = 2;
   __shared__ float A_lds[2][BM * LDA_S];
@@ -227,40 +233,59 @@ void conv_bwd_data_k(const float* __restrict__ dy,
   const int bk = t >> 4, bn = (t & 15) * 4;
   const int stride = ST > 0 ? ST : sh.stride;
 
-  int iws[BM / 32], ihs[BM / 32], nbs[BM / 32];
+  int iwp[BM / 32], ihp[BM / 32];
+  long base[BM / 32];  // nb * OH*OW*KO
   bool mval[BM / 32];
 #pragma unroll
   for (int j = 0; j < BM / 32; ++j) {
     long gm = m_blk + am + j * 32;
     mval[j] = gm < M;
     long gmc = mval[j] ? gm : 0;
-    iws[j] = gmc % sh.W;
-    ihs[j] = (gmc / sh.W) % sh.H;
-    nbs[j] = gmc / ((long)sh.W * sh.H);
+    iwp[j] = (int)(gmc % sh.W) + sh.pad;
+    ihp[j] = (int)((gmc / sh.W) % sh.H) + sh.pad;
+    base[j] = (gmc / ((long)sh.W * sh.H)) * (long)sh.OH * sh.OW * sh.Kout;
   }
 
   float ra[BM / 32][4];
   float4 rb[2];
   auto stage_load = [&](int k0) {
+    if (V4) {
+      int rs = k0 / sh.Kout;
+      int r = rs / sh.S, s = rs % sh.S;
+      int ko0 = k0 - rs * sh.Kout + ak;
 #pragma unroll
-    for (int j = 0; j < BM / 32; ++j) {
-      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
-      if (mval[j]) {
-        int ihp = ihs[j] + sh.pad, iwp = iws[j] + sh.pad;
-        const float* dyp =
-            dy + ((long)nbs[j] * sh.Kout) * sh.OH * sh.OW;
+      for (int j = 0; j < BM / 32; ++j) {
+        int ohn = ihp[j] - r, own = iwp[j] - s;
+        float4 q = {0.f, 0.f, 0.f, 0.f};
+        if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+            own % stride == 0) {
+          int oh = ohn / stride, ow = own / stride;
+          if (oh < sh.OH && ow < sh.OW)
+            q = *(const float4*)(dy + base[j] +
+                                 ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+        }
+        ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      }
+    } else {
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          int k = k0 + ak + e;
-          if (k < Kdim) {
-            int ko, r, s;
-            crs_decomp<RT>(k, sh, ko, r, s);
-            int ohn = ihp - r, own = iwp - s;
-            if (ohn >= 0 && own >= 0 && ohn % stride == 0 &&
-                own % stride == 0) {
-              int oh = ohn / stride, ow = own / stride;
-              if (oh < sh.OH && ow < sh.OW)
-                ra[j][e] = dyp[((long)ko * sh.OH + oh) * sh.OW + ow];
+      for (int j = 0; j < BM / 32; ++j) {
+        ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
+        if (mval[j]) {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            int k = k0 + ak + e;
+            if (k < Kdim) {
+              int ko = k % sh.Kout;
+              int rs = k / sh.Kout;
+              int r = rs / sh.S, s = rs % sh.S;
+              int ohn = ihp[j] - r, own = iwp[j] - s;
+              if (ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+                  own % stride == 0) {
+                int oh = ohn / stride, ow = own / stride;
+                if (oh < sh.OH && ow < sh.OW)
+                  ra[j][e] = dy[base[j] +
+                                ((long)oh * sh.OW + ow) * sh.Kout + ko];
+              }
             }
           }
         }
@@ -321,21 +346,18 @@ void conv_bwd_data_k(const float* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         long m = m_blk + wr * 64 + mi * 16 + l4 * 4 + r;
         if (m >= M) continue;
-        int iw = m % sh.W;
-        int ih = (m / sh.W) % sh.H;
-        int nb = m / ((long)sh.W * sh.H);
-        dx[(((long)nb * sh.C + c) * sh.H + ih) * sh.W + iw] =
-            acc[mi][ni][r];
+        dx[m * sh.C + c] = acc[mi][ni][r];
       }
     }
 }
 
 // ------------------------------------------------------------ bwd-weight
 
-// 64x64 tile (MI=NI=2): Kout rarely exceeds 64 per tile row and a 128-row
-// tile would idle half its MFMAs.  gridDim.z = split-K chunks over
-// m = (nb,oh,ow); partial slabs [z][Kout][C*R*S].
-template <int RT, bool P0>
+// 64x64 tile over (ko, crs=(r,s,c)); Kdim = m = (nb,oh,ow); split-K.
+// A = dy^T gathered NHWC (float4 over ko, transposed into LDS);
+// B = x patches (float4 over c when C%4==0).
+// Slabs/output are in (r,s,c) column order; dwperm converts to (c,r,s).
+template <bool P0, bool V4>
 __global__ __launch_bounds__(256)
 void conv_bwd_weight_k(const float* __restrict__ dy,
                        const float* __restrict__ x, float* __restrict__ out,
@@ -355,90 +377,102 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
   const int m_blk = blockIdx.x * BM;   // over Kout
-  const int n_blk = blockIdx.y * BN;   // over C*R*S
+  const int n_blk = blockIdx.y * BN;   // over (r,s,c)
   const long Kdim = (long)sh.Nb * sh.OH * sh.OW;
   const long k_lo = (long)blockIdx.z * k_per_chunk;
   const long k_hi = min(Kdim, k_lo + k_per_chunk);
-  // A staging: BM*BK = 2048 floats / 256 threads = 8 = 2 x float4
-  const int am = t >> 3, ak = (t & 7) * 4;    // am 0..31, 2 rounds
-  const int bk = t >> 4, bn = (t & 15) * 4;
 
-  // precompute the (c,r,s) for this thread's 4 B columns (fixed all tiles)
-  int bc[4], br[4], bs[4];
-  bool bvalid[4];
-#pragma unroll
-  for (int e = 0; e < 4; ++e) {
-    int crs = n_blk + bn + e;
-    bvalid[e] = crs < Ncrs;
-    crs_decomp<RT>(bvalid[e] ? crs : 0, sh, bc[e], br[e], bs[e]);
+  // A staging: float4 over ko (lanes 0-15 cover 64 consecutive ko); m from
+  // the upper thread bits, 2 rounds cover BK=32 m.
+  const int ako = (t & 15) * 4;
+  const int amr = t >> 4;  // 0..15, +16 per round
+  // B staging: float4 over crs (c fastest); 2 rounds over m.
+  const int bn4 = (t & 15) * 4;
+  const int bmr = t >> 4;
+
+  // (r,s,c0) of this thread's B quad (fixed across tiles)
+  int br_, bs_, bc0_;
+  {
+    int crs = min(n_blk + bn4, Ncrs - 1);
+    int rs = crs / sh.C;
+    br_ = rs / sh.S;
+    bs_ = rs % sh.S;
+    bc0_ = crs - rs * sh.C;
   }
 
-  float ra[2][4];
-  float4 rb[2];
+  float raA[2][4];
+  float4 rbB[2];
   auto stage_load = [&](long k0) {
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int ko = m_blk + am + j * 32;
-      ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
-      if (ko < sh.Kout) {
-        // 4 consecutive m share (nb, oh) almost always; slow path on wrap
-        long k = k0 + ak;
+      long k = k0 + amr + j * 16;  // the m index
+      float4 q = {0.f, 0.f, 0.f, 0.f};
+      if (k < k_hi) {
+        if ((sh.Kout % 4) == 0 && m_blk + ako + 3 < sh.Kout) {
+          q = *(const float4*)(dy + k * sh.Kout + m_blk + ako);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 4; ++e)
+            if (m_blk + ako + e < sh.Kout)
+              (&q.x)[e] = dy[k * sh.Kout + m_blk + ako + e];
+        }
+      }
+      raA[j][0] = q.x; raA[j][1] = q.y; raA[j][2] = q.z; raA[j][3] = q.w;
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      long k = k0 + bmr + j * 16;
+      float4 q = {0.f, 0.f, 0.f, 0.f};
+      if (k < k_hi) {
         int ow = k % sh.OW;
         int oh = (k / sh.OW) % sh.OH;
-        int nb = k / ((long)sh.OW * sh.OH);
-        const float* dyp =
-            dy + (((long)nb * sh.Kout + ko) * sh.OH + oh) * sh.OW;
+        long nb = k / ((long)sh.OW * sh.OH);
+        long xb = nb * (long)sh.H * sh.W * sh.C;
+        if (V4) {
+          int ih = oh * sh.stride - sh.pad + br_;
+          int iw = ow * sh.stride - sh.pad + bs_;
+          if (n_blk + bn4 + 3 < Ncrs &&
+              (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
+            q = *(const float4*)(x + xb + ((long)ih * sh.W + iw) * sh.C +
+                                 bc0_);
+          else if (n_blk + bn4 < Ncrs &&
+                   (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))) {
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          if (k + e < k_hi) {
-            int owe = ow + e;
-            if (owe < sh.OW)
-              ra[j][e] = dyp[owe];
-            else {
-              long ke = k + e;
-              int ow2 = ke % sh.OW;
-              int oh2 = (ke / sh.OW) % sh.OH;
-              int nb2 = ke / ((long)sh.OW * sh.OH);
-              ra[j][e] = dy[(((long)nb2 * sh.Kout + ko) * sh.OH + oh2) *
-                                sh.OW + ow2];
+            for (int e = 0; e < 4; ++e)
+              if (n_blk + bn4 + e < Ncrs)
+                (&q.x)[e] = x[xb + ((long)ih * sh.W + iw) * sh.C + bc0_ + e];
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            int crs = n_blk + bn4 + e;
+            if (crs < Ncrs) {
+              int rs = crs / sh.C;
+              int c = crs - rs * sh.C;
+              int r = rs / sh.S, s = rs % sh.S;
+              int ih = oh * sh.stride - sh.pad + r;
+              int iw = ow * sh.stride - sh.pad + s;
+              if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
+                (&q.x)[e] = x[xb + ((long)ih * sh.W + iw) * sh.C + c];
             }
           }
         }
       }
-    }
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      long k = k0 + bk + j * 16;
-      float v[4] = {0.f, 0.f, 0.f, 0.f};
-      if (k < k_hi) {
-        int ow = k % sh.OW;
-        int oh = (k / sh.OW) % sh.OH;
-        int nb = k / ((long)sh.OW * sh.OH);
-        int ih0 = oh * sh.stride - sh.pad;
-        int iw0 = ow * sh.stride - sh.pad;
-        const float* xp = x + ((long)nb * sh.C) * sh.H * sh.W;
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          if (bvalid[e]) {
-            int ih = ih0 + br[e], iw = iw0 + bs[e];
-            if (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))
-              v[e] = xp[((long)bc[e] * sh.H + ih) * sh.W + iw];
-          }
-        }
-      }
-      rb[j] = {v[0], v[1], v[2], v[3]};
+      rbB[j] = q;
     }
   };
   auto stage_write = [&](int buf) {
+    // A transposed: element (m, ko+i) -> A_lds[ko+i][m]
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
-      ((float2*)dst)[0] = {ra[j][0], ra[j][1]};
-      ((float2*)dst)[1] = {ra[j][2], ra[j][3]};
+      int m = amr + j * 16;
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        A_lds[buf][(ako + i) * LDA_S + m] = raA[j][i];
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j)
-      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+      *(float4*)&B_lds[buf][(bmr + j * 16) * LDB_S + bn4] = rbB[j];
   };
 
   stage_load(k_lo);
@@ -456,6 +490,7 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     buf ^= 1;
   }
 
+  // A/B roles here: A rows = ko, B cols = (r,s,c)
 #pragma unroll
   for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -475,71 +510,77 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     }
 }
 
-// ---- conv bias gradient: db[k] = sum over (nb,oh,ow) of dy ----
-// Two-stage deterministic reduce: stage 1 fills partials[k][chunk] from a
-// (Kout x NCHUNK) grid (plenty of blocks for 256 CUs); stage 2 is one wave
-// per k over the NCHUNK partials.
-constexpr int kDbChunks = 64;
-
-__global__ void conv_db_stage1_k(const float* __restrict__ dy,
-                                 float* __restrict__ partials, int Nb,
-                                 int Kout, int OHW) {
-  int k = blockIdx.x;
-  int chunk = blockIdx.y;
-  long total = (long)Nb * OHW;
-  long per = (total + kDbChunks - 1) / kDbChunks;
-  long lo = chunk * per, hi = min(total, lo + per);
-  __shared__ float sh[kBlock];
-  float acc = 0.f;
-  for (long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-    long nb = i / OHW, px = i % OHW;
-    acc += dy[(nb * Kout + k) * (long)OHW + px];
-  }
-  sh[threadIdx.x] = acc;
-  __syncthreads();
-  for (int off = kBlock / 2; off > 0; off >>= 1) {
-    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
-    __syncthreads();
-  }
-  if (threadIdx.x == 0) partials[(long)k * kDbChunks + chunk] = sh[0];
-}
-
-__global__ void conv_db_stage2_k(const float* __restrict__ partials,
-                                 float* __restrict__ db, int Kout) {
-  int k = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
-  int lane = threadIdx.x % kWave;
-  if (k >= Kout) return;
-  float acc = (lane < kDbChunks) ? partials[(long)k * kDbChunks + lane] : 0.f;
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
-  if (lane == 0) db[k] = acc;
-}
-
-// permute w (Kout,C,R,S) -> dst layouts
-__global__ void wperm_crs_ko_k(const float* __restrict__ w,
-                               float* __restrict__ out, int Kout, int C,
-                               int RS) {
+// dw column permute: [(ko)][(r,s,c)] -> torch layout [(ko)][(c,r,s)]
+__global__ void dwperm_rsc_crs_k(const float* __restrict__ in,
+                                 float* __restrict__ out, int Kout, int C,
+                                 int RS) {
   long n = (long)Kout * C * RS;
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    int ko = i / (C * RS);
-    int crs = i % (C * RS);
-    out[(long)crs * Kout + ko] = w[i];
+    int col = i % (C * RS);   // (r,s,c) flat
+    long ko = i / (C * RS);
+    int c = col % C;
+    int rs = col / C;
+    out[(ko * C + c) * RS + rs] = in[i];
   }
 }
 
-__global__ void wperm_kors_c_k(const float* __restrict__ w,
+// ---- conv bias gradient (NHWC): db[ko] = column sum of dy [M][KO] ----
+constexpr int kDbChunks = 64;
+
+__global__ void conv_db_stage1_k(const float* __restrict__ dy,
+                                 float* __restrict__ partials, long M,
+                                 int Kout) {
+  // grid (chunk, ko-group): consecutive threads cover consecutive ko
+  // (coalesced); each walks its row chunk.
+  int chunk = blockIdx.x;
+  int ko = blockIdx.y * blockDim.x + threadIdx.x;
+  if (ko >= Kout) return;
+  long per = (M + kDbChunks - 1) / kDbChunks;
+  long lo = (long)chunk * per, hi = min(M, lo + per);
+  float acc = 0.f;
+  for (long m = lo; m < hi; ++m) acc += dy[m * Kout + ko];
+  partials[(long)chunk * Kout + ko] = acc;
+}
+
+__global__ void conv_db_stage2_k(const float* __restrict__ partials,
+                                 float* __restrict__ db, int Kout) {
+  int ko = blockIdx.x * blockDim.x + threadIdx.x;
+  if (ko >= Kout) return;
+  float acc = 0.f;
+  for (int c = 0; c < kDbChunks; ++c) acc += partials[(long)c * Kout + ko];
+  db[ko] = acc;
+}
+
+// permute w (Kout,C,R,S) -> staged layouts
+__global__ void wperm_rsc_ko_k(const float* __restrict__ w,
                                float* __restrict__ out, int Kout, int C,
                                int RS) {
+  // out[((rs)*C + c)*Kout + ko] = w[(ko*C + c)*RS + rs]
   long n = (long)Kout * C * RS;
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
     int rs = i % RS;
     int c = (i / RS) % C;
-    int ko = i / ((long)RS * C);
-    out[((long)ko * RS + rs) * C + c] = w[i];
+    long ko = i / ((long)RS * C);
+    out[((long)rs * C + c) * Kout + ko] = w[i];
+  }
+}
+
+__global__ void wperm_rsko_c_k(const float* __restrict__ w,
+                               float* __restrict__ out, int Kout, int C,
+                               int RS) {
+  // out[((rs)*Kout + ko)*C + c] = w[(ko*C + c)*RS + rs]
+  long n = (long)Kout * C * RS;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int rs = i % RS;
+    int c = (i / RS) % C;
+    long ko = i / ((long)RS * C);
+    out[((long)rs * Kout + ko) * C + c] = w[i];
   }
 }
 
@@ -557,14 +598,19 @@ void launch_conv_fwd(const float* x, const float* wt, const float* bias,
   long M = (long)Nb * OH * OW;
   dim3 grid((M + 127) / 128, (Kout + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
-  if (R == 3 && S == 3 && pad == 0)
-    conv_fwd_k<3, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
-  else if (R == 3 && S == 3)
-    conv_fwd_k<3, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
-  else if (R == 1 && S == 1 && pad == 0)
-    conv_fwd_k<1, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+  bool v4 = (C % 32) == 0;
+  if (pad == 0 && v4)
+    conv_fwd_k<true, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
+                                                 relu);
+  else if (pad == 0)
+    conv_fwd_k<true, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
+                                                  relu);
+  else if (v4)
+    conv_fwd_k<false, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
+                                                  relu);
   else
-    conv_fwd_k<0, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim, relu);
+    conv_fwd_k<false, false><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
+                                                   relu);
 }
 
 void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
@@ -576,27 +622,29 @@ void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
   long M = (long)Nb * H * W;
   dim3 grid((M + 127) / 128, (C + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
-  if (R == 3 && S == 3 && stride == 1)
-    conv_bwd_data_k<3, 1><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
-  else if (R == 3 && S == 3 && stride == 2)
-    conv_bwd_data_k<3, 2><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
-  else if (R == 1 && S == 1 && stride == 2)
-    conv_bwd_data_k<1, 2><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
-  else if (R == 1 && S == 1 && stride == 1)
-    conv_bwd_data_k<1, 1><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+  bool v4 = (Kout % 32) == 0;
+  if (stride == 1 && v4)
+    conv_bwd_data_k<1, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+  else if (stride == 2 && v4)
+    conv_bwd_data_k<2, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+  else if (stride == 1)
+    conv_bwd_data_k<1, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+  else if (stride == 2)
+    conv_bwd_data_k<2, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
   else
-    conv_bwd_data_k<0, 0><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<0, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
 }
 
 int conv_bwd_weight_splitk(int Kout, int Ncrs, long Kdim) {
   long tiles = ((Kout + 63) / 64) * (long)((Ncrs + BN - 1) / BN);
   if (tiles >= 192 || Kdim <= 2 * BK) return 1;
-  long want = (512 + tiles - 1) / tiles;  // 2 blocks/CU: latency hiding
+  long want = (512 + tiles - 1) / tiles;  // 2 blocks/CU for latency hiding
   long max_chunks = (Kdim + BK - 1) / BK;
   long sk = want < max_chunks ? want : max_chunks;
   return (int)(sk < 1 ? 1 : (sk > 256 ? 256 : sk));
 }
 
+// ws: SK*Kout*Ncrs + Kout*Ncrs floats (split-K slabs + rsc-ordered temp)
 void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
                             float* ws, int SK, int Nb, int C, int H, int W,
                             int Kout, int R, int S, int OH, int OW,
@@ -608,41 +656,46 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
       SK == 1 ? Kdim : (((Kdim + SK - 1) / SK + BK - 1) / BK) * BK;
   dim3 grid((Kout + 63) / 64, (Ncrs + BN - 1) / BN, SK);
   hipStream_t st = (hipStream_t)s;
-  float* out = SK == 1 ? dw : ws;
-  if (R == 3 && S == 3 && pad == 0)
-    conv_bwd_weight_k<3, true><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                                     k_per_chunk, SK == 1);
-  else if (R == 3 && S == 3)
-    conv_bwd_weight_k<3, false><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                                      k_per_chunk, SK == 1);
-  else if (R == 1 && S == 1 && pad == 0)
-    conv_bwd_weight_k<1, true><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                                     k_per_chunk, SK == 1);
+  float* slabs = ws;                         // SK * Kout * Ncrs
+  float* rsc = ws + (long)SK * Kout * Ncrs;  // Kout * Ncrs, rsc order
+  float* target = SK == 1 ? rsc : slabs;
+  bool v4 = (C % 4) == 0;
+  if (pad == 0 && v4)
+    conv_bwd_weight_k<true, true><<<grid, 256, 0, st>>>(
+        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  else if (pad == 0)
+    conv_bwd_weight_k<true, false><<<grid, 256, 0, st>>>(
+        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  else if (v4)
+    conv_bwd_weight_k<false, true><<<grid, 256, 0, st>>>(
+        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
   else
-    conv_bwd_weight_k<0, false><<<grid, 256, 0, st>>>(dy, x, out, sh, Ncrs,
-                                                      k_per_chunk, SK == 1);
+    conv_bwd_weight_k<false, false><<<grid, 256, 0, st>>>(
+        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
   if (SK > 1)
-    launch_splitk_reduce(ws, dw, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
+    launch_splitk_reduce(slabs, rsc, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
+  dwperm_rsc_crs_k<<<grid_for((long)Kout * Ncrs), kBlock, 0, st>>>(
+      rsc, dw, Kout, C, R * S);
 }
 
 void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
                     int Kout, int OHW, void* s) {
   hipStream_t st = (hipStream_t)s;
-  conv_db_stage1_k<<<dim3(Kout, kDbChunks), kBlock, 0, st>>>(dy, partials,
-                                                             Nb, Kout, OHW);
-  int wpb = kBlock / kWave;
-  conv_db_stage2_k<<<(Kout + wpb - 1) / wpb, kBlock, 0, st>>>(partials, db,
-                                                              Kout);
+  long M = (long)Nb * OHW;
+  dim3 g1(kDbChunks, (Kout + kBlock - 1) / kBlock);
+  conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout);
+  conv_db_stage2_k<<<(Kout + kBlock - 1) / kBlock, kBlock, 0, st>>>(
+      partials, db, Kout);
 }
 
 void launch_wperm_crs_ko(const float* w, float* out, int Kout, int C, int RS,
                          void* s) {
-  wperm_crs_ko_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
+  wperm_rsc_ko_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
                    (hipStream_t)s>>>(w, out, Kout, C, RS);
 }
 void launch_wperm_kors_c(const float* w, float* out, int Kout, int C, int RS,
                          void* s) {
-  wperm_kors_c_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
+  wperm_rsko_c_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
                    (hipStream_t)s>>>(w, out, Kout, C, RS);
 }
 }

@@ -72,23 +72,27 @@ void launch_add_relu(const float* a, const float* b, float* y, long n,
 }
 }
 
-// ------------------------------------------------------------ maxpool 2x2
+// ------------------------------------------------------- maxpool 2x2 NHWC
 
-// One thread per output element; idx stores the 2x2 argmax (0..3) for the
-// gather-style backward (no zero-init, write-once).
+// channels_last: c is the fastest axis, so threads over the flat index are
+// coalesced.  idx stores the 2x2 argmax (0..3) for the gather backward.
 __global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
                                  float* __restrict__ y,
                                  uint8_t* __restrict__ idx,
-                                 int NC, int H, int W, int OH, int OW) {
-  long n_out = (long)NC * OH * OW;
+                                 long B, int H, int W, int OH, int OW,
+                                 int C) {
+  long n_out = B * OH * (long)OW * C;
   long stride = (long)gridDim.x * blockDim.x;
+  int WC = W * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
        i += stride) {
-    int ow = i % OW;
-    int oh = (i / OW) % OH;
-    long nc = i / ((long)OW * OH);
-    const float* p = x + (nc * H + 2 * oh) * W + 2 * ow;
-    float v00 = p[0], v01 = p[1], v10 = p[W], v11 = p[W + 1];
+    int c = i % C;
+    long rest = i / C;
+    int ow = rest % OW;
+    int oh = (rest / OW) % OH;
+    long b = rest / ((long)OW * OH);
+    const float* p = x + ((b * H + 2 * oh) * (long)W + 2 * ow) * C + c;
+    float v00 = p[0], v01 = p[C], v10 = p[WC], v11 = p[WC + C];
     float m = v00; uint8_t a = 0;
     if (v01 > m) { m = v01; a = 1; }
     if (v10 > m) { m = v10; a = 2; }
@@ -98,23 +102,25 @@ __global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
   }
 }
 
-// Gather form: one thread per INPUT element; contributes dy iff it was the
-// argmax of its (unique) window.  Odd H/W tails (never pooled) get 0.
+// Gather form: one thread per INPUT element (write-once, no zero-init).
 __global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
                                  const uint8_t* __restrict__ idx,
                                  float* __restrict__ dx,
-                                 int NC, int H, int W, int OH, int OW) {
-  long n_in = (long)NC * H * W;
+                                 long B, int H, int W, int OH, int OW,
+                                 int C) {
+  long n_in = B * H * (long)W * C;
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_in;
        i += stride) {
-    int iw = i % W;
-    int ih = (i / W) % H;
-    long nc = i / ((long)W * H);
+    int c = i % C;
+    long rest = i / C;
+    int iw = rest % W;
+    int ih = (rest / W) % H;
+    long b = rest / ((long)W * H);
     int oh = ih >> 1, ow = iw >> 1;
     float g = 0.f;
     if (oh < OH && ow < OW) {
-      long o = (nc * OH + oh) * OW + ow;
+      long o = ((b * OH + oh) * (long)OW + ow) * C + c;
       uint8_t a = ((ih & 1) << 1) | (iw & 1);
       if (idx[o] == a) g = dy[o];
     }
@@ -123,18 +129,63 @@ __global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
 }
 
 extern "C" {
-void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, int NC,
-                           int H, int W, int OH, int OW, void* s) {
-  long n = (long)NC * OH * OW;
-  maxpool2x2_fwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, idx, NC,
-                                                               H, W, OH, OW);
+void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, long B,
+                           int H, int W, int OH, int OW, int C, void* s) {
+  long n = B * OH * (long)OW * C;
+  maxpool2x2_fwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, idx, B,
+                                                               H, W, OH, OW,
+                                                               C);
 }
 void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
-                           int NC, int H, int W, int OH, int OW, void* s) {
-  long n = (long)NC * H * W;
+                           long B, int H, int W, int OH, int OW, int C,
+                           void* s) {
+  long n = B * H * (long)W * C;
   maxpool2x2_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx,
-                                                               NC, H, W, OH,
-                                                               OW);
+                                                               B, H, W, OH,
+                                                               OW, C);
+}
+}
+
+// ----------------------------------------------------- global avgpool NHWC
+
+// out[b][c] = mean over HW; thread per (b,c): consecutive threads read
+// consecutive c (coalesced column walk).
+__global__ void gap_fwd_k(const float* __restrict__ x, float* __restrict__ y,
+                          long B, int HW, int C) {
+  long n = B * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / C;
+    const float* p = x + b * (long)HW * C + c;
+    float acc = 0.f;
+    for (int hw = 0; hw < HW; ++hw) acc += p[(long)hw * C];
+    y[i] = acc / HW;
+  }
+}
+
+__global__ void gap_bwd_k(const float* __restrict__ dy,
+                          float* __restrict__ dx, long B, int HW, int C) {
+  long n = B * (long)HW * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / ((long)HW * C);
+    dx[i] = dy[b * C + c] / HW;
+  }
+}
+
+extern "C" {
+void launch_gap_fwd(const float* x, float* y, long B, int HW, int C,
+                    void* s) {
+  gap_fwd_k<<<grid_for(B * C), kBlock, 0, (hipStream_t)s>>>(x, y, B, HW, C);
+}
+void launch_gap_bwd(const float* dy, float* dx, long B, int HW, int C,
+                    void* s) {
+  gap_bwd_k<<<grid_for(B * (long)HW * C), kBlock, 0, (hipStream_t)s>>>(
+      dy, dx, B, HW, C);
 }
 }
 
@@ -220,43 +271,6 @@ void launch_dropout_bwd(const float* dy, const uint8_t* mask, float* dx,
                         long n, float p, void* s) {
   dropout_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, mask, dx, n,
                                                             p);
-}
-}
-
-// ---------------------------------------------------------- global avgpool
-
-__global__ void gap_fwd_k(const float* __restrict__ x, float* __restrict__ y,
-                          int NC, int HW) {
-  // one wave per (n,c) plane
-  int nc = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
-  int lane = threadIdx.x % kWave;
-  if (nc >= NC) return;
-  const float* p = x + (long)nc * HW;
-  float acc = 0.f;
-  for (int i = lane; i < HW; i += kWave) acc += p[i];
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1)
-    acc += __shfl_down(acc, off, kWave);
-  if (lane == 0) y[nc] = acc / HW;
-}
-
-__global__ void gap_bwd_k(const float* __restrict__ dy,
-                          float* __restrict__ dx, int NC, int HW) {
-  long n = (long)NC * HW;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride)
-    dx[i] = dy[i / HW] / HW;
-}
-
-extern "C" {
-void launch_gap_fwd(const float* x, float* y, int NC, int HW, void* s) {
-  int wpb = kBlock / kWave;
-  gap_fwd_k<<<(NC + wpb - 1) / wpb, kBlock, 0, (hipStream_t)s>>>(x, y, NC, HW);
-}
-void launch_gap_bwd(const float* dy, float* dx, int NC, int HW, void* s) {
-  gap_bwd_k<<<grid_for((long)NC * HW), kBlock, 0, (hipStream_t)s>>>(dy, dx,
-                                                                    NC, HW);
 }
 }
 

@@ -68,7 +68,8 @@ __global__ void poison_subf_k(float* __restrict__ data,
   }
 }
 
-// ---- normalize: u8 (B,H,W) or (B,H,W,C) -> f32 NCHW, (x/255 - m)/s ----
+// ---- normalize: u8 HWC raw -> f32 channels_last (NHWC storage): the
+// layouts MATCH, so this is a pure elementwise pass ----
 __global__ void normalize_u8_k(const uint8_t* __restrict__ raw,
                                float* __restrict__ out, long B, int H, int W,
                                int C, const float* __restrict__ mean,
@@ -77,13 +78,8 @@ __global__ void normalize_u8_k(const uint8_t* __restrict__ raw,
   long stride = (long)gridDim.x * blockDim.x;
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
        t += stride) {
-    // out layout NCHW: t = ((b*C + c)*H + h)*W + w
-    long w = t % W;
-    long h = (t / W) % H;
-    long c = (t / ((long)W * H)) % C;
-    long b = t / ((long)W * H * C);
-    uint8_t v = raw[((b * H + h) * W + w) * C + c];  // HWC raw
-    out[t] = (v * (1.0f / 255.0f) - mean[c]) / stdv[c];
+    int c = t % C;
+    out[t] = (raw[t] * (1.0f / 255.0f) - mean[c]) / stdv[c];
   }
 }
 
