@@ -481,7 +481,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   torch::Tensor db;
   if (has_b) {
     db = torch::empty({Kout}, dy.options());
-    auto parts = torch::empty({(long)Kout * 64}, dy.options());
+    auto parts = torch::empty({(long)Kout * 4096}, dy.options());
     launch_conv_db(dy.data_ptr<float>(), db.data_ptr<float>(),
                    parts.data_ptr<float>(), Nb, Kout, OH * OW, st);
   } else {

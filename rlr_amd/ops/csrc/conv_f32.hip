@@ -527,17 +527,15 @@ __global__ void dwperm_rsc_crs_k(const float* __restrict__ in,
 }
 
 // ---- conv bias gradient (NHWC): db[ko] = column sum of dy [M][KO] ----
-constexpr int kDbChunks = 64;
-
+// Dynamic chunk count keeps >=64k threads busy regardless of Kout; stage 2
+// is one wave per ko over the chunk partials (shuffle tree, deterministic).
 __global__ void conv_db_stage1_k(const float* __restrict__ dy,
                                  float* __restrict__ partials, long M,
-                                 int Kout) {
-  // grid (chunk, ko-group): consecutive threads cover consecutive ko
-  // (coalesced); each walks its row chunk.
+                                 int Kout, int chunks) {
   int chunk = blockIdx.x;
   int ko = blockIdx.y * blockDim.x + threadIdx.x;
   if (ko >= Kout) return;
-  long per = (M + kDbChunks - 1) / kDbChunks;
+  long per = (M + chunks - 1) / chunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float acc = 0.f;
   for (long m = lo; m < hi; ++m) acc += dy[m * Kout + ko];
@@ -545,12 +543,17 @@ __global__ void conv_db_stage1_k(const float* __restrict__ dy,
 }
 
 __global__ void conv_db_stage2_k(const float* __restrict__ partials,
-                                 float* __restrict__ db, int Kout) {
-  int ko = blockIdx.x * blockDim.x + threadIdx.x;
+                                 float* __restrict__ db, int Kout,
+                                 int chunks) {
+  int ko = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
   if (ko >= Kout) return;
   float acc = 0.f;
-  for (int c = 0; c < kDbChunks; ++c) acc += partials[(long)c * Kout + ko];
-  db[ko] = acc;
+  for (int c = lane; c < chunks; c += kWave)
+    acc += partials[(long)c * Kout + ko];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) db[ko] = acc;
 }
 
 // permute w (Kout,C,R,S) -> staged layouts
@@ -678,14 +681,25 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
       rsc, dw, Kout, C, R * S);
 }
 
+int conv_db_chunks(long M, int Kout) {
+  long want = 131072 / (Kout < 1 ? 1 : Kout);   // ~128k threads
+  long cap = (M + 63) / 64;                     // >= 64 rows per chunk
+  long c = want < cap ? want : cap;
+  if (c < 64) c = 64;
+  if (c > 4096) c = 4096;
+  return (int)c;
+}
+
 void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
                     int Kout, int OHW, void* s) {
   hipStream_t st = (hipStream_t)s;
   long M = (long)Nb * OHW;
-  dim3 g1(kDbChunks, (Kout + kBlock - 1) / kBlock);
-  conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout);
-  conv_db_stage2_k<<<(Kout + kBlock - 1) / kBlock, kBlock, 0, st>>>(
-      partials, db, Kout);
+  int chunks = conv_db_chunks(M, Kout);
+  dim3 g1(chunks, (Kout + kBlock - 1) / kBlock);
+  conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout, chunks);
+  int wpb = kBlock / kWave;
+  conv_db_stage2_k<<<(Kout + wpb - 1) / wpb, kBlock, 0, st>>>(partials, db,
+                                                              Kout, chunks);
 }
 
 void launch_wperm_crs_ko(const float* w, float* out, int Kout, int C, int RS,
