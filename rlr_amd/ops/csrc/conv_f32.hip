@@ -510,6 +510,127 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     }
 }
 
+// ------------------------- small-C direct kernels (first layers) -------
+// For C*R*S <= 32 the MFMA tile wastes most of its K (conv1: Kdim 9 padded
+// to 32) and most of its N; a direct kernel at the VALU/memory roofline is
+// several times faster.  Thread per output pixel m: all Kout accumulators
+// in registers, weights staged once in LDS, NHWC float4 stores.
+
+template <int KO_T>
+__global__ __launch_bounds__(256)
+void conv_small_fwd_k(const float* __restrict__ x,
+                      const float* __restrict__ wt,  // [(r,s,c)][KO]
+                      const float* __restrict__ bias, float* __restrict__ y,
+                      ConvShape sh, int Kdim, int relu) {
+  __shared__ float w_lds[32 * KO_T];
+  for (int i = threadIdx.x; i < Kdim * sh.Kout; i += blockDim.x)
+    w_lds[i] = wt[i];
+  __syncthreads();
+  long M = (long)sh.Nb * sh.OH * sh.OW;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long m = (long)blockIdx.x * blockDim.x + threadIdx.x; m < M;
+       m += stride) {
+    int ow = m % sh.OW;
+    int oh = (m / sh.OW) % sh.OH;
+    long nb = m / ((long)sh.OW * sh.OH);
+    int oh0 = oh * sh.stride - sh.pad, ow0 = ow * sh.stride - sh.pad;
+    const float* xp = x + nb * (long)sh.H * sh.W * sh.C;
+    float acc[KO_T];
+#pragma unroll
+    for (int k = 0; k < KO_T; ++k) acc[k] = bias ? bias[k] : 0.f;
+    for (int rs = 0; rs < Kdim / sh.C; ++rs) {
+      int r = rs / sh.S, s = rs % sh.S;
+      int ih = oh0 + r, iw = ow0 + s;
+      if (ih < 0 || ih >= sh.H || iw < 0 || iw >= sh.W) continue;
+      const float* xrow = xp + ((long)ih * sh.W + iw) * sh.C;
+      for (int c = 0; c < sh.C; ++c) {
+        float v = xrow[c];
+        const float* wrow = &w_lds[(rs * sh.C + c) * KO_T];
+#pragma unroll
+        for (int k = 0; k < KO_T; ++k) acc[k] = fmaf(v, wrow[k], acc[k]);
+      }
+    }
+    float* yo = y + m * sh.Kout;
+#pragma unroll
+    for (int k = 0; k < KO_T; ++k) {
+      float v = relu ? fmaxf(acc[k], 0.f) : acc[k];
+      yo[k] = v;
+    }
+  }
+}
+
+// dw partials for small Ncrs: block handles an m-chunk; per-thread (ko,crs)
+// pairs accumulate from LDS-staged dy/x tiles; fixed-order combine follows.
+template <int MC>
+__global__ __launch_bounds__(256)
+void conv_small_bwdw_k(const float* __restrict__ dy,
+                       const float* __restrict__ x,
+                       float* __restrict__ partials,  // [chunk][KO*Ncrs]
+                       ConvShape sh, int Ncrs, long k_per_chunk) {
+  __shared__ float dy_lds[MC * 64];   // [m][ko] (KO <= 64)
+  __shared__ float xp_lds[MC * 32];   // [m][crs] (Ncrs <= 32)
+  const long Kdim = (long)sh.Nb * sh.OH * sh.OW;
+  const long k_lo = (long)blockIdx.x * k_per_chunk;
+  const long k_hi = min(Kdim, k_lo + k_per_chunk);
+  const int KO = sh.Kout;
+  const int pairs = KO * Ncrs;
+  float acc[8];  // up to 8 pairs per thread (KO*Ncrs <= 2048)
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = 0.f;
+
+  for (long m0 = k_lo; m0 < k_hi; m0 += MC) {
+    int mc = (int)min((long)MC, k_hi - m0);
+    // stage dy[m][ko]
+    for (int i = threadIdx.x; i < mc * KO; i += blockDim.x)
+      dy_lds[i] = dy[(m0 + i / KO) * KO + (i % KO)];
+    // stage x patches [m][crs]
+    for (int i = threadIdx.x; i < mc * Ncrs; i += blockDim.x) {
+      int m = i / Ncrs, crs = i % Ncrs;
+      long gm = m0 + m;
+      int ow = gm % sh.OW;
+      int oh = (gm / sh.OW) % sh.OH;
+      long nb = gm / ((long)sh.OW * sh.OH);
+      int c = crs % sh.C;
+      int rs = crs / sh.C;
+      int r = rs / sh.S, s = rs % sh.S;
+      int ih = oh * sh.stride - sh.pad + r;
+      int iw = ow * sh.stride - sh.pad + s;
+      float v = 0.f;
+      if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
+        v = x[(nb * (long)sh.H * sh.W + (long)ih * sh.W + iw) * sh.C + c];
+      xp_lds[i] = v;
+    }
+    __syncthreads();
+    for (int p = threadIdx.x, pi = 0; p < pairs; p += blockDim.x, ++pi) {
+      int ko = p / Ncrs, crs = p % Ncrs;
+      float a = acc[pi];
+      for (int m = 0; m < mc; ++m)
+        a = fmaf(dy_lds[m * KO + ko], xp_lds[m * Ncrs + crs], a);
+      acc[pi] = a;
+    }
+    __syncthreads();
+  }
+  for (int p = threadIdx.x, pi = 0; p < pairs; p += blockDim.x, ++pi)
+    partials[(long)blockIdx.x * pairs + p] = acc[pi];
+}
+
+// combine partials (fixed order) + permute (r,s,c)->(c,r,s)
+__global__ void conv_small_bwdw_reduce_k(const float* __restrict__ partials,
+                                         float* __restrict__ dw, int KO,
+                                         int C, int RS, int chunks) {
+  int pairs = KO * C * RS;
+  int p = blockIdx.x * blockDim.x + threadIdx.x;
+  if (p >= pairs) return;
+  float acc = 0.f;
+  for (int ch = 0; ch < chunks; ++ch)
+    acc += partials[(long)ch * pairs + p];
+  int crs = p % (C * RS);   // (r,s,c)
+  int ko = p / (C * RS);
+  int c = crs % C;
+  int rs = crs / C;
+  dw[(ko * C + c) * RS + rs] = acc;
+}
+
 // dw column permute: [(ko)][(r,s,c)] -> torch layout [(ko)][(c,r,s)]
 __global__ void dwperm_rsc_crs_k(const float* __restrict__ in,
                                  float* __restrict__ out, int Kout, int C,
@@ -601,6 +722,15 @@ void launch_conv_fwd(const float* x, const float* wt, const float* bias,
   long M = (long)Nb * OH * OW;
   dim3 grid((M + 127) / 128, (Kout + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
+  if (Kdim <= 32 && (Kout == 32 || Kout == 64)) {
+    if (Kout == 32)
+      conv_small_fwd_k<32><<<grid_for(M), 256, 0, st>>>(x, wt, bias, y, sh,
+                                                        Kdim, relu);
+    else
+      conv_small_fwd_k<64><<<grid_for(M), 256, 0, st>>>(x, wt, bias, y, sh,
+                                                        Kdim, relu);
+    return;
+  }
   bool v4 = (C % 32) == 0;
   if (pad == 0 && v4)
     conv_fwd_k<true, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
@@ -659,6 +789,15 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
       SK == 1 ? Kdim : (((Kdim + SK - 1) / SK + BK - 1) / BK) * BK;
   dim3 grid((Kout + 63) / 64, (Ncrs + BN - 1) / BN, SK);
   hipStream_t st = (hipStream_t)s;
+  if (Ncrs <= 32 && Kout <= 64) {
+    int chunks = 512;
+    long kpc = ((Kdim + chunks - 1) / chunks + 31) / 32 * 32;
+    conv_small_bwdw_k<32><<<chunks, 256, 0, st>>>(dy, x, ws, sh, Ncrs, kpc);
+    int pairs = Kout * Ncrs;
+    conv_small_bwdw_reduce_k<<<(pairs + 255) / 256, 256, 0, st>>>(
+        ws, dw, Kout, C, R * S, chunks);
+    return;
+  }
   float* slabs = ws;                         // SK * Kout * Ncrs
   float* rsc = ws + (long)SK * Kout * Ncrs;  // Kout * Ncrs, rsc order
   float* target = SK == 1 ? rsc : slabs;
