@@ -84,10 +84,13 @@ __global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
   long n_out = B * OH * (long)OW * C;
   long stride = (long)gridDim.x * blockDim.x;
   int WC = W * C;
+  // C is a power of two in every model here: shift/mask decomposition
+  int csh = 31 - __clz(C);
+  bool cp2 = (C & (C - 1)) == 0;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
        i += stride) {
-    int c = i % C;
-    long rest = i / C;
+    int c = cp2 ? (int)(i & (C - 1)) : (int)(i % C);
+    long rest = cp2 ? (i >> csh) : (i / C);
     int ow = rest % OW;
     int oh = (rest / OW) % OH;
     long b = rest / ((long)OW * OH);
@@ -110,10 +113,12 @@ __global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
                                  int C) {
   long n_in = B * H * (long)W * C;
   long stride = (long)gridDim.x * blockDim.x;
+  int csh = 31 - __clz(C);
+  bool cp2 = (C & (C - 1)) == 0;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_in;
        i += stride) {
-    int c = i % C;
-    long rest = i / C;
+    int c = cp2 ? (int)(i & (C - 1)) : (int)(i % C);
+    long rest = cp2 ? (i >> csh) : (i / C);
     int iw = rest % W;
     int ih = (rest / W) % H;
     long b = rest / ((long)W * H);

@@ -298,7 +298,9 @@ void launch_gemm_f32(const float* A, const float* B, float* C,
 void launch_splitk_reduce(const float* ws, float* C, const float* bias,
                           int M, int N, int ldc, int SK, int relu, void* s) {
   long n_out = (long)M * N;
-  if (n_out <= 65536 && SK >= 16) {
+  // wave path only when outputs are few (its z-strided lane loads are
+  // uncoalesced; for many outputs the thread-per-output ILP loop wins)
+  if (n_out <= 8192 && SK >= 16) {
     int wpb = kBlock / kWave;
     splitk_reduce_wave_k<<<(n_out + wpb - 1) / wpb, kBlock, 0,
                            (hipStream_t)s>>>(ws, C, bias, M, N, ldc, SK,
