@@ -614,21 +614,26 @@ void conv_small_bwdw_k(const float* __restrict__ dy,
     partials[(long)blockIdx.x * pairs + p] = acc[pi];
 }
 
-// combine partials (fixed order) + permute (r,s,c)->(c,r,s)
+// combine partials (fixed order, wave per pair) + permute (r,s,c)->(c,r,s)
 __global__ void conv_small_bwdw_reduce_k(const float* __restrict__ partials,
                                          float* __restrict__ dw, int KO,
                                          int C, int RS, int chunks) {
   int pairs = KO * C * RS;
-  int p = blockIdx.x * blockDim.x + threadIdx.x;
+  int p = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
+  int lane = threadIdx.x % kWave;
   if (p >= pairs) return;
   float acc = 0.f;
-  for (int ch = 0; ch < chunks; ++ch)
+  for (int ch = lane; ch < chunks; ch += kWave)
     acc += partials[(long)ch * pairs + p];
-  int crs = p % (C * RS);   // (r,s,c)
-  int ko = p / (C * RS);
-  int c = crs % C;
-  int rs = crs / C;
-  dw[(ko * C + c) * RS + rs] = acc;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) {
+    int crs = p % (C * RS);   // (r,s,c)
+    int ko = p / (C * RS);
+    int c = crs % C;
+    int rs = crs / C;
+    dw[(ko * C + c) * RS + rs] = acc;
+  }
 }
 
 // dw column permute: [(ko)][(r,s,c)] -> torch layout [(ko)][(c,r,s)]
@@ -791,10 +796,12 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
   hipStream_t st = (hipStream_t)s;
   if (Ncrs <= 32 && Kout <= 64) {
     int chunks = 512;
-    long kpc = ((Kdim + chunks - 1) / chunks + 31) / 32 * 32;
-    conv_small_bwdw_k<32><<<chunks, 256, 0, st>>>(dy, x, ws, sh, Ncrs, kpc);
+    long kpc = ((Kdim + chunks - 1) / chunks + 127) / 128 * 128;
+    conv_small_bwdw_k<128><<<chunks, 256, 0, st>>>(dy, x, ws, sh, Ncrs,
+                                                   kpc);
     int pairs = Kout * Ncrs;
-    conv_small_bwdw_reduce_k<<<(pairs + 255) / 256, 256, 0, st>>>(
+    int wpb = kBlock / kWave;
+    conv_small_bwdw_reduce_k<<<(pairs + wpb - 1) / wpb, kBlock, 0, st>>>(
         ws, dw, Kout, C, R * S, chunks);
     return;
   }
@@ -821,7 +828,7 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
 }
 
 int conv_db_chunks(long M, int Kout) {
-  long want = 131072 / (Kout < 1 ? 1 : Kout);   // ~128k threads
+  long want = 32768 / (Kout < 1 ? 1 : Kout);    // ~32k threads
   long cap = (M + 63) / 64;                     // >= 64 rows per chunk
   long c = want < cap ? want : cap;
   if (c < 64) c = 64;
