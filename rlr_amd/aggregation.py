@@ -152,7 +152,9 @@ class Aggregation:
     def aggregate_buffers(self, global_model, buffer_deltas, agent_ids):
         """FedAvg-BN: data-weighted mean of BatchNorm running-stat deltas
         (build extension — the reference has no BN)."""
-        if global_model.n_buffers == 0 or not buffer_deltas:
+        if global_model.n_buffers == 0 or buffer_deltas is None:
+            return
+        if isinstance(buffer_deltas, (list, tuple)) and not buffer_deltas:
             return
         w = self._weights(agent_ids, buffer_deltas.device
                           if isinstance(buffer_deltas, torch.Tensor)

@@ -27,7 +27,8 @@ def _args(**over):
 def test_cifar_cnn_fl_rounds():
     from rlr_amd.federated import run
     h = run(_args(data='cifar10', num_corrupt=1, poison_frac=0.5,
-                  robustLR_threshold=2, pattern_type='plus'))
+                  robustLR_threshold=2, pattern_type='plus',
+                  rounds=4, snap=4, local_ep=2))
     assert torch.isfinite(h['final_params']).all()
     assert h['val_acc'][-1] > 0.2  # learning something
 
