@@ -33,7 +33,7 @@ struct ConvShape {
 
 // Fragment compute over one staged K-tile; wave grid 2x2, each wave owns
 // an (MI*16 x NI*16) subtile.
-template <int MI, int NI>
+template <int MI, int NI, int LDA = LDA_S, int LDB = LDB_S>
 __device__ __forceinline__ void mfma_tile(const float* __restrict__ Abuf,
                                           const float* __restrict__ Bbuf,
                                           int wr, int wc, int l15, int l4,
@@ -44,11 +44,11 @@ __device__ __forceinline__ void mfma_tile(const float* __restrict__ Abuf,
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
       a_frag[mi] =
-          Abuf[(wr * MI * 16 + mi * 16 + l15) * LDA_S + kk * 4 + l4];
+          Abuf[(wr * MI * 16 + mi * 16 + l15) * LDA + kk * 4 + l4];
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni)
       b_frag[ni] =
-          Bbuf[(kk * 4 + l4) * LDB_S + wc * NI * 16 + ni * 16 + l15];
+          Bbuf[(kk * 4 + l4) * LDB + wc * NI * 16 + ni * 16 + l15];
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -351,6 +351,151 @@ void conv_bwd_data_k(const float* __restrict__ dy,
     }
 }
 
+// bwd-data, BN=32 (C <= 32): wave grid 4x1 (MI=2, NI=2 over 32 cols) —
+// halves the MFMA work vs padding C=32 into a 64-wide tile.
+constexpr int LDB32_S = 48;  // 32 + 16: consecutive k rows 16 banks apart
+
+template <int ST, bool V4>
+__global__ __launch_bounds__(256)
+void conv_bwd_data32_k(const float* __restrict__ dy,
+                       const float* __restrict__ wp,  // [(r,s,ko)][C]
+                       float* __restrict__ dx, ConvShape sh, int Kdim) {
+  constexpr int BM = 128, MI = 2, NI = 2, BN32 = 32;
+  __shared__ float A_lds[2][BM * LDA_S];
+  __shared__ float B_lds[2][BK * LDB32_S];
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave, wc = 0;  // 4x1 wave grid
+  const int l15 = lane & 15, l4 = lane >> 4;
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int m_blk = blockIdx.x * BM;
+  const int n_blk = blockIdx.y * BN32;
+  const long M = (long)sh.Nb * sh.H * sh.W;
+  const int am = t >> 3, ak = (t & 7) * 4;
+  const int bk = t >> 3, bn = (t & 7) * 4;  // 32x32 tile, 1 round
+  const int stride = ST > 0 ? ST : sh.stride;
+
+  int iwp[BM / 32], ihp[BM / 32];
+  long base[BM / 32];
+  bool mval[BM / 32];
+#pragma unroll
+  for (int j = 0; j < BM / 32; ++j) {
+    long gm = m_blk + am + j * 32;
+    mval[j] = gm < M;
+    long gmc = mval[j] ? gm : 0;
+    iwp[j] = (int)(gmc % sh.W) + sh.pad;
+    ihp[j] = (int)((gmc / sh.W) % sh.H) + sh.pad;
+    base[j] = (gmc / ((long)sh.W * sh.H)) * (long)sh.OH * sh.OW * sh.Kout;
+  }
+
+  float ra[BM / 32][4];
+  float4 rb;
+  auto stage_load = [&](int k0) {
+    if (V4) {
+      int rs = k0 / sh.Kout;
+      int r = rs / sh.S, s = rs % sh.S;
+      int ko0 = k0 - rs * sh.Kout + ak;
+#pragma unroll
+      for (int j = 0; j < BM / 32; ++j) {
+        int ohn = ihp[j] - r, own = iwp[j] - s;
+        float4 q = {0.f, 0.f, 0.f, 0.f};
+        if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+            own % stride == 0) {
+          int oh = ohn / stride, ow = own / stride;
+          if (oh < sh.OH && ow < sh.OW)
+            q = *(const float4*)(dy + base[j] +
+                                 ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+        }
+        ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < BM / 32; ++j) {
+        ra[j][0] = ra[j][1] = ra[j][2] = ra[j][3] = 0.f;
+        if (mval[j]) {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            int k = k0 + ak + e;
+            if (k < Kdim) {
+              int ko = k % sh.Kout;
+              int rs = k / sh.Kout;
+              int r = rs / sh.S, s = rs % sh.S;
+              int ohn = ihp[j] - r, own = iwp[j] - s;
+              if (ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+                  own % stride == 0) {
+                int oh = ohn / stride, ow = own / stride;
+                if (oh < sh.OH && ow < sh.OW)
+                  ra[j][e] = dy[base[j] +
+                                ((long)oh * sh.OW + ow) * sh.Kout + ko];
+              }
+            }
+          }
+        }
+      }
+    }
+    {
+      int gk = k0 + bk;
+      float4 q = {0.f, 0.f, 0.f, 0.f};
+      if (gk < Kdim) {
+        const float* src = wp + (long)gk * sh.C + n_blk + bn;
+        if (n_blk + bn + 3 < sh.C && (sh.C % 4) == 0)
+          q = *(const float4*)src;
+        else {
+          if (n_blk + bn + 0 < sh.C) q.x = src[0];
+          if (n_blk + bn + 1 < sh.C) q.y = src[1];
+          if (n_blk + bn + 2 < sh.C) q.z = src[2];
+          if (n_blk + bn + 3 < sh.C) q.w = src[3];
+        }
+      }
+      rb = q;
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < BM / 32; ++j) {
+      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+      ((float2*)dst)[0] = {ra[j][0], ra[j][1]};
+      ((float2*)dst)[1] = {ra[j][2], ra[j][3]};
+    }
+    *(float4*)&B_lds[buf][bk * LDB32_S + bn] = rb;
+  };
+
+  stage_load(0);
+  stage_write(0);
+  if (BK < Kdim) stage_load(BK);
+  __syncthreads();
+  int buf = 0;
+  for (int k0 = 0; k0 < Kdim; k0 += BK) {
+    if (k0 + BK < Kdim) {
+      stage_write(buf ^ 1);
+      if (k0 + 2 * BK < Kdim) stage_load(k0 + 2 * BK);
+    }
+    mfma_tile<MI, NI, LDA_S, LDB32_S>(A_lds[buf], B_lds[buf], wr, wc, l15,
+                                      l4, acc);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int c = n_blk + ni * 16 + l15;
+      if (c >= sh.C) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long m = m_blk + wr * 32 + mi * 16 + l4 * 4 + r;
+        if (m >= M) continue;
+        dx[m * sh.C + c] = acc[mi][ni][r];
+      }
+    }
+}
+
 // ------------------------------------------------------------ bwd-weight
 
 // 64x64 tile over (ko, crs=(r,s,c)); Kdim = m = (nb,oh,ow); split-K.
@@ -364,7 +509,9 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
                        ConvShape sh, int Ncrs, long k_per_chunk,
                        int direct_out) {
   constexpr int BM = 64, MI = 2, NI = 2;
-  __shared__ float A_lds[2][BM * LDA_S];
+  constexpr int LDAW = 33;  // odd stride: transposed scalar writes 2-way
+                            // instead of 4-way bank conflicted
+  __shared__ float A_lds[2][BM * LDAW];
   __shared__ float B_lds[2][BK * LDB_S];
   const int t = threadIdx.x;
   const int wave = t >> 6, lane = t & 63;
@@ -468,7 +615,7 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
       int m = amr + j * 16;
 #pragma unroll
       for (int i = 0; i < 4; ++i)
-        A_lds[buf][(ako + i) * LDA_S + m] = raA[j][i];
+        A_lds[buf][(ako + i) * LDAW + m] = raA[j][i];
     }
 #pragma unroll
     for (int j = 0; j < 2; ++j)
@@ -485,7 +632,7 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
       stage_write(buf ^ 1);
       if (k0 + 2 * BK < k_hi) stage_load(k0 + 2 * BK);
     }
-    mfma_tile<MI, NI>(A_lds[buf], B_lds[buf], wr, wc, l15, l4, acc);
+    mfma_tile<MI, NI, LDAW>(A_lds[buf], B_lds[buf], wr, wc, l15, l4, acc);
     __syncthreads();
     buf ^= 1;
   }
@@ -761,6 +908,20 @@ void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
   dim3 grid((M + 127) / 128, (C + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
   bool v4 = (Kout % 32) == 0;
+  if (C <= 32) {
+    dim3 g32((M + 127) / 128, (C + 31) / 32, 1);
+    if (stride == 1 && v4)
+      conv_bwd_data32_k<1, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    else if (stride == 2 && v4)
+      conv_bwd_data32_k<2, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    else if (stride == 1)
+      conv_bwd_data32_k<1, false><<<g32, 256, 0, st>>>(dy, wp, dx, sh,
+                                                       Kdim);
+    else
+      conv_bwd_data32_k<0, false><<<g32, 256, 0, st>>>(dy, wp, dx, sh,
+                                                       Kdim);
+    return;
+  }
   if (stride == 1 && v4)
     conv_bwd_data_k<1, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
   else if (stride == 2 && v4)
