@@ -205,7 +205,7 @@ def dropout(x, p, training, rng: DropoutCtx):
     from ..utils.rng import derive_seed
     g = torch.Generator(device='cpu')
     g.manual_seed(derive_seed(rng.seed, 'dropout', rng.next_offset()))
-    mask = (torch.rand(x.shape, generator=g) >= p).to(x.dtype)
+    mask = (torch.rand(x.shape, generator=g) >= p).to(x.dtype).to(x.device)
     return x * mask / (1.0 - p)
 
 

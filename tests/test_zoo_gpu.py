@@ -26,12 +26,14 @@ def _args(**over):
 
 def test_cifar_cnn_fl_rounds():
     from rlr_amd.federated import run
+    # DBA poisoning on, defense off (RLR with K=2 agents flips every
+    # disputed coordinate and destroys learning by design — the defense
+    # semantics are covered at K=10 in test_e2e_gpu)
     h = run(_args(data='cifar10', num_corrupt=1, poison_frac=0.5,
-                  robustLR_threshold=2, pattern_type='plus',
-                  rounds=6, snap=3, local_ep=2))
+                  pattern_type='plus', rounds=6, snap=3, local_ep=2))
     assert torch.isfinite(h['final_params']).all()
-    # the deep CIFAR CNN learns the synthetic task slowly (CPU reference:
-    # ~0.24 by round 6); just require clearly-above-chance
+    # deep CIFAR CNN learns the synthetic task slowly (CPU ref: ~0.2 by
+    # round 6); require clearly-above-chance
     assert h['val_acc'][-1] > 0.14, h['val_acc']
 
 
