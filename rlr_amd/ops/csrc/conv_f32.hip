@@ -25,7 +25,10 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 constexpr int BK = 32;
 constexpr int LDA_S = BK + 2;
 constexpr int BN = 64;
-constexpr int LDB_S = BN + 16;
+// +8 pad: consecutive k rows land 8 banks apart (mild 2-way conflict on
+// half of each b32 lane group) but the 128x64x32 tiles' LDS drops to
+// 53.2 KB -> 3 blocks/CU instead of 2 (more TLP to hide staging).
+constexpr int LDB_S = BN + 8;
 
 struct ConvShape {
   int Nb, C, H, W, Kout, R, S, OH, OW, stride, pad;
