@@ -821,15 +821,19 @@ __global__ void conv_db_stage1_k(const float* __restrict__ dy,
 __global__ void conv_db_stage2_k(const float* __restrict__ partials,
                                  float* __restrict__ db, int Kout,
                                  int chunks) {
-  int ko = blockIdx.x * (blockDim.x / kWave) + threadIdx.x / kWave;
-  int lane = threadIdx.x % kWave;
-  if (ko >= Kout) return;
+  // one BLOCK per ko: 256 threads stride the chunks, LDS tree combine
+  int ko = blockIdx.x;
+  __shared__ float sh[kBlock];
   float acc = 0.f;
-  for (int c = lane; c < chunks; c += kWave)
+  for (int c = threadIdx.x; c < chunks; c += blockDim.x)
     acc += partials[(long)c * Kout + ko];
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
-  if (lane == 0) db[ko] = acc;
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = kBlock / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) db[ko] = sh[0];
 }
 
 // permute w (Kout,C,R,S) -> staged layouts
@@ -992,7 +996,7 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
 }
 
 int conv_db_chunks(long M, int Kout) {
-  long want = 32768 / (Kout < 1 ? 1 : Kout);    // ~32k threads
+  long want = 131072 / (Kout < 1 ? 1 : Kout);   // ~128k threads
   long cap = (M + 63) / 64;                     // >= 64 rows per chunk
   long c = want < cap ? want : cap;
   if (c < 64) c = 64;
@@ -1007,9 +1011,7 @@ void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
   int chunks = conv_db_chunks(M, Kout);
   dim3 g1(chunks, (Kout + kBlock - 1) / kBlock);
   conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout, chunks);
-  int wpb = kBlock / kWave;
-  conv_db_stage2_k<<<(Kout + wpb - 1) / wpb, kBlock, 0, st>>>(partials, db,
-                                                              Kout, chunks);
+  conv_db_stage2_k<<<Kout, kBlock, 0, st>>>(partials, db, Kout, chunks);
 }
 
 void launch_wperm_crs_ko(const float* w, float* out, int Kout, int C, int RS,
