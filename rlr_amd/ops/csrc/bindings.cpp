@@ -96,6 +96,12 @@ void launch_conv_bwd_weight(const float*, const float*, float*, float*, int,
 void launch_conv_db(const float*, float*, float*, int, int, int, void*);
 void launch_wperm_crs_ko(const float*, float*, int, int, int, void*);
 void launch_wperm_kors_c(const float*, float*, int, int, int, void*);
+// gemm_bf16.hip
+void launch_gemm_bf16(const unsigned short*, const unsigned short*, float*,
+                      const float*, unsigned short*, float*, int, int, int,
+                      int, int, int, int, int, void*);
+void launch_f32_to_bf16(const float*, unsigned short*, long, void*);
+void launch_bf16_to_f32(const unsigned short*, float*, long, void*);
 // batchnorm.hip
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, int, int, int, float,
@@ -392,6 +398,42 @@ torch::Tensor gemm(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// bf16 GEMM: A (M,K) bf16, B (K,N) bf16 -> C fp32 (or bf16 if out_bf16)
+torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor B,
+                        c10::optional<torch::Tensor> bias, bool relu,
+                        bool out_bf16) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              B.scalar_type() == torch::kBFloat16);
+  int M = A.size(0), K = A.size(1), N = B.size(1);
+  CHK(B.size(0) == K);
+  int SK = gemm_f32_splitk(M, N, K);
+  torch::Tensor ws;
+  float* wsp = nullptr;
+  if (SK > 1) {
+    ws = torch::empty({(long)SK * M * N}, A.options().dtype(torch::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  torch::Tensor C;
+  if (out_bf16 && SK == 1) {
+    C = torch::empty({M, N}, A.options());
+    launch_gemm_bf16((const unsigned short*)A.data_ptr(),
+                     (const unsigned short*)B.data_ptr(), nullptr,
+                     bias ? bias->data_ptr<float>() : nullptr,
+                     (unsigned short*)C.data_ptr(), wsp, M, N, K, K, N, N,
+                     SK, relu ? 1 : 0, stream_of(A));
+  } else {
+    C = torch::empty({M, N}, A.options().dtype(torch::kFloat));
+    launch_gemm_bf16((const unsigned short*)A.data_ptr(),
+                     (const unsigned short*)B.data_ptr(),
+                     C.data_ptr<float>(),
+                     bias ? bias->data_ptr<float>() : nullptr, nullptr, wsp,
+                     M, N, K, K, N, N, SK, relu ? 1 : 0, stream_of(A));
+    if (out_bf16) C = C.to(torch::kBFloat16);
+  }
+  return C;
+}
+
 torch::Tensor transpose2d(torch::Tensor A) {
   CHK_CUDA(A);
   auto out = torch::empty({A.size(1), A.size(0)}, A.options());
@@ -616,6 +658,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("apply_update", &apply_update);
   m.def("add_noise", &add_noise);
   m.def("gemm", &gemm);
+  m.def("gemm_bf16", &gemm_bf16);
   m.def("transpose2d", &transpose2d);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
