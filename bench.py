@@ -34,6 +34,10 @@ def main():
     ap.add_argument('--model', type=str, default=None,
                     help="e.g. resnet18 (BASELINE configs 3-4: CIFAR10 "
                          "ResNet18); default: the dataset's reference CNN")
+    ap.add_argument('--dtype', type=str, default='fp32',
+                    choices=['fp32', 'bf16'],
+                    help="client compute dtype; the headline FMNIST config "
+                         "is fp32 (the reference's own precision)")
     a = ap.parse_args()
 
     from rlr_amd.federated import build_world, run_round
@@ -53,7 +57,7 @@ def main():
     samples_per_agent = 6000 if a.data == 'fmnist' else 1250
     D.DEFAULT_SIZES[a.data] = (samples_per_agent * num_agents, 10000)
     args = default_args(
-        data=a.data, model=a.model, num_agents=num_agents,
+        data=a.data, model=a.model, dtype=a.dtype, num_agents=num_agents,
         num_corrupt=1 * n, poison_frac=0.5, robustLR_threshold=4,
         aggr='avg', local_ep=2, bs=256, agent_frac=1.0, synthetic=True,
         no_tb=True, snap=10 ** 9, device=device)
@@ -95,7 +99,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no throughput numbers
-            "dtype": "fp32",
+            "dtype": a.dtype,
             "data": "synthetic",
             "config": {
                 "model": a.model or ("CNN_MNIST" if a.data != 'cifar10'

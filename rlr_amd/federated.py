@@ -78,6 +78,8 @@ def build_world(args):
     # deterministic init on every rank (no broadcast needed)
     torch.manual_seed(derive_seed(args.seed, 'init'))
     model = get_model(args.data, getattr(args, 'model', None))
+    if getattr(args, 'dtype', 'fp32') == 'bf16' and device.type == 'cuda':
+        model.set_compute_dtype(torch.bfloat16)
     global_model = FlatParamModel(model, device)
 
     aggregator = Aggregation(agent_data_sizes, global_model.n_params,
@@ -97,6 +99,8 @@ def _replica_pool(world, args, gm):
         reps = []
         for _ in range(max(1, R)):
             m = get_model(args.data, getattr(args, 'model', None))
+            if getattr(args, 'dtype', 'fp32') == 'bf16':
+                m.set_compute_dtype(torch.bfloat16)
             rep = FlatParamModel(m, gm.device)
             reps.append((rep, torch.cuda.Stream(device=gm.device)))
         world['replicas'] = reps

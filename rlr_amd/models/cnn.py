@@ -16,9 +16,19 @@ class _OpsModel(nn.Module):
     def __init__(self):
         super().__init__()
         self.rng = Fo.DropoutCtx()
+        self.compute_dtype = None  # None = fp32; torch.bfloat16 for the
+                                   # bf16 MFMA path (--dtype bf16, GPU only)
 
     def set_dropout_seed(self, seed: int):
         self.rng.reset(seed)
+
+    def set_compute_dtype(self, dtype):
+        self.compute_dtype = dtype
+
+    def _cast_in(self, x):
+        if self.compute_dtype is not None and x.dtype != self.compute_dtype:
+            x = x.to(self.compute_dtype)
+        return x
 
 
 class CNN_MNIST(_OpsModel):
@@ -35,6 +45,7 @@ class CNN_MNIST(_OpsModel):
         self.p_drop = 0.5
 
     def forward(self, x):
+        x = self._cast_in(x)
         x = Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True)
         x = Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True)
         x = Fo.max_pool2d_2x2(x)
@@ -62,6 +73,7 @@ class CNN_CIFAR(_OpsModel):
         self.p_drop = 0.5
 
     def forward(self, x):
+        x = self._cast_in(x)
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True))
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True))
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv3.weight, self.conv3.bias, relu=True))

@@ -71,6 +71,7 @@ class ResNet18(_OpsModel):
         self.fc = nn.Linear(512, num_classes)
 
     def forward(self, x):
+        x = self._cast_in(x)
         x = Fo.conv2d(x, self.conv1.weight, None, 1, 1)
         x = Fo.relu(self.bn1(x))
         x = self.layers(x)

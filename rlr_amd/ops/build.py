@@ -22,7 +22,8 @@ ARCH = os.environ.get('PYTORCH_ROCM_ARCH', 'gfx950')
 
 KERNEL_SOURCES = [
     'elementwise.hip', 'flatopt.hip', 'aggregation.hip', 'gemm_f32.hip',
-    'gemm_bf16.hip', 'conv_f32.hip', 'batchnorm.hip', 'poison.hip',
+    'gemm_bf16.hip', 'conv_f32.hip', 'conv_bf16.hip', 'batchnorm.hip',
+    'poison.hip',
 ]
 BINDINGS = 'bindings.cpp'
 

@@ -9,7 +9,8 @@
 constexpr int kBnChunks = 64;
 
 // ---- stage 1: partials[chunk][2C] = (sum x, sum x^2) over a row chunk ----
-__global__ void bn_stats1_k(const float* __restrict__ x,
+template <typename T>
+__global__ void bn_stats1_k(const T* __restrict__ x,
                             float* __restrict__ partials, long M, int C) {
   int chunk = blockIdx.x;
   int c = blockIdx.y * blockDim.x + threadIdx.x;
@@ -18,7 +19,7 @@ __global__ void bn_stats1_k(const float* __restrict__ x,
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float s = 0.f, ss = 0.f;
   for (long m = lo; m < hi; ++m) {
-    float v = x[m * C + c];
+    float v = ldv(&x[m * C + c]);
     s += v;
     ss += v * v;
   }
@@ -73,30 +74,32 @@ __global__ void bn_eval_stats_k(const float* __restrict__ running_mean,
 
 // ---- normalize: y = w[c]*(x-mean[c])*rstd[c] + b[c]; stride % C == 0 so
 // each thread's channel is fixed across its grid-stride loop ----
-__global__ void bn_norm_k(const float* __restrict__ x,
+template <typename T>
+__global__ void bn_norm_k(const T* __restrict__ x,
                           const float* __restrict__ w,
                           const float* __restrict__ b,
                           const float* __restrict__ mean,
                           const float* __restrict__ rstd,
-                          float* __restrict__ y, long n, int C) {
+                          T* __restrict__ y, long n, int C) {
   long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   if (stride % C == 0) {  // launcher arranges this for power-of-two C
     int c = (int)(i0 % C);
     float wc = w[c] * rstd[c];
     float bc = b[c] - mean[c] * wc;
-    for (long i = i0; i < n; i += stride) y[i] = x[i] * wc + bc;
+    for (long i = i0; i < n; i += stride) stv(&y[i], ldv(&x[i]) * wc + bc);
   } else {
     for (long i = i0; i < n; i += stride) {
       int c = (int)(i % C);
-      y[i] = w[c] * rstd[c] * (x[i] - mean[c]) + b[c];
+      stv(&y[i], w[c] * rstd[c] * (ldv(&x[i]) - mean[c]) + b[c]);
     }
   }
 }
 
 // ---- bwd stage 1: per-channel sum(dy), sum(dy * xhat) ----
-__global__ void bn_bwd_stats1_k(const float* __restrict__ x,
-                                const float* __restrict__ dy,
+template <typename T>
+__global__ void bn_bwd_stats1_k(const T* __restrict__ x,
+                                const T* __restrict__ dy,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ rstd,
                                 float* __restrict__ partials, long M,
@@ -109,9 +112,9 @@ __global__ void bn_bwd_stats1_k(const float* __restrict__ x,
   float m_ = mean[c], rs = rstd[c];
   float sdy = 0.f, sdyx = 0.f;
   for (long m = lo; m < hi; ++m) {
-    float g = dy[m * C + c];
+    float g = ldv(&dy[m * C + c]);
     sdy += g;
-    sdyx += g * (x[m * C + c] - m_) * rs;
+    sdyx += g * (ldv(&x[m * C + c]) - m_) * rs;
   }
   partials[((long)chunk * 2) * C + c] = sdy;
   partials[((long)chunk * 2 + 1) * C + c] = sdyx;
@@ -120,13 +123,14 @@ __global__ void bn_bwd_stats1_k(const float* __restrict__ x,
 // (bn_stats2_k combines these too: same [chunk][2C] layout)
 
 // ---- bwd: dx = w*rstd*(dy - (db + xhat*dwdot)/Nc) ----
-__global__ void bn_bwd_dx_k(const float* __restrict__ x,
-                            const float* __restrict__ dy,
+template <typename T>
+__global__ void bn_bwd_dx_k(const T* __restrict__ x,
+                            const T* __restrict__ dy,
                             const float* __restrict__ w,
                             const float* __restrict__ mean,
                             const float* __restrict__ rstd,
                             const float* __restrict__ stats,
-                            float* __restrict__ dx, long n, int C,
+                            T* __restrict__ dx, long n, int C,
                             float inv_count) {
   long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
@@ -136,15 +140,16 @@ __global__ void bn_bwd_dx_k(const float* __restrict__ x,
     float m_ = mean[c], rs = rstd[c];
     float db = stats[c], dwdot = stats[C + c];
     for (long i = i0; i < n; i += stride) {
-      float xhat = (x[i] - m_) * rs;
-      dx[i] = wr * (dy[i] - inv_count * (db + xhat * dwdot));
+      float xhat = (ldv(&x[i]) - m_) * rs;
+      stv(&dx[i], wr * (ldv(&dy[i]) - inv_count * (db + xhat * dwdot)));
     }
   } else {
     for (long i = i0; i < n; i += stride) {
       int c = (int)(i % C);
-      float xhat = (x[i] - mean[c]) * rstd[c];
-      dx[i] = w[c] * rstd[c] *
-              (dy[i] - inv_count * (stats[c] + xhat * stats[C + c]));
+      float xhat = (ldv(&x[i]) - mean[c]) * rstd[c];
+      stv(&dx[i], w[c] * rstd[c] *
+                      (ldv(&dy[i]) -
+                       inv_count * (stats[c] + xhat * stats[C + c])));
     }
   }
 }
@@ -162,20 +167,19 @@ static int bn_grid(long n, int C) {
   return (int)(want > 0 ? want : 1);
 }
 
-extern "C" {
-void launch_bn_fwd(const float* x, const float* w, const float* b,
-                   float* running_mean, float* running_var, float* save_mean,
-                   float* save_rstd, float* y, float* scratch, int Nb,
-                   int C, int HW, float momentum, float eps, int training,
-                   void* s) {
-  hipStream_t st = (hipStream_t)s;
+template <typename T>
+static void bn_fwd_impl(const T* x, const float* w, const float* b,
+                        float* running_mean, float* running_var,
+                        float* save_mean, float* save_rstd, T* y,
+                        float* scratch, int Nb, int C, int HW,
+                        float momentum, float eps, int training,
+                        hipStream_t st) {
   long M = (long)Nb * HW;
-  // scratch: kBnChunks*2*C partials + 2*C sums
   float* partials = scratch;
   float* sums = scratch + (long)kBnChunks * 2 * C;
   if (training) {
     dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
-    bn_stats1_k<<<g1, kBlock, 0, st>>>(x, partials, M, C);
+    bn_stats1_k<T><<<g1, kBlock, 0, st>>>(x, partials, M, C);
     bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials,
                                                               sums, C);
     bn_finalize_k<<<(C + 255) / 256, 256, 0, st>>>(
@@ -186,30 +190,66 @@ void launch_bn_fwd(const float* x, const float* w, const float* b,
         running_mean, running_var, save_mean, save_rstd, C, eps);
   }
   long n = M * C;
-  bn_norm_k<<<bn_grid(n, C), kBlock, 0, st>>>(x, w, b, save_mean, save_rstd,
-                                              y, n, C);
+  bn_norm_k<T><<<bn_grid(n, C), kBlock, 0, st>>>(x, w, b, save_mean,
+                                                 save_rstd, y, n, C);
 }
 
-void launch_bn_bwd(const float* x, const float* dy, const float* w,
-                   const float* save_mean, const float* save_rstd,
-                   float* scratch, float* dx, float* dw, float* db, int Nb,
-                   int C, int HW, int training, void* s) {
-  hipStream_t st = (hipStream_t)s;
+template <typename T>
+static void bn_bwd_impl(const T* x, const T* dy, const float* w,
+                        const float* save_mean, const float* save_rstd,
+                        float* scratch, T* dx, float* dw, float* db, int Nb,
+                        int C, int HW, int training, hipStream_t st) {
   long M = (long)Nb * HW;
   float* partials = scratch;
   float* stats = scratch + (long)kBnChunks * 2 * C;
   dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
-  bn_bwd_stats1_k<<<g1, kBlock, 0, st>>>(x, dy, save_mean, save_rstd,
-                                         partials, M, C);
+  bn_bwd_stats1_k<T><<<g1, kBlock, 0, st>>>(x, dy, save_mean, save_rstd,
+                                            partials, M, C);
   bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials, stats,
                                                             C);
   long n = M * C;
-  bn_bwd_dx_k<<<bn_grid(n, C), kBlock, 0, st>>>(
+  bn_bwd_dx_k<T><<<bn_grid(n, C), kBlock, 0, st>>>(
       x, dy, w, save_mean, save_rstd, stats, dx, n, C,
       training ? 1.f / (float)M : 0.f);
   HIP_CHECK(hipMemcpyAsync(db, stats, C * sizeof(float),
                            hipMemcpyDeviceToDevice, st));
   HIP_CHECK(hipMemcpyAsync(dw, stats + C, C * sizeof(float),
                            hipMemcpyDeviceToDevice, st));
+}
+
+extern "C" {
+void launch_bn_fwd(const float* x, const float* w, const float* b,
+                   float* running_mean, float* running_var, float* save_mean,
+                   float* save_rstd, float* y, float* scratch, int Nb,
+                   int C, int HW, float momentum, float eps, int training,
+                   void* s) {
+  bn_fwd_impl<float>(x, w, b, running_mean, running_var, save_mean,
+                     save_rstd, y, scratch, Nb, C, HW, momentum, eps,
+                     training, (hipStream_t)s);
+}
+void launch_bn_fwd_bf16(const unsigned short* x, const float* w,
+                        const float* b, float* running_mean,
+                        float* running_var, float* save_mean,
+                        float* save_rstd, unsigned short* y, float* scratch,
+                        int Nb, int C, int HW, float momentum, float eps,
+                        int training, void* s) {
+  bn_fwd_impl<unsigned short>(x, w, b, running_mean, running_var, save_mean,
+                              save_rstd, y, scratch, Nb, C, HW, momentum,
+                              eps, training, (hipStream_t)s);
+}
+void launch_bn_bwd(const float* x, const float* dy, const float* w,
+                   const float* save_mean, const float* save_rstd,
+                   float* scratch, float* dx, float* dw, float* db, int Nb,
+                   int C, int HW, int training, void* s) {
+  bn_bwd_impl<float>(x, dy, w, save_mean, save_rstd, scratch, dx, dw, db,
+                     Nb, C, HW, training, (hipStream_t)s);
+}
+void launch_bn_bwd_bf16(const unsigned short* x, const unsigned short* dy,
+                        const float* w, const float* save_mean,
+                        const float* save_rstd, float* scratch,
+                        unsigned short* dx, float* dw, float* db, int Nb,
+                        int C, int HW, int training, void* s) {
+  bn_bwd_impl<unsigned short>(x, dy, w, save_mean, save_rstd, scratch, dx,
+                              dw, db, Nb, C, HW, training, (hipStream_t)s);
 }
 }

@@ -102,6 +102,53 @@ void launch_gemm_bf16(const unsigned short*, const unsigned short*, float*,
                       int, int, int, int, int, void*);
 void launch_f32_to_bf16(const float*, unsigned short*, long, void*);
 void launch_bf16_to_f32(const unsigned short*, float*, long, void*);
+// conv_bf16.hip
+void launch_conv_fwd_bf16(const unsigned short*, const unsigned short*,
+                          const float*, unsigned short*, int, int, int, int,
+                          int, int, int, int, int, int, int, int, void*);
+void launch_conv_bwd_data_bf16(const unsigned short*, const unsigned short*,
+                               unsigned short*, int, int, int, int, int,
+                               int, int, int, int, int, int, void*);
+void launch_conv_bwd_weight_bf16(const unsigned short*,
+                                 const unsigned short*, float*, float*, int,
+                                 int, int, int, int, int, int, int, int,
+                                 int, int, int, void*);
+void launch_wperm_rsc_ko_bf16(const float*, unsigned short*, int, int, int,
+                              void*);
+void launch_wperm_rsko_c_bf16(const float*, unsigned short*, int, int, int,
+                              void*);
+void launch_conv_db_bf16(const unsigned short*, float*, float*, int, int,
+                         int, void*);
+// elementwise bf16
+void launch_relu_fwd_bf16(const unsigned short*, unsigned short*, long,
+                          void*);
+void launch_relu_bwd_bf16(const unsigned short*, const unsigned short*,
+                          unsigned short*, long, void*);
+void launch_add_relu_bf16(const unsigned short*, const unsigned short*,
+                          unsigned short*, long, void*);
+void launch_dropout_fwd_dev_bf16(const unsigned short*, unsigned short*,
+                                 uint8_t*, long, float,
+                                 const unsigned long long*, int, void*);
+void launch_dropout_bwd_bf16(const unsigned short*, const uint8_t*,
+                             unsigned short*, long, float, void*);
+void launch_gap_fwd_bf16(const unsigned short*, unsigned short*, long, int,
+                         int, void*);
+void launch_gap_bwd_bf16(const unsigned short*, unsigned short*, long, int,
+                         int, void*);
+void launch_maxpool2x2_fwd_bf16(const unsigned short*, unsigned short*,
+                                uint8_t*, long, int, int, int, int, int,
+                                void*);
+void launch_maxpool2x2_bwd_bf16(const unsigned short*, const uint8_t*,
+                                unsigned short*, long, int, int, int, int,
+                                int, void*);
+// batchnorm bf16
+void launch_bn_fwd_bf16(const unsigned short*, const float*, const float*,
+                        float*, float*, float*, float*, unsigned short*,
+                        float*, int, int, int, float, float, int, void*);
+void launch_bn_bwd_bf16(const unsigned short*, const unsigned short*,
+                        const float*, const float*, const float*, float*,
+                        unsigned short*, float*, float*, int, int, int, int,
+                        void*);
 // batchnorm.hip
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, int, int, int, float,
@@ -124,11 +171,22 @@ void launch_normalize_u8(const uint8_t*, float*, long, int, int, int,
 
 namespace {
 
+static bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+
 // ------------------------------------------------------------ elementwise
 
 torch::Tensor relu_fwd(torch::Tensor x) {
   CHK_CUDA(x);
   auto y = torch::empty_like(x);
+  if (is_bf16(x)) {
+    launch_relu_fwd_bf16((const unsigned short*)x.data_ptr(),
+                         (unsigned short*)y.data_ptr(), x.numel(),
+                         stream_of(x));
+    return y;
+  }
   launch_relu_fwd(x.data_ptr<float>(), y.data_ptr<float>(), x.numel(),
                   stream_of(x));
   return y;
@@ -138,6 +196,13 @@ torch::Tensor relu_bwd(torch::Tensor y, torch::Tensor dy) {
   CHK_CUDA(y);
   dy = match_layout(y, dy);
   auto dx = torch::empty_like(y);
+  if (is_bf16(y)) {
+    launch_relu_bwd_bf16((const unsigned short*)y.data_ptr(),
+                         (const unsigned short*)dy.data_ptr(),
+                         (unsigned short*)dx.data_ptr(), y.numel(),
+                         stream_of(y));
+    return dx;
+  }
   launch_relu_bwd(y.data_ptr<float>(), dy.data_ptr<float>(),
                   dx.data_ptr<float>(), y.numel(), stream_of(y));
   return dx;
@@ -147,6 +212,13 @@ torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
   CHK_CUDA(a);
   b = match_layout(a, b);
   auto y = torch::empty_like(a);
+  if (is_bf16(a)) {
+    launch_add_relu_bf16((const unsigned short*)a.data_ptr(),
+                         (const unsigned short*)b.data_ptr(),
+                         (unsigned short*)y.data_ptr(), a.numel(),
+                         stream_of(a));
+    return y;
+  }
   launch_add_relu(a.data_ptr<float>(), b.data_ptr<float>(),
                   y.data_ptr<float>(), a.numel(), stream_of(a));
   return y;
@@ -159,9 +231,15 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
   int OH = H / 2, OW = W / 2;
   auto y = empty_cl({Nb, C, OH, OW}, x.options());
   auto idx = empty_cl({Nb, C, OH, OW}, x.options().dtype(torch::kUInt8));
-  launch_maxpool2x2_fwd(x.data_ptr<float>(), y.data_ptr<float>(),
-                        idx.data_ptr<uint8_t>(), Nb, H, W, OH, OW, C,
-                        stream_of(x));
+  if (is_bf16(x))
+    launch_maxpool2x2_fwd_bf16((const unsigned short*)x.data_ptr(),
+                               (unsigned short*)y.data_ptr(),
+                               idx.data_ptr<uint8_t>(), Nb, H, W, OH, OW, C,
+                               stream_of(x));
+  else
+    launch_maxpool2x2_fwd(x.data_ptr<float>(), y.data_ptr<float>(),
+                          idx.data_ptr<uint8_t>(), Nb, H, W, OH, OW, C,
+                          stream_of(x));
   return {y, idx};
 }
 
@@ -172,9 +250,15 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
   int Nb = in_shape[0], C = in_shape[1], H = in_shape[2], W = in_shape[3];
   int OH = dy.size(2), OW = dy.size(3);
   auto dx = empty_cl({Nb, C, H, W}, dy.options());
-  launch_maxpool2x2_bwd(dy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
-                        dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
-                        stream_of(dy));
+  if (is_bf16(dy))
+    launch_maxpool2x2_bwd_bf16((const unsigned short*)dy.data_ptr(),
+                               idx.data_ptr<uint8_t>(),
+                               (unsigned short*)dx.data_ptr(), Nb, H, W, OH,
+                               OW, C, stream_of(dy));
+  else
+    launch_maxpool2x2_bwd(dy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
+                          dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
+                          stream_of(dy));
   return dx;
 }
 
@@ -199,19 +283,33 @@ std::tuple<torch::Tensor, torch::Tensor> dropout_fwd_dev(torch::Tensor x,
       state.scalar_type() == torch::kInt64);
   auto y = torch::empty_like(x);
   auto mask = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
-  launch_dropout_fwd_dev(
-      x.data_ptr<float>(), y.data_ptr<float>(), mask.data_ptr<uint8_t>(),
-      x.numel(), (float)p,
-      (const unsigned long long*)state.data_ptr(), (int)site, stream_of(x));
+  if (is_bf16(x))
+    launch_dropout_fwd_dev_bf16(
+        (const unsigned short*)x.data_ptr(), (unsigned short*)y.data_ptr(),
+        mask.data_ptr<uint8_t>(), x.numel(), (float)p,
+        (const unsigned long long*)state.data_ptr(), (int)site,
+        stream_of(x));
+  else
+    launch_dropout_fwd_dev(
+        x.data_ptr<float>(), y.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+        x.numel(), (float)p,
+        (const unsigned long long*)state.data_ptr(), (int)site,
+        stream_of(x));
   return {y, mask};
 }
 
 torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
   CHK_CUDA(dy);
   auto dx = torch::empty_like(dy);
-  launch_dropout_bwd(dy.data_ptr<float>(), mask.data_ptr<uint8_t>(),
-                     dx.data_ptr<float>(), dy.numel(), (float)p,
-                     stream_of(dy));
+  if (is_bf16(dy))
+    launch_dropout_bwd_bf16((const unsigned short*)dy.data_ptr(),
+                            mask.data_ptr<uint8_t>(),
+                            (unsigned short*)dx.data_ptr(), dy.numel(),
+                            (float)p, stream_of(dy));
+  else
+    launch_dropout_bwd(dy.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+                       dx.data_ptr<float>(), dy.numel(), (float)p,
+                       stream_of(dy));
   return dx;
 }
 
@@ -220,8 +318,13 @@ torch::Tensor gap_fwd(torch::Tensor x) {
   x = cl(x);
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = torch::empty({Nb, C}, x.options());
-  launch_gap_fwd(x.data_ptr<float>(), y.data_ptr<float>(), Nb, HW, C,
-                 stream_of(x));
+  if (is_bf16(x))
+    launch_gap_fwd_bf16((const unsigned short*)x.data_ptr(),
+                        (unsigned short*)y.data_ptr(), Nb, HW, C,
+                        stream_of(x));
+  else
+    launch_gap_fwd(x.data_ptr<float>(), y.data_ptr<float>(), Nb, HW, C,
+                   stream_of(x));
   return y;
 }
 
@@ -230,8 +333,13 @@ torch::Tensor gap_bwd(torch::Tensor dy, std::vector<int64_t> in_shape) {
   dy = dy.contiguous();
   int Nb = in_shape[0], C = in_shape[1], HW = in_shape[2] * in_shape[3];
   auto dx = empty_cl({Nb, C, in_shape[2], in_shape[3]}, dy.options());
-  launch_gap_bwd(dy.data_ptr<float>(), dx.data_ptr<float>(), Nb, HW, C,
-                 stream_of(dy));
+  if (is_bf16(dy))
+    launch_gap_bwd_bf16((const unsigned short*)dy.data_ptr(),
+                        (unsigned short*)dx.data_ptr(), Nb, HW, C,
+                        stream_of(dy));
+  else
+    launch_gap_bwd(dy.data_ptr<float>(), dx.data_ptr<float>(), Nb, HW, C,
+                   stream_of(dy));
   return dx;
 }
 
@@ -445,6 +553,10 @@ torch::Tensor transpose2d(torch::Tensor A) {
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> b, bool relu) {
   // y = x @ w^T (+b); w is (out, in) — transpose once, then plain GEMM
+  if (is_bf16(x)) {
+    auto wt16 = w.t().contiguous().to(torch::kBFloat16);
+    return gemm_bf16(x.contiguous(), wt16, b, relu, /*out_bf16=*/true);
+  }
   auto wt = transpose2d(w);
   return gemm(x, wt, b, relu);
 }
@@ -452,6 +564,14 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy) {
   dy = dy.contiguous();
+  if (is_bf16(x)) {
+    auto w16 = w.to(torch::kBFloat16).contiguous();
+    auto dx = gemm_bf16(dy, w16, c10::nullopt, false, /*out_bf16=*/true);
+    auto dyt = dy.t().contiguous();
+    auto dw = gemm_bf16(dyt, x.contiguous(), c10::nullopt, false, false);
+    auto db = dy.to(torch::kFloat).sum(0);
+    return {dx, dw, db};
+  }
   auto dx = gemm(dy, w, c10::nullopt, false);        // (M,out)x(out,in)
   auto dyt = transpose2d(dy);                        // (out, M)
   auto dw = gemm(dyt, x, c10::nullopt, false);       // (out, in)
@@ -465,6 +585,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> b, int64_t stride,
+                         int64_t pad, bool relu);
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> b, int64_t stride,
                          int64_t pad, bool relu) {
   TORCH_CHECK(x.is_cuda() && w.is_cuda());
   x = cl(x);
@@ -473,6 +596,26 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
   int OH = (H + 2 * pad - R) / stride + 1;
   int OW = (W + 2 * pad - S) / stride + 1;
+  if (is_bf16(x)) {
+    if ((C % 32) == 0) {
+      auto wt = torch::empty({(long)C * R * S, Kout},
+                             w.options().dtype(torch::kBFloat16));
+      launch_wperm_rsc_ko_bf16(w.data_ptr<float>(),
+                               (unsigned short*)wt.data_ptr(), Kout, C,
+                               R * S, stream_of(x));
+      auto y = empty_cl({Nb, Kout, OH, OW}, x.options());
+      launch_conv_fwd_bf16((const unsigned short*)x.data_ptr(),
+                           (const unsigned short*)wt.data_ptr(),
+                           b ? b->data_ptr<float>() : nullptr,
+                           (unsigned short*)y.data_ptr(), Nb, C, H, W, Kout,
+                           R, S, OH, OW, (int)stride, (int)pad,
+                           relu ? 1 : 0, stream_of(x));
+      return y;
+    }
+    // first-layer fallback (C=1,3): fp32 kernels, bf16 in/out casts
+    auto y32 = conv2d_fwd(x.to(torch::kFloat), w, b, stride, pad, relu);
+    return y32.to(torch::kBFloat16);
+  }
   auto wt = torch::empty({(long)C * R * S, Kout}, w.options());
   launch_wperm_crs_ko(w.data_ptr<float>(), wt.data_ptr<float>(), Kout, C,
                       R * S, stream_of(x));
@@ -486,6 +629,9 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, int64_t stride,
+    int64_t pad, bool has_b, bool need_dx);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor dy, int64_t stride,
     int64_t pad, bool has_b, bool need_dx) {
   TORCH_CHECK(x.is_cuda());
   x = cl(x);
@@ -495,6 +641,52 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
   int OH = dy.size(2), OW = dy.size(3);
   auto st = stream_of(x);
+
+  if (is_bf16(x)) {
+    bool fast = (Kout % 32) == 0 && (C % 8) == 0;
+    if (!fast) {
+      // fp32 fallback for the tiny first layers
+      auto r32 = conv2d_bwd(x.to(torch::kFloat), w, dy.to(torch::kFloat),
+                            stride, pad, has_b, need_dx);
+      auto dx32 = std::get<0>(r32);
+      return {dx32.defined() ? dx32.to(torch::kBFloat16) : dx32,
+              std::get<1>(r32), std::get<2>(r32)};
+    }
+    torch::Tensor dxb;
+    if (need_dx) {
+      auto wp = torch::empty({(long)Kout * R * S, C},
+                             w.options().dtype(torch::kBFloat16));
+      launch_wperm_rsko_c_bf16(w.data_ptr<float>(),
+                               (unsigned short*)wp.data_ptr(), Kout, C,
+                               R * S, st);
+      dxb = empty_cl({Nb, C, H, W}, x.options());
+      launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
+                                (const unsigned short*)wp.data_ptr(),
+                                (unsigned short*)dxb.data_ptr(), Nb, C, H,
+                                W, Kout, R, S, OH, OW, (int)stride,
+                                (int)pad, st);
+    }
+    int Ncrs = C * R * S;
+    long Kdim = (long)Nb * OH * OW;
+    int SK = conv_bwd_weight_splitk(Kout, Ncrs, Kdim);
+    auto dw = torch::empty_like(w);
+    auto ws = torch::empty({((long)SK + 1) * Kout * Ncrs},
+                           w.options());  // fp32 slabs + rsc temp
+    launch_conv_bwd_weight_bf16((const unsigned short*)dy.data_ptr(),
+                                (const unsigned short*)x.data_ptr(),
+                                dw.data_ptr<float>(), ws.data_ptr<float>(),
+                                SK, Nb, C, H, W, Kout, R, S, OH, OW,
+                                (int)stride, (int)pad, st);
+    torch::Tensor db;
+    if (has_b) {
+      db = torch::empty({Kout}, w.options());
+      auto parts = torch::empty({(long)Kout * 4096}, w.options());
+      launch_conv_db_bf16((const unsigned short*)dy.data_ptr(),
+                          db.data_ptr<float>(), parts.data_ptr<float>(), Nb,
+                          Kout, OH * OW, st);
+    }
+    return {dxb, dw, db};
+  }
 
   torch::Tensor dx;
   if (need_dx) {
@@ -547,13 +739,27 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                     x.options());
   auto save_mean = torch::empty({C}, x.options());
   auto save_rstd = torch::empty({C}, x.options());
-  auto scratch = torch::empty({(64 * 2 + 2) * (long)C}, x.options());
-  launch_bn_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
-                b.data_ptr<float>(), running_mean.data_ptr<float>(),
-                running_var.data_ptr<float>(), save_mean.data_ptr<float>(),
-                save_rstd.data_ptr<float>(), y.data_ptr<float>(),
-                scratch.data_ptr<float>(), Nb, C, HW, (float)momentum,
-                (float)eps, training ? 1 : 0, stream_of(x));
+  auto scratch = torch::empty({(64 * 2 + 2) * (long)C},
+                              x.options().dtype(torch::kFloat));
+  if (is_bf16(x))
+    launch_bn_fwd_bf16((const unsigned short*)x.data_ptr(),
+                       w.data_ptr<float>(), b.data_ptr<float>(),
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(),
+                       save_mean.data_ptr<float>(),
+                       save_rstd.data_ptr<float>(),
+                       (unsigned short*)y.data_ptr(),
+                       scratch.data_ptr<float>(), Nb, C, HW,
+                       (float)momentum, (float)eps, training ? 1 : 0,
+                       stream_of(x));
+  else
+    launch_bn_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
+                  b.data_ptr<float>(), running_mean.data_ptr<float>(),
+                  running_var.data_ptr<float>(),
+                  save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
+                  y.data_ptr<float>(), scratch.data_ptr<float>(), Nb, C,
+                  HW, (float)momentum, (float)eps, training ? 1 : 0,
+                  stream_of(x));
   return {y, save_mean, save_rstd};
 }
 
@@ -566,14 +772,25 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto dx = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
                      x.options());
-  auto dw = torch::empty({C}, x.options());
-  auto db = torch::empty({C}, x.options());
-  auto scratch = torch::empty({(64 * 2 + 2) * (long)C}, x.options());
-  launch_bn_bwd(x.data_ptr<float>(), dy.data_ptr<float>(),
-                w.data_ptr<float>(), save_mean.data_ptr<float>(),
-                save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
-                dx.data_ptr<float>(), dw.data_ptr<float>(),
-                db.data_ptr<float>(), Nb, C, HW, 1, stream_of(x));
+  auto dw = torch::empty({C}, x.options().dtype(torch::kFloat));
+  auto db = torch::empty({C}, x.options().dtype(torch::kFloat));
+  auto scratch = torch::empty({(64 * 2 + 2) * (long)C},
+                              x.options().dtype(torch::kFloat));
+  if (is_bf16(x))
+    launch_bn_bwd_bf16((const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)dy.data_ptr(),
+                       w.data_ptr<float>(), save_mean.data_ptr<float>(),
+                       save_rstd.data_ptr<float>(),
+                       scratch.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(),
+                       dw.data_ptr<float>(), db.data_ptr<float>(), Nb, C,
+                       HW, 1, stream_of(x));
+  else
+    launch_bn_bwd(x.data_ptr<float>(), dy.data_ptr<float>(),
+                  w.data_ptr<float>(), save_mean.data_ptr<float>(),
+                  save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
+                  dx.data_ptr<float>(), dw.data_ptr<float>(),
+                  db.data_ptr<float>(), Nb, C, HW, 1, stream_of(x));
   return {dx, dw, db};
 }
 

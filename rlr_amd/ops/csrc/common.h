@@ -61,6 +61,37 @@ __device__ __forceinline__ Philox4 philox4(uint64_t seed, uint64_t offset,
   return {c0, c1, c2, c3};
 }
 
+// ------------------------------------------------- bf16 <-> f32 helpers
+// bf16 is carried as raw ushort; loads/stores go through these so kernels
+// can be templated on the storage type (float or ushort-bf16).
+__device__ __forceinline__ float bf2f_(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+__device__ __forceinline__ unsigned short f2bf_(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  unsigned int r = v.i + 0x7FFF + ((v.i >> 16) & 1);  // round-nearest-even
+  return (unsigned short)(r >> 16);
+}
+template <typename T>
+__device__ __forceinline__ float ldv(const T* p);
+template <>
+__device__ __forceinline__ float ldv<float>(const float* p) { return *p; }
+template <>
+__device__ __forceinline__ float ldv<unsigned short>(
+    const unsigned short* p) { return bf2f_(*p); }
+template <typename T>
+__device__ __forceinline__ void stv(T* p, float v);
+template <>
+__device__ __forceinline__ void stv<float>(float* p, float v) { *p = v; }
+template <>
+__device__ __forceinline__ void stv<unsigned short>(unsigned short* p,
+                                                    float v) {
+  *p = f2bf_(v);
+}
+
 __device__ __forceinline__ float u32_to_uniform(uint32_t v) {
   // (0,1]: matches the usual counter-RNG convention
   return (v >> 8) * (1.0f / 16777216.0f) + (0.5f / 16777216.0f);

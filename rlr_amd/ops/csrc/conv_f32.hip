@@ -1014,6 +1014,18 @@ void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
   conv_db_stage2_k<<<Kout, kBlock, 0, st>>>(partials, db, Kout, chunks);
 }
 
+void launch_dwperm_rsc_crs(const float* in, float* out, int Kout, int C,
+                           int RS, void* s) {
+  dwperm_rsc_crs_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
+                     (hipStream_t)s>>>(in, out, Kout, C, RS);
+}
+
+void launch_conv_db_stage2(const float* partials, float* db, int Kout,
+                           int chunks, void* s) {
+  conv_db_stage2_k<<<Kout, kBlock, 0, (hipStream_t)s>>>(partials, db, Kout,
+                                                        chunks);
+}
+
 void launch_wperm_crs_ko(const float* w, float* out, int Kout, int C, int RS,
                          void* s) {
   wperm_rsc_ko_k<<<grid_for((long)Kout * C * RS), kBlock, 0,

@@ -76,8 +76,9 @@ void launch_add_relu(const float* a, const float* b, float* y, long n,
 
 // channels_last: c is the fastest axis, so threads over the flat index are
 // coalesced.  idx stores the 2x2 argmax (0..3) for the gather backward.
-__global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
-                                 float* __restrict__ y,
+template <typename T>
+__global__ void maxpool2x2_fwd_k(const T* __restrict__ x,
+                                 T* __restrict__ y,
                                  uint8_t* __restrict__ idx,
                                  long B, int H, int W, int OH, int OW,
                                  int C) {
@@ -94,21 +95,23 @@ __global__ void maxpool2x2_fwd_k(const float* __restrict__ x,
     int ow = rest % OW;
     int oh = (rest / OW) % OH;
     long b = rest / ((long)OW * OH);
-    const float* p = x + ((b * H + 2 * oh) * (long)W + 2 * ow) * C + c;
-    float v00 = p[0], v01 = p[C], v10 = p[WC], v11 = p[WC + C];
+    const T* p = x + ((b * H + 2 * oh) * (long)W + 2 * ow) * C + c;
+    float v00 = ldv(p), v01 = ldv(p + C), v10 = ldv(p + WC),
+          v11 = ldv(p + WC + C);
     float m = v00; uint8_t a = 0;
     if (v01 > m) { m = v01; a = 1; }
     if (v10 > m) { m = v10; a = 2; }
     if (v11 > m) { m = v11; a = 3; }
-    y[i] = m;
+    stv(&y[i], m);
     idx[i] = a;
   }
 }
 
 // Gather form: one thread per INPUT element (write-once, no zero-init).
-__global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
+template <typename T>
+__global__ void maxpool2x2_bwd_k(const T* __restrict__ dy,
                                  const uint8_t* __restrict__ idx,
-                                 float* __restrict__ dx,
+                                 T* __restrict__ dx,
                                  long B, int H, int W, int OH, int OW,
                                  int C) {
   long n_in = B * H * (long)W * C;
@@ -127,9 +130,9 @@ __global__ void maxpool2x2_bwd_k(const float* __restrict__ dy,
     if (oh < OH && ow < OW) {
       long o = ((b * OH + oh) * (long)OW + ow) * C + c;
       uint8_t a = ((ih & 1) << 1) | (iw & 1);
-      if (idx[o] == a) g = dy[o];
+      if (idx[o] == a) g = ldv(&dy[o]);
     }
-    dx[i] = g;
+    stv(&dx[i], g);
   }
 }
 
@@ -137,17 +140,32 @@ extern "C" {
 void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, long B,
                            int H, int W, int OH, int OW, int C, void* s) {
   long n = B * OH * (long)OW * C;
-  maxpool2x2_fwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, idx, B,
-                                                               H, W, OH, OW,
-                                                               C);
+  maxpool2x2_fwd_k<float><<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
+      x, y, idx, B, H, W, OH, OW, C);
 }
 void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
                            long B, int H, int W, int OH, int OW, int C,
                            void* s) {
   long n = B * H * (long)W * C;
-  maxpool2x2_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx,
-                                                               B, H, W, OH,
-                                                               OW, C);
+  maxpool2x2_bwd_k<float><<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
+      dy, idx, dx, B, H, W, OH, OW, C);
+}
+void launch_maxpool2x2_fwd_bf16(const unsigned short* x, unsigned short* y,
+                                uint8_t* idx, long B, int H, int W, int OH,
+                                int OW, int C, void* s) {
+  long n = B * OH * (long)OW * C;
+  maxpool2x2_fwd_k<unsigned short>
+      <<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, idx, B, H, W, OH,
+                                                   OW, C);
+}
+void launch_maxpool2x2_bwd_bf16(const unsigned short* dy,
+                                const uint8_t* idx, unsigned short* dx,
+                                long B, int H, int W, int OH, int OW, int C,
+                                void* s) {
+  long n = B * H * (long)W * C;
+  maxpool2x2_bwd_k<unsigned short>
+      <<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx, B, H, W, OH,
+                                                   OW, C);
 }
 }
 
@@ -397,5 +415,137 @@ void launch_eval_update(const float* logits, const long* labels, float* conf,
   eval_update_k<<<1, kBlock, 0, (hipStream_t)s>>>(logits, labels, conf,
                                                   loss_sum, B, C,
                                                   num_classes);
+}
+}
+
+// -------------------------------------------------- bf16 scalar variants
+
+__global__ void relu_fwd_bf16_k(const unsigned short* __restrict__ x,
+                                unsigned short* __restrict__ y, long n) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    y[i] = bf2f_(x[i]) > 0.f ? x[i] : (unsigned short)0;
+}
+
+__global__ void relu_bwd_bf16_k(const unsigned short* __restrict__ y,
+                                const unsigned short* __restrict__ dy,
+                                unsigned short* __restrict__ dx, long n) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dx[i] = bf2f_(y[i]) > 0.f ? dy[i] : (unsigned short)0;
+}
+
+__global__ void add_relu_bf16_k(const unsigned short* __restrict__ a,
+                                const unsigned short* __restrict__ b,
+                                unsigned short* __restrict__ y, long n) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    y[i] = f2bf_(fmaxf(bf2f_(a[i]) + bf2f_(b[i]), 0.f));
+}
+
+__global__ void dropout_fwd_dev_bf16_k(const unsigned short* __restrict__ x,
+                                       unsigned short* __restrict__ y,
+                                       uint8_t* __restrict__ mask, long n,
+                                       float p,
+                                       const unsigned long long* __restrict__
+                                           state, int site) {
+  uint64_t seed = state[0];
+  uint64_t offset = state[1] + site;
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = (n + 3) / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    Philox4 r = philox4(seed, offset, (uint32_t)i);
+    uint32_t rv[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long k = i * 4 + j;
+      if (k < n) {
+        bool keep = u32_to_uniform(rv[j]) >= p;
+        mask[k] = keep;
+        y[k] = keep ? f2bf_(bf2f_(x[k]) * scale) : (unsigned short)0;
+      }
+    }
+  }
+}
+
+__global__ void dropout_bwd_bf16_k(const unsigned short* __restrict__ dy,
+                                   const uint8_t* __restrict__ mask,
+                                   unsigned short* __restrict__ dx, long n,
+                                   float p) {
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dx[i] = mask[i] ? f2bf_(bf2f_(dy[i]) * scale) : (unsigned short)0;
+}
+
+__global__ void gap_fwd_bf16_k(const unsigned short* __restrict__ x,
+                               unsigned short* __restrict__ y, long B,
+                               int HW, int C) {
+  long n = B * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / C;
+    const unsigned short* p = x + b * (long)HW * C + c;
+    float acc = 0.f;
+    for (int hw = 0; hw < HW; ++hw) acc += bf2f_(p[(long)hw * C]);
+    y[i] = f2bf_(acc / HW);
+  }
+}
+
+__global__ void gap_bwd_bf16_k(const unsigned short* __restrict__ dy,
+                               unsigned short* __restrict__ dx, long B,
+                               int HW, int C) {
+  long n = B * (long)HW * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / ((long)HW * C);
+    dx[i] = f2bf_(bf2f_(dy[b * C + c]) / HW);
+  }
+}
+
+extern "C" {
+void launch_relu_fwd_bf16(const unsigned short* x, unsigned short* y,
+                          long n, void* s) {
+  relu_fwd_bf16_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(x, y, n);
+}
+void launch_relu_bwd_bf16(const unsigned short* y, const unsigned short* dy,
+                          unsigned short* dx, long n, void* s) {
+  relu_bwd_bf16_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(y, dy, dx, n);
+}
+void launch_add_relu_bf16(const unsigned short* a, const unsigned short* b,
+                          unsigned short* y, long n, void* s) {
+  add_relu_bf16_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(a, b, y, n);
+}
+void launch_dropout_fwd_dev_bf16(const unsigned short* x, unsigned short* y,
+                                 uint8_t* mask, long n, float p,
+                                 const unsigned long long* state, int site,
+                                 void* s) {
+  dropout_fwd_dev_bf16_k<<<grid_for((n + 3) / 4), kBlock, 0,
+                           (hipStream_t)s>>>(x, y, mask, n, p, state, site);
+}
+void launch_dropout_bwd_bf16(const unsigned short* dy, const uint8_t* mask,
+                             unsigned short* dx, long n, float p, void* s) {
+  dropout_bwd_bf16_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, mask,
+                                                                 dx, n, p);
+}
+void launch_gap_fwd_bf16(const unsigned short* x, unsigned short* y, long B,
+                         int HW, int C, void* s) {
+  gap_fwd_bf16_k<<<grid_for(B * C), kBlock, 0, (hipStream_t)s>>>(x, y, B,
+                                                                 HW, C);
+}
+void launch_gap_bwd_bf16(const unsigned short* dy, unsigned short* dx,
+                         long B, int HW, int C, void* s) {
+  gap_bwd_bf16_k<<<grid_for(B * (long)HW * C), kBlock, 0,
+                   (hipStream_t)s>>>(dy, dx, B, HW, C);
 }
 }

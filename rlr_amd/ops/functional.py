@@ -225,8 +225,11 @@ class _CrossEntropy(torch.autograd.Function):
 
 
 def cross_entropy(logits, labels):
-    """Mean-reduced CE (the reference's criterion, federated.py:61)."""
+    """Mean-reduced CE (the reference's criterion, federated.py:61).
+    bf16 logits are cast up (autograd casts the gradient back down)."""
     if _gpu(logits):
+        if logits.dtype == torch.bfloat16:
+            logits = logits.float()
         return _CrossEntropy.apply(logits, labels)
     return F.cross_entropy(logits, labels)
 

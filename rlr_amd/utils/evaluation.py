@@ -29,8 +29,10 @@ def get_loss_n_accuracy(model, X, Y, args, num_classes=10, bs=None):
     for lo in range(0, n, bs):
         inputs, labels = X[lo:lo + bs], Y[lo:lo + bs]
         outputs = model(inputs)
+        if outputs.dtype != torch.float32:
+            outputs = outputs.float()
         if use_hip:
-            ext().eval_update(outputs, labels, conf, loss_sum)
+            ext().eval_update(outputs.contiguous(), labels, conf, loss_sum)
         else:
             loss_sum += torch.nn.functional.cross_entropy(
                 outputs, labels, reduction='sum').double()
