@@ -1,0 +1,31 @@
+#!/usr/bin/env bash
+# GPU-side CI batch, run via: gpurun -- 'bash scripts/gpu_ci.sh [what]'
+set -x
+mkdir -p gpurun_out
+WHAT="${1:-all}"
+
+if [ "$WHAT" = "bf16" ] || [ "$WHAT" = "all" ]; then
+  timeout 600 python -m pytest tests/test_bf16_gpu.py tests/test_bf16_conv_gpu.py -q \
+      > gpurun_out/ci_bf16.log 2>&1
+  echo "bf16 rc=$?" >> gpurun_out/ci_bf16.log
+  tail -12 gpurun_out/ci_bf16.log
+fi
+
+if [ "$WHAT" = "suite" ] || [ "$WHAT" = "all" ]; then
+  timeout 700 python -m pytest tests/ -m gpu -q > gpurun_out/ci_suite.log 2>&1
+  echo "suite rc=$?" >> gpurun_out/ci_suite.log
+  tail -6 gpurun_out/ci_suite.log
+fi
+
+if [ "$WHAT" = "bench" ] || [ "$WHAT" = "all" ]; then
+  timeout 300 python bench.py --steps 5 --warmup 2 > gpurun_out/ci_bench.log 2>&1
+  grep -o '"value": [0-9.]*' gpurun_out/ci_bench.log
+fi
+
+if [ "$WHAT" = "benchr" ] || [ "$WHAT" = "all" ]; then
+  timeout 300 python bench.py --data cifar10 --model resnet18 --dtype bf16 \
+      --steps 3 --warmup 1 > gpurun_out/ci_benchr.log 2>&1
+  echo "benchr rc=$?"
+  grep -o '"value": [0-9.]*\|"ms_per_step": [0-9.]*' gpurun_out/ci_benchr.log
+  tail -2 gpurun_out/ci_benchr.log | head -1
+fi
