@@ -737,8 +737,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
                     x.options());
-  auto save_mean = torch::empty({C}, x.options());
-  auto save_rstd = torch::empty({C}, x.options());
+  auto save_mean = torch::empty({C}, x.options().dtype(torch::kFloat));
+  auto save_rstd = torch::empty({C}, x.options().dtype(torch::kFloat));
   auto scratch = torch::empty({(64 * 2 + 2) * (long)C},
                               x.options().dtype(torch::kFloat));
   if (is_bf16(x))
