@@ -34,6 +34,7 @@ def main():
     ap.add_argument('--model', type=str, default=None,
                     help="e.g. resnet18 (BASELINE configs 3-4: CIFAR10 "
                          "ResNet18); default: the dataset's reference CNN")
+    ap.add_argument('--agents_per_stream_override', type=int, default=0)
     ap.add_argument('--dtype', type=str, default='fp32',
                     choices=['fp32', 'bf16'],
                     help="client compute dtype; the headline FMNIST config "
@@ -60,7 +61,8 @@ def main():
         data=a.data, model=a.model, dtype=a.dtype, num_agents=num_agents,
         num_corrupt=1 * n, poison_frac=0.5, robustLR_threshold=4,
         aggr='avg', local_ep=2, bs=256, agent_frac=1.0, synthetic=True,
-        no_tb=True, snap=10 ** 9, device=device)
+        no_tb=True, snap=10 ** 9, device=device,
+        agents_per_stream=a.agents_per_stream_override)
 
     world_state = build_world(args)
 

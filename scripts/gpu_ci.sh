@@ -29,3 +29,11 @@ if [ "$WHAT" = "benchr" ] || [ "$WHAT" = "all" ]; then
   grep -o '"value": [0-9.]*\|"ms_per_step": [0-9.]*' gpurun_out/ci_benchr.log
   tail -2 gpurun_out/ci_benchr.log | head -1
 fi
+
+if [ "$WHAT" = "sweep" ]; then
+  for R in 2 4 6 8; do
+    timeout 200 python bench.py --steps 4 --warmup 2 --agents_per_stream_override $R \
+      > gpurun_out/sweep_$R.log 2>&1 || true
+  done
+  grep -H -o '"value": [0-9.]*' gpurun_out/sweep_*.log
+fi
