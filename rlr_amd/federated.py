@@ -95,7 +95,10 @@ def _replica_pool(world, args, gm):
     federated.py:68-72; the tiny CNN's kernels underfill 256 CUs, so
     stream-level concurrency is the chip-filling lever)."""
     if 'replicas' not in world:
-        R = args.agents_per_stream if args.agents_per_stream > 0 else 4
+        # sweep on MI355X (BASELINE.md): 2 replicas > 4/6/8 — the big conv
+        # kernels already fill the chip; 2 streams cover the small-kernel
+        # tails without cache thrash
+        R = args.agents_per_stream if args.agents_per_stream > 0 else 2
         reps = []
         for _ in range(max(1, R)):
             m = get_model(args.data, getattr(args, 'model', None))
