@@ -80,7 +80,8 @@ void launch_add_noise(double*, double, long, uint64_t, uint64_t, void*);
 // gemm_f32.hip
 int gemm_f32_splitk(int, int, int);
 void launch_gemm_f32(const float*, const float*, float*, const float*,
-                     float*, int, int, int, int, int, int, int, int, void*);
+                     float*, int, int, int, int, int, int, int, int, int,
+                     void*);
 void launch_transpose_f32(const float*, float*, int, int, void*);
 void launch_colsum(const float*, float*, int, int, void*);
 // conv_f32.hip
@@ -502,7 +503,28 @@ torch::Tensor gemm(torch::Tensor A, torch::Tensor B,
   launch_gemm_f32(A.data_ptr<float>(), B.data_ptr<float>(),
                   C.data_ptr<float>(),
                   bias ? bias->data_ptr<float>() : nullptr, wsp, M, N, K, K,
-                  N, N, SK, relu ? 1 : 0, stream_of(A));
+                  N, N, SK, relu ? 1 : 0, 0, stream_of(A));
+  return C;
+}
+
+// shared fp32 GEMM driver with explicit operand layout
+static torch::Tensor gemm_layout(const torch::Tensor& A,
+                                 const torch::Tensor& B,
+                                 c10::optional<torch::Tensor> bias,
+                                 bool relu, int M, int N, int K, int lda,
+                                 int ldb, int layout) {
+  auto C = torch::empty({M, N}, A.options());
+  int SK = gemm_f32_splitk(M, N, K);
+  torch::Tensor ws;
+  float* wsp = nullptr;
+  if (SK > 1) {
+    ws = torch::empty({(long)SK * M * N}, A.options());
+    wsp = ws.data_ptr<float>();
+  }
+  launch_gemm_f32(A.data_ptr<float>(), B.data_ptr<float>(),
+                  C.data_ptr<float>(),
+                  bias ? bias->data_ptr<float>() : nullptr, wsp, M, N, K,
+                  lda, ldb, N, SK, relu ? 1 : 0, layout, stream_of(A));
   return C;
 }
 
@@ -557,8 +579,11 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
     auto wt16 = w.t().contiguous().to(torch::kBFloat16);
     return gemm_bf16(x.contiguous(), wt16, b, relu, /*out_bf16=*/true);
   }
-  auto wt = transpose2d(w);
-  return gemm(x, wt, b, relu);
+  // w (out,in) consumed directly as B^T (layout 2) — no transpose pass
+  x = x.contiguous();
+  w = w.contiguous();
+  int M = x.size(0), K = x.size(1), N = w.size(0);
+  return gemm_layout(x, w, b, relu, M, N, K, K, K, 2);
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
@@ -572,12 +597,16 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
     auto db = dy.to(torch::kFloat).sum(0);
     return {dx, dw, db};
   }
-  auto dx = gemm(dy, w, c10::nullopt, false);        // (M,out)x(out,in)
-  auto dyt = transpose2d(dy);                        // (out, M)
-  auto dw = gemm(dyt, x, c10::nullopt, false);       // (out, in)
-  auto db = torch::empty({dy.size(1)}, dy.options());
-  launch_colsum(dy.data_ptr<float>(), db.data_ptr<float>(), dy.size(0),
-                dy.size(1), stream_of(dy));
+  x = x.contiguous();
+  w = w.contiguous();
+  int bM = x.size(0), in = x.size(1), out = w.size(0);
+  // dx = dY (M,out) @ W (out,in): plain NN
+  auto dx = gemm_layout(dy, w, c10::nullopt, false, bM, in, out, out, in, 0);
+  // dw = dY^T @ X: A = dY consumed transposed (layout 1)
+  auto dw = gemm_layout(dy, x, c10::nullopt, false, out, in, bM, out, in, 1);
+  auto db = torch::empty({out}, dy.options());
+  launch_colsum(dy.data_ptr<float>(), db.data_ptr<float>(), bM, out,
+                stream_of(dy));
   return {dx, dw, db};
 }
 

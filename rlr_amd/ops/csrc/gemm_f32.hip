@@ -26,7 +26,10 @@ constexpr int LDB_S = BN + 16;  // B_lds row stride (floats)
 
 // Staging guards handle M/N/K edges by zero-fill; VEC selects float4 global
 // loads (requires 16B-aligned rows: ld % 4 == 0).
-template <bool VEC>
+// AT: A is stored transposed [K][M] (lda = its row length = M direction);
+// BT: B is stored transposed [N][K] (e.g. a torch Linear weight (out,in)
+// consumed directly — no separate transpose pass).
+template <bool VEC, bool AT = false, bool BT = false>
 __global__ __launch_bounds__(256)
 void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
                 float* __restrict__ C, const float* __restrict__ bias,
@@ -59,56 +62,127 @@ void gemm_f32_k(const float* __restrict__ A, const float* __restrict__ B,
 
   // async-STAGE split (guide T14/G15): global loads go to registers a
   // K-tile early; the LDS write lands after the barrier, under the MFMAs.
+  // AT/BT staging loads along the transposed storage rows (still float4
+  // coalesced) and scatter-writes the LDS image transposed.
   float4 ra[4], rb[2];
+  // transposed-staging thread coordinates
+  const int tak = t >> 3;           // AT: k index, 0..31
+  const int tam = (t & 7) * 4;      // AT: m base, +32 per round (4 rounds)
+  const int tbn = t >> 3;           // BT: n index base, +32 per round (2)
+  const int tbk = (t & 7) * 4;      // BT: k base
   auto stage_load = [&](long k0) {
+    if (AT) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int m = am + j * 32;
-      long gm = m_blk + m;
-      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
-      if (gm < M) {
-        long base = gm * (long)lda + k0 + ak;
-        if (VEC && k0 + ak + 3 < k_hi) {
-          const float4 q = *(const float4*)(A + base);
-          v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
-        } else {
-          if (k0 + ak + 0 < k_hi) v0 = A[base + 0];
-          if (k0 + ak + 1 < k_hi) v1 = A[base + 1];
-          if (k0 + ak + 2 < k_hi) v2 = A[base + 2];
-          if (k0 + ak + 3 < k_hi) v3 = A[base + 3];
+      for (int j = 0; j < 4; ++j) {
+        long gk = k0 + tak;
+        long gm = m_blk + tam + j * 32;
+        float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+        if (gk < k_hi) {
+          long base = gk * (long)lda + gm;
+          if (VEC && gm + 3 < M) {
+            const float4 q = *(const float4*)(A + base);
+            v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+          } else {
+            if (gm + 0 < M) v0 = A[base + 0];
+            if (gm + 1 < M) v1 = A[base + 1];
+            if (gm + 2 < M) v2 = A[base + 2];
+            if (gm + 3 < M) v3 = A[base + 3];
+          }
         }
+        ra[j] = {v0, v1, v2, v3};
       }
-      ra[j] = {v0, v1, v2, v3};
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int m = am + j * 32;
+        long gm = m_blk + m;
+        float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+        if (gm < M) {
+          long base = gm * (long)lda + k0 + ak;
+          if (VEC && k0 + ak + 3 < k_hi) {
+            const float4 q = *(const float4*)(A + base);
+            v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+          } else {
+            if (k0 + ak + 0 < k_hi) v0 = A[base + 0];
+            if (k0 + ak + 1 < k_hi) v1 = A[base + 1];
+            if (k0 + ak + 2 < k_hi) v2 = A[base + 2];
+            if (k0 + ak + 3 < k_hi) v3 = A[base + 3];
+          }
+        }
+        ra[j] = {v0, v1, v2, v3};
+      }
     }
+    if (BT) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      long gk = k0 + bk + j * 16;
-      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
-      if (gk < k_hi) {
-        long base = gk * (long)ldb + n_blk + bn;
-        if (VEC && n_blk + bn + 3 < N) {
-          const float4 q = *(const float4*)(B + base);
-          v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
-        } else {
-          if (n_blk + bn + 0 < N) v0 = B[base + 0];
-          if (n_blk + bn + 1 < N) v1 = B[base + 1];
-          if (n_blk + bn + 2 < N) v2 = B[base + 2];
-          if (n_blk + bn + 3 < N) v3 = B[base + 3];
+      for (int j = 0; j < 2; ++j) {
+        long gn = n_blk + tbn + j * 32;
+        long gk = k0 + tbk;
+        float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+        if (gn < N) {
+          long base = gn * (long)ldb + gk;
+          if (VEC && gk + 3 < k_hi) {
+            const float4 q = *(const float4*)(B + base);
+            v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+          } else {
+            if (gk + 0 < k_hi) v0 = B[base + 0];
+            if (gk + 1 < k_hi) v1 = B[base + 1];
+            if (gk + 2 < k_hi) v2 = B[base + 2];
+            if (gk + 3 < k_hi) v3 = B[base + 3];
+          }
         }
+        rb[j] = {v0, v1, v2, v3};
       }
-      rb[j] = {v0, v1, v2, v3};
+    } else {
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        long gk = k0 + bk + j * 16;
+        float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+        if (gk < k_hi) {
+          long base = gk * (long)ldb + n_blk + bn;
+          if (VEC && n_blk + bn + 3 < N) {
+            const float4 q = *(const float4*)(B + base);
+            v0 = q.x; v1 = q.y; v2 = q.z; v3 = q.w;
+          } else {
+            if (n_blk + bn + 0 < N) v0 = B[base + 0];
+            if (n_blk + bn + 1 < N) v1 = B[base + 1];
+            if (n_blk + bn + 2 < N) v2 = B[base + 2];
+            if (n_blk + bn + 3 < N) v3 = B[base + 3];
+          }
+        }
+        rb[j] = {v0, v1, v2, v3};
+      }
     }
   };
   auto stage_write = [&](int buf) {
+    if (AT) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
-      ((float2*)dst)[0] = {ra[j].x, ra[j].y};
-      ((float2*)dst)[1] = {ra[j].z, ra[j].w};
+      for (int j = 0; j < 4; ++j) {
+        const float v[4] = {ra[j].x, ra[j].y, ra[j].z, ra[j].w};
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          A_lds[buf][(tam + j * 32 + e) * LDA_S + tak] = v[e];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float* dst = &A_lds[buf][(am + j * 32) * LDA_S + ak];
+        ((float2*)dst)[0] = {ra[j].x, ra[j].y};
+        ((float2*)dst)[1] = {ra[j].z, ra[j].w};
+      }
     }
+    if (BT) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j)
-      *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+      for (int j = 0; j < 2; ++j) {
+        const float v[4] = {rb[j].x, rb[j].y, rb[j].z, rb[j].w};
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          B_lds[buf][(tbk + e) * LDB_S + tbn + j * 32] = v[e];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        *(float4*)&B_lds[buf][(bk + j * 16) * LDB_S + bn] = rb[j];
+    }
   };
 
   stage_load(k_lo);
@@ -271,16 +345,36 @@ int gemm_f32_splitk(int M, int N, int K) {
 }
 
 // ws: null unless SK>1, then SK*M*N floats.
+// layout: 0 = NN, 1 = A transposed ([K][M]), 2 = B transposed ([N][K]).
 void launch_gemm_f32(const float* A, const float* B, float* C,
                      const float* bias, float* ws, int M, int N, int K,
-                     int lda, int ldb, int ldc, int SK, int relu, void* s) {
+                     int lda, int ldb, int ldc, int SK, int relu,
+                     int layout, void* s) {
   hipStream_t st = (hipStream_t)s;
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, SK);
   long k_per_chunk = SK == 1 ? (long)K
                              : ((((long)K + SK - 1) / SK + BK - 1) / BK) * BK;
   bool vec = (lda % 4 == 0) && (ldb % 4 == 0);
   float* out = SK == 1 ? C : ws;
-  if (vec)
+  if (layout == 1) {
+    if (vec)
+      gemm_f32_k<true, true, false><<<grid, 256, 0, st>>>(
+          A, B, out, bias, M, N, K, lda, ldb, ldc, k_per_chunk, relu,
+          SK == 1);
+    else
+      gemm_f32_k<false, true, false><<<grid, 256, 0, st>>>(
+          A, B, out, bias, M, N, K, lda, ldb, ldc, k_per_chunk, relu,
+          SK == 1);
+  } else if (layout == 2) {
+    if (vec)
+      gemm_f32_k<true, false, true><<<grid, 256, 0, st>>>(
+          A, B, out, bias, M, N, K, lda, ldb, ldc, k_per_chunk, relu,
+          SK == 1);
+    else
+      gemm_f32_k<false, false, true><<<grid, 256, 0, st>>>(
+          A, B, out, bias, M, N, K, lda, ldb, ldc, k_per_chunk, relu,
+          SK == 1);
+  } else if (vec)
     gemm_f32_k<true><<<grid, 256, 0, st>>>(A, B, out, bias, M, N, K, lda,
                                            ldb, ldc, k_per_chunk, relu,
                                            SK == 1);
