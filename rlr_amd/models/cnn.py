@@ -49,7 +49,7 @@ class CNN_MNIST(_OpsModel):
         x = Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True)
         x = Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True)
         x = Fo.max_pool2d_2x2(x)
-        x = x.reshape(x.shape[0], -1)
+        x = Fo.flatten2d(x)
         x = Fo.dropout(x, self.p_drop, self.training, self.rng)
         x = Fo.linear(x, self.fc1.weight, self.fc1.bias, relu=True)
         x = Fo.dropout(x, self.p_drop, self.training, self.rng)
@@ -77,7 +77,7 @@ class CNN_CIFAR(_OpsModel):
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv1.weight, self.conv1.bias, relu=True))
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv2.weight, self.conv2.bias, relu=True))
         x = Fo.max_pool2d_2x2(Fo.conv2d(x, self.conv3.weight, self.conv3.bias, relu=True))
-        x = x.reshape(x.shape[0], -1)
+        x = Fo.flatten2d(x)
         x = Fo.dropout(x, self.p_drop, self.training, self.rng)
         x = Fo.linear(x, self.fc1.weight, self.fc1.bias, relu=True)
         x = Fo.dropout(x, self.p_drop, self.training, self.rng)

@@ -142,6 +142,14 @@ void launch_maxpool2x2_fwd_bf16(const unsigned short*, unsigned short*,
 void launch_maxpool2x2_bwd_bf16(const unsigned short*, const uint8_t*,
                                 unsigned short*, long, int, int, int, int,
                                 int, void*);
+// flatten
+void launch_nhwc_flatten(const float*, float*, long, int, int, int, void*);
+void launch_nhwc_unflatten(const float*, float*, long, int, int, int,
+                           void*);
+void launch_nhwc_flatten_bf16(const unsigned short*, unsigned short*, long,
+                              int, int, int, void*);
+void launch_nhwc_unflatten_bf16(const unsigned short*, unsigned short*,
+                                long, int, int, int, void*);
 // batchnorm bf16
 void launch_bn_fwd_bf16(const unsigned short*, const float*, const float*,
                         float*, float*, float*, float*, unsigned short*,
@@ -379,6 +387,39 @@ void eval_update(torch::Tensor logits, torch::Tensor labels,
   launch_eval_update(logits.data_ptr<float>(), labels.data_ptr<int64_t>(),
                      conf.data_ptr<float>(), loss_sum.data_ptr<double>(), B,
                      C, num_classes, stream_of(logits));
+}
+
+// flatten: channels_last (B,C,H,W) -> (B, C*H*W) in CHW order
+torch::Tensor nhwc_flatten(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  x = cl(x);
+  long B = x.size(0);
+  int C = x.size(1), H = x.size(2), W = x.size(3);
+  auto out = torch::empty({B, (long)C * H * W}, x.options());
+  if (is_bf16(x))
+    launch_nhwc_flatten_bf16((const unsigned short*)x.data_ptr(),
+                             (unsigned short*)out.data_ptr(), B, C, H, W,
+                             stream_of(x));
+  else
+    launch_nhwc_flatten(x.data_ptr<float>(), out.data_ptr<float>(), B, C,
+                        H, W, stream_of(x));
+  return out;
+}
+
+torch::Tensor nhwc_unflatten(torch::Tensor g, int64_t C, int64_t H,
+                             int64_t W) {
+  TORCH_CHECK(g.is_cuda());
+  g = g.contiguous();
+  long B = g.size(0);
+  auto out = empty_cl({B, C, H, W}, g.options());
+  if (is_bf16(g))
+    launch_nhwc_unflatten_bf16((const unsigned short*)g.data_ptr(),
+                               (unsigned short*)out.data_ptr(), B, (int)C,
+                               (int)H, (int)W, stream_of(g));
+  else
+    launch_nhwc_unflatten(g.data_ptr<float>(), out.data_ptr<float>(), B,
+                          (int)C, (int)H, (int)W, stream_of(g));
+  return out;
 }
 
 // --------------------------------------------------------------- flatopt
@@ -906,6 +947,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &gemm);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("transpose2d", &transpose2d);
+  m.def("nhwc_flatten", &nhwc_flatten);
+  m.def("nhwc_unflatten", &nhwc_unflatten);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("conv2d_fwd", &conv2d_fwd);

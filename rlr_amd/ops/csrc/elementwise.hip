@@ -549,3 +549,68 @@ void launch_gap_bwd_bf16(const unsigned short* dy, unsigned short* dx,
                    (hipStream_t)s>>>(dy, dx, B, HW, C);
 }
 }
+
+// ---------------------------------------------- NHWC <-> flat CHW permute
+// The fc layers consume the conv trunk's channels_last output flattened in
+// the reference's CHW order (models.py:21,50 parity).  Torch's generic
+// permute kernel measured ~12.5 us for this shape; these are ~roofline.
+
+template <typename T>
+__global__ void nhwc_flatten_k(const T* __restrict__ in, T* __restrict__ out,
+                               long B, int C, int H, int W) {
+  long n = B * C * (long)H * W;
+  long stride = (long)gridDim.x * blockDim.x;
+  int HW = H * W;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < n;
+       o += stride) {
+    int w = o % W;
+    int h = (o / W) % H;
+    int c = (o / HW) % C;
+    long b = o / ((long)HW * C);
+    out[o] = in[((b * H + h) * (long)W + w) * C + c];
+  }
+}
+
+template <typename T>
+__global__ void nhwc_unflatten_k(const T* __restrict__ in,  // CHW flat
+                                 T* __restrict__ out,       // NHWC storage
+                                 long B, int C, int H, int W) {
+  long n = B * C * (long)H * W;
+  long stride = (long)gridDim.x * blockDim.x;
+  int HW = H * W;
+  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < n;
+       o += stride) {
+    // o indexes the NHWC output (coalesced writes)
+    int c = o % C;
+    long rest = o / C;
+    int w = rest % W;
+    int h = (rest / W) % H;
+    long b = rest / ((long)W * H);
+    out[o] = in[((b * C + c) * (long)H + h) * W + w];
+  }
+}
+
+extern "C" {
+void launch_nhwc_flatten(const float* in, float* out, long B, int C, int H,
+                         int W, void* s) {
+  nhwc_flatten_k<float><<<grid_for(B * C * (long)H * W), kBlock, 0,
+                          (hipStream_t)s>>>(in, out, B, C, H, W);
+}
+void launch_nhwc_unflatten(const float* in, float* out, long B, int C,
+                           int H, int W, void* s) {
+  nhwc_unflatten_k<float><<<grid_for(B * C * (long)H * W), kBlock, 0,
+                            (hipStream_t)s>>>(in, out, B, C, H, W);
+}
+void launch_nhwc_flatten_bf16(const unsigned short* in, unsigned short* out,
+                              long B, int C, int H, int W, void* s) {
+  nhwc_flatten_k<unsigned short><<<grid_for(B * C * (long)H * W), kBlock, 0,
+                                   (hipStream_t)s>>>(in, out, B, C, H, W);
+}
+void launch_nhwc_unflatten_bf16(const unsigned short* in,
+                                unsigned short* out, long B, int C, int H,
+                                int W, void* s) {
+  nhwc_unflatten_k<unsigned short>
+      <<<grid_for(B * C * (long)H * W), kBlock, 0, (hipStream_t)s>>>(
+          in, out, B, C, H, W);
+}
+}

@@ -209,6 +209,29 @@ def dropout(x, p, training, rng: DropoutCtx):
     return x * mask / (1.0 - p)
 
 
+# ---------------------------------------------------------------- flatten
+
+class _Flatten(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.chw = (x.shape[1], x.shape[2], x.shape[3])
+        return ext().nhwc_flatten(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        c, h, w = ctx.chw
+        return ext().nhwc_unflatten(dy.contiguous(), c, h, w)
+
+
+def flatten2d(x):
+    """(B,C,H,W) -> (B, C*H*W) flattened in the reference's CHW order
+    (models.py:21,50); on the GPU path this is one permute kernel off the
+    channels_last storage."""
+    if _gpu(x) and x.dim() == 4:
+        return _Flatten.apply(x)
+    return x.reshape(x.shape[0], -1)
+
+
 # ---------------------------------------------------------- cross entropy
 
 class _CrossEntropy(torch.autograd.Function):
