@@ -37,3 +37,12 @@ if [ "$WHAT" = "sweep" ]; then
   done
   grep -H -o '"value": [0-9.]*' gpurun_out/sweep_*.log
 fi
+
+if [ "$WHAT" = "fedemnist" ]; then
+  timeout 900 python -m rlr_amd.federated --data fedemnist --num_agents 3383 \
+    --agent_frac 0.01 --num_corrupt 338 --poison_frac 0.5 \
+    --robustLR_threshold 8 --local_ep 10 --bs 64 --rounds 5 --snap 5 \
+    --pattern_type square --synthetic --no_tb > gpurun_out/ci_fed.log 2>&1
+  echo "fedemnist rc=$?"
+  tail -8 gpurun_out/ci_fed.log
+fi
