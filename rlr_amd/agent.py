@@ -95,6 +95,8 @@ class Agent:
         engine = gm.get_engine(args) if use_graphs else None
         if engine is not None:
             engine.begin_round(theta0)
+        else:
+            gm.ensure_grad_views()
         for ep in range(args.local_ep):
             perm = np_rng(args.seed, 'shuffle', self.id, rnd, ep).permutation(n)
             perm_t = torch.as_tensor(perm, device=self.device)

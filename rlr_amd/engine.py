@@ -38,10 +38,18 @@ class TrainEngine:
 
     def _step_body(self, static_x, static_y):
         gm, args = self.gm, self.args
-        gm.flat_grads.zero_()
+        # adopt-and-gather: with p.grad detached, autograd ADOPTS each
+        # backward output (no zero + no per-param accumulate-add); one
+        # kernel then gathers the 8..62 grad tensors into flat_grads.
+        # Bitwise identical to zero_grad + accumulate (0 + g == g).
+        from .ops import ext
+        gm.clear_grads()
         out = gm(static_x)
         loss = Fo.cross_entropy(out, static_y)
         loss.backward()
+        ext().gather_grads(gm.flat_grads,
+                           [p.grad for p in gm.params],
+                           gm.param_offsets)
         flat_ops.clipped_sgd_step_(gm.flat_params, gm.flat_grads,
                                    gm.momentum, args.client_lr,
                                    args.client_moment, 10.0)

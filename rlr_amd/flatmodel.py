@@ -28,9 +28,12 @@ class FlatParamModel:
         self.flat_grads = torch.zeros(self.n_params, device=device)
         self.momentum = torch.zeros(self.n_params, device=device)
 
+        self.params = params
+        self.param_offsets = []
         offset = 0
         for p in params:
             n = p.numel()
+            self.param_offsets.append(offset)
             self.flat_params[offset:offset + n].copy_(p.data.reshape(-1))
             p.data = self.flat_params[offset:offset + n].view(p.shape)
             p.grad = self.flat_grads[offset:offset + n].view(p.shape)
@@ -68,6 +71,18 @@ class FlatParamModel:
 
     def zero_grad(self):
         self.flat_grads.zero_()
+
+    def ensure_grad_views(self):
+        """Re-point every p.grad at its flat_grads view (the graphed step
+        detaches them so autograd ADOPTS fresh grad tensors instead of
+        accumulating — engine.py gathers those into flat_grads in one
+        kernel)."""
+        for p, off in zip(self.params, self.param_offsets):
+            p.grad = self.flat_grads[off:off + p.numel()].view(p.shape)
+
+    def clear_grads(self):
+        for p in self.params:
+            p.grad = None
 
     def zero_momentum(self):
         self.momentum.zero_()

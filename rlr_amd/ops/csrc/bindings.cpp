@@ -64,6 +64,8 @@ void launch_clipped_sgd(float*, const float*, float*, float*, float, float,
                         float, long, void*);
 void launch_pgd_project(float*, const float*, float*, float, long, void*);
 void launch_delta64(const float*, const double*, double*, long, void*);
+void launch_gather_grads(const float* const*, const long*, const long*,
+                         int, float*, void*);
 // aggregation.hip
 void launch_fused_avg_rlr_apply(const double*, const double*, int, long,
                                 double, int, double, double, float*, double*,
@@ -440,6 +442,27 @@ void pgd_project(torch::Tensor p, torch::Tensor t0, double clip) {
   launch_pgd_project(p.data_ptr<float>(), t0.data_ptr<float>(),
                      scratch.data_ptr<float>(), (float)clip, p.numel(),
                      stream_of(p));
+}
+
+void gather_grads(torch::Tensor flat, std::vector<torch::Tensor> grads,
+                  std::vector<int64_t> offsets) {
+  CHK_CUDA(flat);
+  CHK(grads.size() == offsets.size());
+  size_t i = 0;
+  while (i < grads.size()) {
+    const float* srcs[16];
+    long offs[16], lens[16];
+    int cnt = 0;
+    for (; cnt < 16 && i < grads.size(); ++cnt, ++i) {
+      auto g = grads[i].contiguous();
+      TORCH_CHECK(g.scalar_type() == torch::kFloat);
+      srcs[cnt] = g.data_ptr<float>();
+      offs[cnt] = offsets[i];
+      lens[cnt] = g.numel();
+    }
+    launch_gather_grads(srcs, offs, lens, cnt, flat.data_ptr<float>(),
+                        stream_of(flat));
+  }
 }
 
 torch::Tensor delta64(torch::Tensor p, torch::Tensor t0) {
@@ -937,6 +960,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("clipped_sgd_step", &clipped_sgd_step);
   m.def("pgd_project", &pgd_project);
   m.def("delta64", &delta64);
+  m.def("gather_grads", &gather_grads);
   m.def("fused_avg_rlr_apply", &fused_avg_rlr_apply);
   m.def("rlr_vote", &rlr_vote);
   m.def("agg_avg", &agg_avg);

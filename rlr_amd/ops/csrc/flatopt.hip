@@ -128,3 +128,49 @@ void launch_delta64(const float* p, const double* t0, double* out, long n,
   delta64_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(p, t0, out, n);
 }
 }
+
+// ---- multi-segment gather: scattered autograd grad tensors -> the flat
+// grad buffer in ONE launch (replaces zero_grad + one accumulate-add per
+// parameter; pointers ride in kernargs so the kernel is hipGraph-stable).
+struct GatherSegs {
+  const float* src[16];
+  long off[16];
+  long len[16];
+  int count;
+};
+
+__global__ void gather_grads_k(GatherSegs segs, float* __restrict__ flat) {
+  long total = 0;
+#pragma unroll
+  for (int i = 0; i < 16; ++i)
+    if (i < segs.count) total += segs.len[i];
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    long rem = t;
+    int seg = 0;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) {
+      if (i < segs.count && rem >= segs.len[i]) {
+        rem -= segs.len[i];
+        seg = i + 1;
+      }
+    }
+    flat[segs.off[seg] + rem] = segs.src[seg][rem];
+  }
+}
+
+extern "C" void launch_gather_grads(const float* const* srcs,
+                                    const long* offs, const long* lens,
+                                    int count, float* flat, void* s) {
+  GatherSegs g;
+  g.count = count;
+  long total = 0;
+  for (int i = 0; i < count && i < 16; ++i) {
+    g.src[i] = srcs[i];
+    g.off[i] = offs[i];
+    g.len[i] = lens[i];
+    total += lens[i];
+  }
+  gather_grads_k<<<grid_for(total), kBlock, 0, (hipStream_t)s>>>(g, flat);
+}
