@@ -37,13 +37,13 @@ class TrainEngine:
     # ------------------------------------------------------------- step
 
     def _step_body(self, static_x, static_y):
-        gm, args = self.gm, self.args
-        # adopt-and-gather: with p.grad detached, autograd ADOPTS each
-        # backward output (no zero + no per-param accumulate-add); one
-        # kernel then gathers the 8..62 grad tensors into flat_grads.
+        # adopt-and-gather: with p.grad detached (cleared OUTSIDE the
+        # capture — freeing tensors mid-capture aborts), autograd ADOPTS
+        # each backward output (no zero + no per-param accumulate-add);
+        # one kernel gathers the 8..62 grad tensors into flat_grads.
         # Bitwise identical to zero_grad + accumulate (0 + g == g).
+        gm, args = self.gm, self.args
         from .ops import ext
-        gm.clear_grads()
         out = gm(static_x)
         loss = Fo.cross_entropy(out, static_y)
         loss.backward()
@@ -79,9 +79,13 @@ class TrainEngine:
         side.wait_stream(torch.cuda.current_stream(device))
         with torch.cuda.stream(side):
             for _ in range(2):
+                gm.clear_grads()
                 self._step_body(static_x, static_y)
                 rng.site = snap_site
         torch.cuda.current_stream(device).wait_stream(side)
+        # drop the warmup grads BEFORE capture opens: any allocator free
+        # that lands inside the capture window aborts it
+        gm.clear_grads()
 
         g = torch.cuda.CUDAGraph()
         # a GC cycle during capture frees pre-capture tensors -> hipFree
