@@ -83,9 +83,11 @@ class TrainEngine:
                 self._step_body(static_x, static_y)
                 rng.site = snap_site
         torch.cuda.current_stream(device).wait_stream(side)
-        # drop the warmup grads BEFORE capture opens: any allocator free
-        # that lands inside the capture window aborts it
+        # drop the warmup grads BEFORE capture opens (any allocator free
+        # inside the capture window aborts it) and settle the allocator's
+        # pending stream events with a full sync
         gm.clear_grads()
+        torch.cuda.synchronize(device)
 
         g = torch.cuda.CUDAGraph()
         # a GC cycle during capture frees pre-capture tensors -> hipFree
