@@ -37,19 +37,19 @@ class TrainEngine:
     # ------------------------------------------------------------- step
 
     def _step_body(self, static_x, static_y):
-        # adopt-and-gather: with p.grad detached (cleared OUTSIDE the
-        # capture — freeing tensors mid-capture aborts), autograd ADOPTS
-        # each backward output (no zero + no per-param accumulate-add);
-        # one kernel gathers the 8..62 grad tensors into flat_grads.
-        # Bitwise identical to zero_grad + accumulate (0 + g == g).
+        # NOTE on a rejected optimization: detaching p.grad so autograd
+        # ADOPTS fresh gradients (then gathering them with ONE kernel,
+        # ops ext.gather_grads) removes zero_grad + the 8..62 per-param
+        # accumulate-adds (~7% of a CNN step), but torch's allocator
+        # aborts the hipGraph capture when autograd allocates adopted
+        # grads inside the capture window on ROCm 7.0 — revisit with a
+        # newer torch.  The preset-view + accumulate scheme below is
+        # capture-stable.
         gm, args = self.gm, self.args
-        from .ops import ext
+        gm.flat_grads.zero_()
         out = gm(static_x)
         loss = Fo.cross_entropy(out, static_y)
         loss.backward()
-        ext().gather_grads(gm.flat_grads,
-                           [p.grad for p in gm.params],
-                           gm.param_offsets)
         flat_ops.clipped_sgd_step_(gm.flat_params, gm.flat_grads,
                                    gm.momentum, args.client_lr,
                                    args.client_moment, 10.0)
@@ -75,18 +75,14 @@ class TrainEngine:
         snap_state = rng.gpu_state(device).clone()
         snap_site = rng.site
 
+        gm.ensure_grad_views()
         side = torch.cuda.Stream(device=device)
         side.wait_stream(torch.cuda.current_stream(device))
         with torch.cuda.stream(side):
             for _ in range(2):
-                gm.clear_grads()
                 self._step_body(static_x, static_y)
                 rng.site = snap_site
         torch.cuda.current_stream(device).wait_stream(side)
-        # drop the warmup grads BEFORE capture opens (any allocator free
-        # inside the capture window aborts it) and settle the allocator's
-        # pending stream events with a full sync
-        gm.clear_grads()
         torch.cuda.synchronize(device)
 
         g = torch.cuda.CUDAGraph()
