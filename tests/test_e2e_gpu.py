@@ -112,3 +112,16 @@ def test_defense_semantics_gpu():
         poisons[thr] = h['poison_acc']
     assert max(poisons[0][-3:]) > 0.4, poisons
     assert max(poisons[5][-2:]) < 0.2, poisons
+
+
+def test_gpu_resume_bitwise(tmp_path):
+    """checkpoint -> resume on GPU (graphs + streams on) is bitwise equal
+    to the uninterrupted run."""
+    from rlr_amd.federated import run
+    import os
+    h_full = run(_args(rounds=4, snap=2, ckpt_dir=str(tmp_path / 'a')))
+    run(_args(rounds=2, snap=2, ckpt_dir=str(tmp_path / 'b')))
+    ck = os.path.join(str(tmp_path / 'b'), 'round_000002.pt')
+    h_res = run(_args(rounds=4, snap=2, resume=ck,
+                      ckpt_dir=str(tmp_path / 'c')))
+    assert torch.equal(h_full['final_params'], h_res['final_params'])
