@@ -46,3 +46,12 @@ if [ "$WHAT" = "fedemnist" ]; then
   echo "fedemnist rc=$?"
   tail -8 gpurun_out/ci_fed.log
 fi
+
+if [ "$WHAT" = "final" ]; then
+  timeout 400 python bench.py --steps 20 --warmup 5 > gpurun_out/final_bench.log 2>&1
+  grep -o '"value": [0-9.]*\|"ms_per_step": [0-9.]*' gpurun_out/final_bench.log
+  cd /tmp && export TMPDIR=/tmp
+  timeout 400 rocprofv3 --kernel-trace --stats -d $GRAFT_REPO_ROOT/gpurun_out/proff -o bf \
+    -- python $GRAFT_REPO_ROOT/bench.py --steps 3 --warmup 2 > /dev/null 2>&1
+  echo "profile rc=$?"
+fi
