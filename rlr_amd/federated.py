@@ -17,7 +17,6 @@ Per round (reference federated.py:65-92):
 """
 
 import copy
-import math
 import os
 import time
 

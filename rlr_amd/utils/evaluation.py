@@ -10,7 +10,6 @@ kernel per batch (ops/csrc: eval_update)."""
 import torch
 
 from ..ops import ext, force_eager
-from ..ops import functional as Fo
 
 
 @torch.no_grad()
