@@ -55,3 +55,14 @@ if [ "$WHAT" = "final" ]; then
     -- python $GRAFT_REPO_ROOT/bench.py --steps 3 --warmup 2 > /dev/null 2>&1
   echo "profile rc=$?"
 fi
+
+if [ "$WHAT" = "dba" ]; then
+  # BASELINE config 4 semantics on ONE GPU (the driver scales it to 8):
+  # CIFAR10, 40 agents, 4 corrupt with the DBA plus-pattern split, RLR 8
+  timeout 600 python -m rlr_amd.federated --data cifar10 --num_agents 40 \
+    --num_corrupt 4 --poison_frac 0.5 --pattern_type plus \
+    --robustLR_threshold 8 --rounds 3 --snap 3 --synthetic --no_tb \
+    > gpurun_out/ci_dba.log 2>&1
+  echo "dba rc=$?"
+  tail -5 gpurun_out/ci_dba.log
+fi
