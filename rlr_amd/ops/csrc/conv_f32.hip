@@ -943,8 +943,9 @@ void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
 
 int conv_bwd_weight_splitk(int Kout, int Ncrs, long Kdim) {
   long tiles = ((Kout + 63) / 64) * (long)((Ncrs + BN - 1) / BN);
-  if (tiles >= 192 || Kdim <= 2 * BK) return 1;
-  long want = (512 + tiles - 1) / tiles;  // 2 blocks/CU for latency hiding
+  if (tiles >= 768 || Kdim <= 2 * BK) return 1;
+  // the 64x64 tile runs 4 blocks/CU (35 KB LDS): target ~1024 resident
+  long want = (1024 + tiles - 1) / tiles;
   long max_chunks = (Kdim + BK - 1) / BK;
   long sk = want < max_chunks ? want : max_chunks;
   return (int)(sk < 1 ? 1 : (sk > 256 ? 256 : sk));
