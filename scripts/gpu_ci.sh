@@ -66,3 +66,12 @@ if [ "$WHAT" = "dba" ]; then
   echo "dba rc=$?"
   tail -5 gpurun_out/ci_dba.log
 fi
+
+if [ "$WHAT" = "pmc" ]; then
+  cd /tmp && export TMPDIR=/tmp
+  timeout 300 rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY SQ_ACTIVE_INST_ANY SQ_LDS_BANK_CONFLICT SQ_INSTS_VALU SQ_INSTS_MFMA SQ_INSTS_LDS \
+    -d $GRAFT_REPO_ROOT/gpurun_out/pmc3 -o p3 --output-format csv \
+    -- python $GRAFT_REPO_ROOT/tests/perf/conv_micro.py all 15 \
+    > $GRAFT_REPO_ROOT/gpurun_out/pmc3.log 2>&1
+  echo "pmc rc=$?"
+fi
