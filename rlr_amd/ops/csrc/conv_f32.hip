@@ -208,6 +208,154 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
     }
 }
 
+// -------------------------------------------- fwd, x-resident variant
+// For stride-1 pad-0 C%32==0 layers whose image has >= 128 output pixels
+// (conv2-class): the block's whole input region is loaded into LDS ONCE
+// (instead of re-gathering each patch element per K-tile — a 9x re-read
+// through L3) and A fragments are read directly with a per-pixel +1 bank
+// pad.  One image per block; B (weights) stays double-buffer streamed.
+__global__ __launch_bounds__(256)
+void conv_fwd_res_k(const float* __restrict__ x,
+                    const float* __restrict__ wt,  // [(r,s,c)][KO]
+                    const float* __restrict__ bias, float* __restrict__ y,
+                    ConvShape sh, int Kdim, int relu, int ptiles,
+                    int lds_rows) {
+  constexpr int BM = 128, MI = 4, NI = 2;
+  extern __shared__ __attribute__((aligned(16))) float dynLds[];
+  float* x_lds = dynLds;                       // [lds_rows][W][C+1-padded]
+  // B buffers start 16-B aligned after the (odd-strided) x region
+  long xf = ((long)lds_rows * sh.W * (sh.C + 1) + 3) & ~3L;
+  float* B_lds0 = dynLds + xf;
+  // B double buffer after the x region
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  f32x4 acc[MI][NI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nb = blockIdx.x / ptiles;
+  const int p0 = (blockIdx.x % ptiles) * BM;
+  const int OHOW = sh.OH * sh.OW;
+  const int n_blk = blockIdx.y * BN;
+  const int Cp = sh.C + 1;
+
+  const int oh_lo = p0 / sh.OW;
+  // ---- load the x region once (always in-bounds: pad==0, stride==1) ----
+  {
+    const float* xsrc =
+        x + ((long)nb * sh.H + oh_lo) * sh.W * sh.C;
+    int total4 = lds_rows * sh.W * sh.C / 4;
+    int avail_rows = sh.H - oh_lo;
+    int total4_avail = min(total4, avail_rows * sh.W * sh.C / 4);
+    for (int i = t; i < total4_avail; i += 256) {
+      float4 q = *(const float4*)(xsrc + (long)i * 4);
+      int c = (i * 4) % sh.C;
+      int pix = (i * 4) / sh.C;
+      float* dst = &x_lds[pix * Cp + c];
+      dst[0] = q.x; dst[1] = q.y; dst[2] = q.z; dst[3] = q.w;
+    }
+  }
+
+  // per-lane fragment bases: lds pixel offset of each mi's output pixel
+  int pixrel[MI];
+  bool mval[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    int m = p0 + wr * 64 + mi * 16 + l15;
+    mval[mi] = m < OHOW;
+    int mm = mval[mi] ? m : 0;
+    int oh = mm / sh.OW, ow = mm % sh.OW;
+    pixrel[mi] = ((oh - oh_lo) * sh.W + ow) * Cp;
+  }
+
+  // ---- B staging (T14 split, double-buffered) ----
+  const int bk = t >> 4, bn = (t & 15) * 4;
+  float4 rb[2];
+  auto stage_loadB = [&](int k0) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int gk = k0 + bk + j * 16;
+      float4 q = {0.f, 0.f, 0.f, 0.f};
+      if (gk < Kdim) {
+        const float* srcp = wt + (long)gk * sh.Kout + n_blk + bn;
+        if (n_blk + bn + 3 < sh.Kout && (sh.Kout % 4) == 0)
+          q = *(const float4*)srcp;
+        else {
+          if (n_blk + bn + 0 < sh.Kout) q.x = srcp[0];
+          if (n_blk + bn + 1 < sh.Kout) q.y = srcp[1];
+          if (n_blk + bn + 2 < sh.Kout) q.z = srcp[2];
+          if (n_blk + bn + 3 < sh.Kout) q.w = srcp[3];
+        }
+      }
+      rb[j] = q;
+    }
+  };
+  auto stage_writeB = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      *(float4*)&B_lds0[buf * BK * LDB_S + (bk + j * 16) * LDB_S + bn] =
+          rb[j];
+  };
+
+  stage_loadB(0);
+  stage_writeB(0);
+  if (BK < Kdim) stage_loadB(BK);
+  __syncthreads();
+  int buf = 0;
+  for (int k0 = 0; k0 < Kdim; k0 += BK) {
+    if (k0 + BK < Kdim) {
+      stage_writeB(buf ^ 1);
+      if (k0 + 2 * BK < Kdim) stage_loadB(k0 + 2 * BK);
+    }
+    // one tap per 32-wide K-tile
+    int rs = k0 / sh.C;
+    int r = rs / sh.S, s = rs % sh.S;
+    int c0 = k0 - rs * sh.C;
+    int tap = (r * sh.W + s) * Cp + c0 + l4;
+    const float* Bbuf = &B_lds0[buf * BK * LDB_S];
+#pragma unroll
+    for (int kk = 0; kk < BK / 4; ++kk) {
+      float a_frag[MI], b_frag[NI];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        a_frag[mi] = x_lds[pixrel[mi] + tap + kk * 4];
+#pragma unroll
+      for (int ni = 0; ni < NI; ++ni)
+        b_frag[ni] =
+            Bbuf[(kk * 4 + l4) * LDB_S + wc * 32 + ni * 16 + l15];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NI; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NI; ++ni) {
+      int ko = n_blk + wc * 32 + ni * 16 + l15;
+      if (ko >= sh.Kout) continue;
+#pragma unroll
+      for (int r2 = 0; r2 < 4; ++r2) {
+        int m = p0 + wr * 64 + mi * 16 + l4 * 4 + r2;
+        if (m >= OHOW) continue;
+        float v = acc[mi][ni][r2];
+        if (bias) v += bias[ko];
+        if (relu) v = fmaxf(v, 0.f);
+        y[((long)nb * OHOW + m) * sh.Kout + ko] = v;
+      }
+    }
+}
+
 // -------------------------------------------------------------- bwd-data
 
 // Kdim order (r,s,ko); V4: KO % 32 == 0.  ST: compile-time stride.
@@ -891,6 +1039,20 @@ void launch_conv_fwd(const float* x, const float* wt, const float* bias,
     return;
   }
   bool v4 = (C % 32) == 0;
+  if (pad == 0 && stride == 1 && v4 && OH * OW >= 128) {
+    // x-resident fast path: LDS = x region + 2 B buffers
+    int ptiles = (OH * OW + 127) / 128;
+    int oh_span = (127 / OW) + 2;               // output rows per tile (+1)
+    int lds_rows = oh_span + R - 1;
+    long xf = ((long)lds_rows * W * (C + 1) + 3) & ~3L;
+    long lds_bytes = (xf + 2L * BK * LDB_S) * 4;
+    if (lds_bytes <= 100 * 1024) {
+      dim3 g2((long)Nb * ptiles, (Kout + BN - 1) / BN, 1);
+      conv_fwd_res_k<<<g2, 256, lds_bytes, st>>>(x, wt, bias, y, sh, Kdim,
+                                                 relu, ptiles, lds_rows);
+      return;
+    }
+  }
   if (pad == 0 && v4)
     conv_fwd_k<true, true><<<grid, 256, 0, st>>>(x, wt, bias, y, sh, Kdim,
                                                  relu);
