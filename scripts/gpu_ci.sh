@@ -75,3 +75,18 @@ if [ "$WHAT" = "pmc" ]; then
     > $GRAFT_REPO_ROOT/gpurun_out/pmc3.log 2>&1
   echo "pmc rc=$?"
 fi
+
+if [ "$WHAT" = "curves" ]; then
+  # the reference's FMNIST triple at full scale (runner.sh:12-18):
+  # no-attack / attack / attack+RLR(theta=4), 10 agents, 200 rounds
+  for CFG in "na:--num_corrupt 0 --poison_frac 0" \
+             "atk:--num_corrupt 1 --poison_frac 0.5" \
+             "rlr:--num_corrupt 1 --poison_frac 0.5 --robustLR_threshold 4"; do
+    TAG="${CFG%%:*}"; FLAGS="${CFG#*:}"
+    timeout 500 python -m rlr_amd.federated --data fmnist --num_agents 10 \
+      --rounds 200 --snap 50 --synthetic --no_tb $FLAGS \
+      > gpurun_out/curve_$TAG.log 2>&1
+    echo "== $TAG rc=$?"
+    grep -E "Val_Loss|Poison Loss" gpurun_out/curve_$TAG.log | tail -4
+  done
+fi
