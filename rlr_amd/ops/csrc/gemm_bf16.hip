@@ -26,18 +26,6 @@ constexpr int BM16 = 128, BN16 = 64, BK16 = 32;
 constexpr int LDA16 = BK16 + 8;   // bf16 elements per A_lds row
 constexpr int LDB16 = BK16 + 8;   // bf16 elements per B_lds row (B^T image)
 
-__device__ __forceinline__ float bf2f(unsigned short u) {
-  union { unsigned int i; float f; } v;
-  v.i = ((unsigned int)u) << 16;
-  return v.f;
-}
-__device__ __forceinline__ unsigned short f2bf(float f) {
-  union { float f; unsigned int i; } v;
-  v.f = f;
-  unsigned int rounded = v.i + 0x7FFF + ((v.i >> 16) & 1);  // RNE
-  return (unsigned short)(rounded >> 16);
-}
-
 // A and B both staged as [row][k] bf16 (B transposed at stage time), so a
 // fragment read is ONE 16-byte load of 8 contiguous bf16.
 template <bool VEC>
@@ -168,7 +156,7 @@ void gemm_bf16_k(const unsigned short* __restrict__ A,
           if (bias) v += bias[col];
           if (relu) v = fmaxf(v, 0.f);
           if (C16)
-            C16[(long)row * ldc + col] = f2bf(v);
+            C16[(long)row * ldc + col] = f2bf_(v);
           else
             C[(long)row * ldc + col] = v;
         } else {
@@ -184,7 +172,7 @@ __global__ void f32_to_bf16_k(const float* __restrict__ in,
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride)
-    out[i] = f2bf(in[i]);
+    out[i] = f2bf_(in[i]);
 }
 
 __global__ void bf16_to_f32_k(const unsigned short* __restrict__ in,
@@ -192,7 +180,7 @@ __global__ void bf16_to_f32_k(const unsigned short* __restrict__ in,
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride)
-    out[i] = bf2f(in[i]);
+    out[i] = bf2f_(in[i]);
 }
 
 extern "C" {
