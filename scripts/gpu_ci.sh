@@ -67,6 +67,17 @@ if [ "$WHAT" = "dba" ]; then
   tail -5 gpurun_out/ci_dba.log
 fi
 
+if [ "$WHAT" = "profr" ]; then
+  # kernel-time breakdown of the bf16 ResNet18 path (BASELINE config 3) —
+  # evidence for the round-2 optimization roadmap
+  cd /tmp && export TMPDIR=/tmp
+  timeout 500 rocprofv3 --kernel-trace --stats -d $GRAFT_REPO_ROOT/gpurun_out/profr -o rn \
+    -- python $GRAFT_REPO_ROOT/bench.py --data cifar10 --model resnet18 \
+    --dtype bf16 --steps 2 --warmup 1 > $GRAFT_REPO_ROOT/gpurun_out/profr.log 2>&1
+  echo "profr rc=$?"
+  grep -o '"ms_per_step": [0-9.]*' $GRAFT_REPO_ROOT/gpurun_out/profr.log
+fi
+
 if [ "$WHAT" = "pmc" ]; then
   cd /tmp && export TMPDIR=/tmp
   timeout 300 rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY SQ_ACTIVE_INST_ANY SQ_LDS_BANK_CONFLICT SQ_INSTS_VALU SQ_INSTS_MFMA SQ_INSTS_LDS \
