@@ -4,20 +4,33 @@
 // two-stage column sums (chunk partials -> combine), deterministic, no
 // atomics.  Elementwise passes use a stride-multiple-of-C grid so each
 // thread's channel is computed ONCE (launcher rounds the grid).
+//
+// Stage-1 parallelism (rocprofv3 evidence, profiles/): 512 row chunks, and
+// when C < 256 a block carries 256/C chunk SLICES so all four waves work —
+// consecutive lanes read consecutive channels (coalesced 128 B per wave
+// for bf16 at C>=64).  The first-cut 64-chunk/one-wave version left 3/4 of
+// each block idle and only 64 blocks on a 256-CU chip: 322 us for a 33 MB
+// pass that bn_norm_k covers in 8 us.
 #include "common.h"
 
-constexpr int kBnChunks = 64;
+constexpr int kBnChunks = 512;  // total row chunks (stage-1 partials)
 
 // ---- stage 1: partials[chunk][2C] = (sum x, sum x^2) over a row chunk ----
+// grid: x = kBnChunks / slices_per_block, y = ceil(C / C_blk);
+// thread -> (slice = tid / C_blk, c = y*C_blk + tid % C_blk).
 template <typename T>
 __global__ void bn_stats1_k(const T* __restrict__ x,
-                            float* __restrict__ partials, long M, int C) {
-  int chunk = blockIdx.x;
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                            float* __restrict__ partials, long M, int C,
+                            int C_blk) {
+  int sub_per = blockDim.x / C_blk;
+  int c = blockIdx.y * C_blk + threadIdx.x % C_blk;
+  int sub = threadIdx.x / C_blk;
+  int chunk = blockIdx.x * sub_per + sub;
+  if (c >= C || chunk >= kBnChunks) return;
   long per = (M + kBnChunks - 1) / kBnChunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float s = 0.f, ss = 0.f;
+#pragma unroll 4
   for (long m = lo; m < hi; ++m) {
     float v = ldv(&x[m * C + c]);
     s += v;
@@ -27,17 +40,40 @@ __global__ void bn_stats1_k(const T* __restrict__ x,
   partials[((long)chunk * 2 + 1) * C + c] = ss;
 }
 
+// ---- stage 2: sums[c], sums[C+c] = fixed-order combine of the chunk
+// partials.  Each of the block's 256/C_blk slices sums a strided subset,
+// then the slices are combined in LDS in fixed slice order (deterministic).
 __global__ void bn_stats2_k(const float* __restrict__ partials,
-                            float* __restrict__ sums, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                            float* __restrict__ sums, int C, int C_blk,
+                            float* __restrict__ copy0,
+                            float* __restrict__ copy1) {
+  __shared__ float l_s[256], l_ss[256];
+  int sub_per = blockDim.x / C_blk;
+  int lc = threadIdx.x % C_blk;
+  int c = blockIdx.x * C_blk + lc;
+  int sub = threadIdx.x / C_blk;
   float s = 0.f, ss = 0.f;
-  for (int ch = 0; ch < kBnChunks; ++ch) {
-    s += partials[((long)ch * 2) * C + c];
-    ss += partials[((long)ch * 2 + 1) * C + c];
+  if (c < C) {
+    for (int ch = sub; ch < kBnChunks; ch += sub_per) {
+      s += partials[((long)ch * 2) * C + c];
+      ss += partials[((long)ch * 2 + 1) * C + c];
+    }
   }
-  sums[c] = s;
-  sums[C + c] = ss;
+  l_s[threadIdx.x] = s;
+  l_ss[threadIdx.x] = ss;
+  __syncthreads();
+  if (sub == 0 && c < C) {
+    for (int j = 1; j < sub_per; ++j) {
+      s += l_s[j * C_blk + lc];
+      ss += l_ss[j * C_blk + lc];
+    }
+    sums[c] = s;
+    sums[C + c] = ss;
+    // bwd: db = sum(dy), dw = sum(dy*xhat) — written here instead of two
+    // D2D copies after the fact (12k extra launches per bench, profiles/)
+    if (copy0) copy0[c] = s;
+    if (copy1) copy1[c] = ss;
+  }
 }
 
 // ---- finalize mean/rstd + running stats (torch: running_var unbiased) ----
@@ -102,15 +138,18 @@ __global__ void bn_bwd_stats1_k(const T* __restrict__ x,
                                 const T* __restrict__ dy,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ rstd,
-                                float* __restrict__ partials, long M,
-                                int C) {
-  int chunk = blockIdx.x;
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                float* __restrict__ partials, long M, int C,
+                                int C_blk) {
+  int sub_per = blockDim.x / C_blk;
+  int c = blockIdx.y * C_blk + threadIdx.x % C_blk;
+  int sub = threadIdx.x / C_blk;
+  int chunk = blockIdx.x * sub_per + sub;
+  if (c >= C || chunk >= kBnChunks) return;
   long per = (M + kBnChunks - 1) / kBnChunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float m_ = mean[c], rs = rstd[c];
   float sdy = 0.f, sdyx = 0.f;
+#pragma unroll 4
   for (long m = lo; m < hi; ++m) {
     float g = ldv(&dy[m * C + c]);
     sdy += g;
@@ -167,6 +206,14 @@ static int bn_grid(long n, int C) {
   return (int)(want > 0 ? want : 1);
 }
 
+// channels per block for the stats kernels: all of C up to the block size,
+// and a divisor of 256 so slices tile the block exactly (non-power-of-two
+// C falls back to one slice per block).
+static int bn_cblk(int C) {
+  if (C >= 256) return 256;
+  return (256 % C == 0) ? C : 256;
+}
+
 template <typename T>
 static void bn_fwd_impl(const T* x, const float* w, const float* b,
                         float* running_mean, float* running_var,
@@ -178,10 +225,12 @@ static void bn_fwd_impl(const T* x, const float* w, const float* b,
   float* partials = scratch;
   float* sums = scratch + (long)kBnChunks * 2 * C;
   if (training) {
-    dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
-    bn_stats1_k<T><<<g1, kBlock, 0, st>>>(x, partials, M, C);
-    bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials,
-                                                              sums, C);
+    int C_blk = bn_cblk(C);
+    int sub_per = 256 / C_blk;
+    dim3 g1(kBnChunks / sub_per, (C + C_blk - 1) / C_blk);
+    bn_stats1_k<T><<<g1, 256, 0, st>>>(x, partials, M, C, C_blk);
+    bn_stats2_k<<<(C + C_blk - 1) / C_blk, 256, 0, st>>>(
+        partials, sums, C, C_blk, nullptr, nullptr);
     bn_finalize_k<<<(C + 255) / 256, 256, 0, st>>>(
         sums, save_mean, save_rstd, running_mean, running_var, C, M,
         momentum, eps);
@@ -202,22 +251,22 @@ static void bn_bwd_impl(const T* x, const T* dy, const float* w,
   long M = (long)Nb * HW;
   float* partials = scratch;
   float* stats = scratch + (long)kBnChunks * 2 * C;
-  dim3 g1(kBnChunks, (C + kBlock - 1) / kBlock);
-  bn_bwd_stats1_k<T><<<g1, kBlock, 0, st>>>(x, dy, save_mean, save_rstd,
-                                            partials, M, C);
-  bn_stats2_k<<<(C + kBlock - 1) / kBlock, kBlock, 0, st>>>(partials, stats,
-                                                            C);
+  int C_blk = bn_cblk(C);
+  int sub_per = 256 / C_blk;
+  dim3 g1(kBnChunks / sub_per, (C + C_blk - 1) / C_blk);
+  bn_bwd_stats1_k<T><<<g1, 256, 0, st>>>(x, dy, save_mean, save_rstd,
+                                         partials, M, C, C_blk);
+  bn_stats2_k<<<(C + C_blk - 1) / C_blk, 256, 0, st>>>(partials, stats, C,
+                                                       C_blk, db, dw);
   long n = M * C;
   bn_bwd_dx_k<T><<<bn_grid(n, C), kBlock, 0, st>>>(
       x, dy, w, save_mean, save_rstd, stats, dx, n, C,
       training ? 1.f / (float)M : 0.f);
-  HIP_CHECK(hipMemcpyAsync(db, stats, C * sizeof(float),
-                           hipMemcpyDeviceToDevice, st));
-  HIP_CHECK(hipMemcpyAsync(dw, stats + C, C * sizeof(float),
-                           hipMemcpyDeviceToDevice, st));
 }
 
 extern "C" {
+int bn_scratch_floats(int C) { return (kBnChunks * 2 + 2) * C; }
+
 void launch_bn_fwd(const float* x, const float* w, const float* b,
                    float* running_mean, float* running_var, float* save_mean,
                    float* save_rstd, float* y, float* scratch, int Nb,

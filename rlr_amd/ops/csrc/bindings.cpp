@@ -161,6 +161,7 @@ void launch_bn_bwd_bf16(const unsigned short*, const unsigned short*,
                         unsigned short*, float*, float*, int, int, int, int,
                         void*);
 // batchnorm.hip
+int bn_scratch_floats(int);
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, int, int, int, float,
                    float, int, void*);
@@ -832,7 +833,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                     x.options());
   auto save_mean = torch::empty({C}, x.options().dtype(torch::kFloat));
   auto save_rstd = torch::empty({C}, x.options().dtype(torch::kFloat));
-  auto scratch = torch::empty({(64 * 2 + 2) * (long)C},
+  auto scratch = torch::empty({(long)bn_scratch_floats(C)},
                               x.options().dtype(torch::kFloat));
   if (is_bf16(x))
     launch_bn_fwd_bf16((const unsigned short*)x.data_ptr(),
@@ -867,7 +868,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
                      x.options());
   auto dw = torch::empty({C}, x.options().dtype(torch::kFloat));
   auto db = torch::empty({C}, x.options().dtype(torch::kFloat));
-  auto scratch = torch::empty({(64 * 2 + 2) * (long)C},
+  auto scratch = torch::empty({(long)bn_scratch_floats(C)},
                               x.options().dtype(torch::kFloat));
   if (is_bf16(x))
     launch_bn_bwd_bf16((const unsigned short*)x.data_ptr(),
