@@ -18,7 +18,22 @@ static void* stream_of(const torch::Tensor& t) {
 }
 
 // 4-D activations are channels_last (NHWC storage) on the HIP path.
-static torch::Tensor cl(torch::Tensor t) {
+// RLR_AMD_DEBUG_LAYOUT=1 prints every call that actually COPIES (a copy
+// here means some producer broke the channels_last chain — hot-path bug).
+static bool layout_debug() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("RLR_AMD_DEBUG_LAYOUT");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+static torch::Tensor cl(torch::Tensor t, const char* site = "?") {
+  if (layout_debug() && t.dim() == 4 &&
+      !t.is_contiguous(torch::MemoryFormat::ChannelsLast))
+    fprintf(stderr, "[layout] cl(%s) copy %ldx%ldx%ldx%ld dtype=%d\n",
+            site, (long)t.size(0), (long)t.size(1), (long)t.size(2),
+            (long)t.size(3), (int)t.scalar_type());
   return t.contiguous(torch::MemoryFormat::ChannelsLast);
 }
 static torch::Tensor empty_cl(std::vector<int64_t> sizes,
@@ -238,7 +253,7 @@ torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
 
 std::tuple<torch::Tensor, torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4);
-  x = cl(x);
+  x = cl(x, "mp_fwd.x");
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int OH = H / 2, OW = W / 2;
   auto y = empty_cl({Nb, C, OH, OW}, x.options());
@@ -258,7 +273,7 @@ std::tuple<torch::Tensor, torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
                              std::vector<int64_t> in_shape) {
   TORCH_CHECK(dy.is_cuda());
-  dy = cl(dy);
+  dy = cl(dy, "mp_bwd.dy");
   int Nb = in_shape[0], C = in_shape[1], H = in_shape[2], W = in_shape[3];
   int OH = dy.size(2), OW = dy.size(3);
   auto dx = empty_cl({Nb, C, H, W}, dy.options());
@@ -327,7 +342,7 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
 
 torch::Tensor gap_fwd(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda());
-  x = cl(x);
+  x = cl(x, "mpB.x");
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = torch::empty({Nb, C}, x.options());
   if (is_bf16(x))
@@ -395,7 +410,7 @@ void eval_update(torch::Tensor logits, torch::Tensor labels,
 // flatten: channels_last (B,C,H,W) -> (B, C*H*W) in CHW order
 torch::Tensor nhwc_flatten(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4);
-  x = cl(x);
+  x = cl(x, "gap.x");
   long B = x.size(0);
   int C = x.size(1), H = x.size(2), W = x.size(3);
   auto out = torch::empty({B, (long)C * H * W}, x.options());
@@ -684,7 +699,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> b, int64_t stride,
                          int64_t pad, bool relu) {
   TORCH_CHECK(x.is_cuda() && w.is_cuda());
-  x = cl(x);
+  x = cl(x, "conv_fwd.x");
   w = w.contiguous();
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
@@ -728,9 +743,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor dy, int64_t stride,
     int64_t pad, bool has_b, bool need_dx) {
   TORCH_CHECK(x.is_cuda());
-  x = cl(x);
+  x = cl(x, "conv_bwd.x");
   w = w.contiguous();
-  dy = cl(dy);
+  dy = cl(dy, "conv_bwd.dy");
   int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int Kout = w.size(0), R = w.size(2), S = w.size(3);
   int OH = dy.size(2), OW = dy.size(3);
@@ -827,7 +842,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
     double eps, bool training) {
   TORCH_CHECK(x.is_cuda());
-  x = cl(x);
+  x = cl(x, "bn_fwd.x");
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto y = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
                     x.options());
@@ -861,8 +876,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor save_mean,
     torch::Tensor save_rstd, torch::Tensor dy) {
   TORCH_CHECK(x.is_cuda());
-  x = cl(x);
-  dy = cl(dy);
+  x = cl(x, "bn_bwd.x");
+  dy = cl(dy, "bn_bwd.dy");
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto dx = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
                      x.options());

@@ -170,7 +170,9 @@ class _MaxPool2x2(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (idx,) = ctx.saved_tensors
-        return ext().maxpool2x2_bwd(dy.contiguous(), idx, list(ctx.in_shape))
+        # no .contiguous() here: plain contiguous() would force NCHW and
+        # the binding's cl() would copy straight back to channels_last
+        return ext().maxpool2x2_bwd(dy, idx, list(ctx.in_shape))
 
 
 def max_pool2d_2x2(x):
@@ -271,8 +273,10 @@ class _BatchNorm(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, save_mean, save_rstd = ctx.saved_tensors
-        dx, dw, db = ext().batchnorm_bwd(x, w, save_mean, save_rstd,
-                                         dy.contiguous())
+        # dy is normalized to channels_last inside the binding; a plain
+        # .contiguous() here would bounce it through NCHW (2 extra copies
+        # per BN — rocprofv3 evidence in profiles/)
+        dx, dw, db = ext().batchnorm_bwd(x, w, save_mean, save_rstd, dy)
         return dx, dw, db, None, None, None, None, None
 
 
