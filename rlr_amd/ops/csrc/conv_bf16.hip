@@ -289,16 +289,18 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
 
 // ------------------------------------------------------------ bwd-weight
 
-// 64(ko) x 64(crs) x BK=32(m); A = dy^T, B = x-patch^T — both staged
+// 64(ko) x 128(crs) x BK=32(m); A = dy^T, B = x-patch^T — both staged
 // transposed for contiguous 8-element m fragments.  fp32 slabs/output
 // (weight grads stay fp32); reuses the f32 split-K reduce + dw permute.
+// The 128-wide crs tile doubles MFMA work per barrier (2x4 fragments,
+// 8 MFMA/wave/k-step) and halves dy re-reads vs the first-cut 64x64.
 template <bool P0>
 __global__ __launch_bounds__(256)
 void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
                             const unsigned short* __restrict__ x,
                             float* __restrict__ out, ConvShapeB sh,
                             int Ncrs, long k_per_chunk, int direct_out) {
-  constexpr int BM = 64, BN = 64, MI = 2, NI = 2;
+  constexpr int BM = 64, BN = 128, MI = 2, NI = 4;
   __shared__ unsigned short A_lds[2][BM * LDA_B];  // [ko][m]
   __shared__ unsigned short B_lds[2][BN * LDT_B];  // [crs][m]
   const int t = threadIdx.x;
@@ -319,8 +321,9 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
 
   // A: thread loads dy[m][ko8..ko8+7]; ko8 = (t&7)*8, m = t>>3 (32 m)
   const int ako = (t & 7) * 8, amr = t >> 3;
-  // B: thread loads x-patch[m][crs8..+7] (8 contiguous c); same split
-  const int bcr = (t & 7) * 8, bmr = t >> 3;
+  // B: thread loads x-patch[m][crs8..+7] (8 contiguous c) for TWO m rows
+  // (128 crs x 32 m needs 512 b128 loads from 256 threads)
+  const int bcr = (t & 15) * 8, bmr = t >> 4;
 
   // (r,s,c0) for this thread's B columns (fixed)
   int br_, bs_, bc0_;
@@ -332,7 +335,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     bc0_ = crs - rs * sh.C;
   }
 
-  unsigned short raA[8], rbB[8];
+  unsigned short raA[8], rbB[2][8];
   auto stage_load = [&](long k0) {
     {
       long k = k0 + amr;
@@ -349,8 +352,9 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
       }
       *(bf16x8*)raA = q;
     }
-    {
-      long k = k0 + bmr;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      long k = k0 + bmr + j * 16;
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
       if (k < k_hi) {
         int ow = k % sh.OW;
@@ -363,7 +367,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
           q = *(const bf16x8*)(x + (nb * sh.H * sh.W + (long)ih * sh.W +
                                     iw) * sh.C + bc0_);
       }
-      *(bf16x8*)rbB = q;
+      *(bf16x8*)rbB[j] = q;
     }
   };
   auto stage_write = [&](int buf) {
@@ -372,8 +376,10 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     for (int e = 0; e < 8; ++e)
       A_lds[buf][(ako + e) * LDA_B + m] = raA[e];
 #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      B_lds[buf][(bcr + e) * LDT_B + bmr] = rbB[e];
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        B_lds[buf][(bcr + e) * LDT_B + bmr + j * 16] = rbB[j][e];
   };
 
   stage_load(k_lo);
