@@ -108,6 +108,7 @@ void launch_conv_fwd(const float*, const float*, const float*, float*, int,
 void launch_conv_bwd_data(const float*, const float*, float*, int, int, int,
                           int, int, int, int, int, int, int, int, void*);
 int conv_bwd_weight_splitk(int, int, long);
+int conv_bwd_weight_bf16_splitk(int, int, long);
 void launch_conv_bwd_weight(const float*, const float*, float*, float*, int,
                             int, int, int, int, int, int, int, int, int, int,
                             int, void*);
@@ -777,7 +778,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     }
     int Ncrs = C * R * S;
     long Kdim = (long)Nb * OH * OW;
-    int SK = conv_bwd_weight_splitk(Kout, Ncrs, Kdim);
+    int SK = conv_bwd_weight_bf16_splitk(Kout, Ncrs, Kdim);
     auto dw = torch::empty_like(w);
     auto ws = torch::empty({((long)SK + 1) * Kout * Ncrs},
                            w.options());  // fp32 slabs + rsc temp

@@ -289,20 +289,23 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
 
 // ------------------------------------------------------------ bwd-weight
 
-// 64(ko) x 128(crs) x BK=32(m); A = dy^T, B = x-patch^T — both staged
+// BM=64(ko) x BNW(crs) x BK=32(m); A = dy^T, B = x-patch^T — both staged
 // transposed for contiguous 8-element m fragments.  fp32 slabs/output
 // (weight grads stay fp32); reuses the f32 split-K reduce + dw permute.
-// The 128-wide crs tile doubles MFMA work per barrier (2x4 fragments,
-// 8 MFMA/wave/k-step) and halves dy re-reads vs the first-cut 64x64.
-template <bool P0>
+// BNW=64: 2x2 fragments (4 MFMA/wave/step, 19 KB LDS); BNW=128: 2x4
+// fragments (8 MFMA/wave/step, 31 KB LDS, halves dy re-reads) — variants
+// compared on-box by tests/perf/bwdw_micro.
+template <bool P0, int BNW>
 __global__ __launch_bounds__(256)
 void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
                             const unsigned short* __restrict__ x,
                             float* __restrict__ out, ConvShapeB sh,
                             int Ncrs, long k_per_chunk, int direct_out) {
-  constexpr int BM = 64, BN = 128, MI = 2, NI = 4;
-  __shared__ unsigned short A_lds[2][BM * LDA_B];  // [ko][m]
-  __shared__ unsigned short B_lds[2][BN * LDT_B];  // [crs][m]
+  constexpr int BM = 64, MI = 2, NI = BNW / 32;
+  constexpr int TB = BNW / 8;      // threads per m-row of B staging
+  constexpr int BR = TB / 8;       // B staging rounds (m rows per thread)
+  __shared__ unsigned short A_lds[2][BM * LDA_B];   // [ko][m]
+  __shared__ unsigned short B_lds[2][BNW * LDT_B];  // [crs][m]
   const int t = threadIdx.x;
   const int wave = t >> 6, lane = t & 63;
   const int wr = wave >> 1, wc = wave & 1;
@@ -314,16 +317,15 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     for (int ni = 0; ni < NI; ++ni) acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
 
   const int m_blk = blockIdx.x * BM;   // over ko
-  const int n_blk = blockIdx.y * BN;   // over crs
+  const int n_blk = blockIdx.y * BNW;  // over crs
   const long Kdim = (long)sh.Nb * sh.OH * sh.OW;
   const long k_lo = (long)blockIdx.z * k_per_chunk;
   const long k_hi = min(Kdim, k_lo + k_per_chunk);
 
   // A: thread loads dy[m][ko8..ko8+7]; ko8 = (t&7)*8, m = t>>3 (32 m)
   const int ako = (t & 7) * 8, amr = t >> 3;
-  // B: thread loads x-patch[m][crs8..+7] (8 contiguous c) for TWO m rows
-  // (128 crs x 32 m needs 512 b128 loads from 256 threads)
-  const int bcr = (t & 15) * 8, bmr = t >> 4;
+  // B: thread loads x-patch[m][crs8..+7] (8 contiguous c) for BR m rows
+  const int bcr = (t % TB) * 8, bmr = t / TB;
 
   // (r,s,c0) for this thread's B columns (fixed)
   int br_, bs_, bc0_;
@@ -335,7 +337,22 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     bc0_ = crs - rs * sh.C;
   }
 
-  unsigned short raA[8], rbB[2][8];
+  unsigned short raA[8], rbB[BR][8];
+  // B-side (n,oh,ow) coordinates are advanced INCREMENTALLY by BKB per
+  // stage instead of div/mod per load: the k->(n,oh,ow) divisions cost
+  // ~60 VALU cycles/step vs ~64 MFMA cycles/step — they were the
+  // bottleneck (tests/perf/bwdw_micro: all layers plateaued ~170 us).
+  int c_ow[BR], c_oh[BR];
+  long c_nb[BR], c_k[BR];
+  const int dow = BKB % sh.OW, doh = BKB / sh.OW;
+#pragma unroll
+  for (int j = 0; j < BR; ++j) {
+    long k = k_lo + bmr + j * (256 / TB);
+    c_k[j] = k;
+    c_ow[j] = (int)(k % sh.OW);
+    c_oh[j] = (int)((k / sh.OW) % sh.OH);
+    c_nb[j] = k / ((long)sh.OW * sh.OH);
+  }
   auto stage_load = [&](long k0) {
     {
       long k = k0 + amr;
@@ -353,21 +370,23 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
       *(bf16x8*)raA = q;
     }
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      long k = k0 + bmr + j * 16;
+    for (int j = 0; j < BR; ++j) {
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (k < k_hi) {
-        int ow = k % sh.OW;
-        int oh = (k / sh.OW) % sh.OH;
-        long nb = k / ((long)sh.OW * sh.OH);
-        int ih = oh * sh.stride - sh.pad + br_;
-        int iw = ow * sh.stride - sh.pad + bs_;
+      if (c_k[j] < k_hi) {
+        int ih = c_oh[j] * sh.stride - sh.pad + br_;
+        int iw = c_ow[j] * sh.stride - sh.pad + bs_;
         if (n_blk + bcr < Ncrs &&
             (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
-          q = *(const bf16x8*)(x + (nb * sh.H * sh.W + (long)ih * sh.W +
-                                    iw) * sh.C + bc0_);
+          q = *(const bf16x8*)(x + (c_nb[j] * sh.H * sh.W +
+                                    (long)ih * sh.W + iw) * sh.C + bc0_);
       }
       *(bf16x8*)rbB[j] = q;
+      // advance to this slot's next stage (k += BKB)
+      c_k[j] += BKB;
+      c_ow[j] += dow;
+      if (c_ow[j] >= sh.OW) { c_ow[j] -= sh.OW; ++c_oh[j]; }
+      c_oh[j] += doh;
+      while (c_oh[j] >= sh.OH) { c_oh[j] -= sh.OH; ++c_nb[j]; }
     }
   };
   auto stage_write = [&](int buf) {
@@ -376,10 +395,10 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     for (int e = 0; e < 8; ++e)
       A_lds[buf][(ako + e) * LDA_B + m] = raA[e];
 #pragma unroll
-    for (int j = 0; j < 2; ++j)
+    for (int j = 0; j < BR; ++j)
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        B_lds[buf][(bcr + e) * LDT_B + bmr + j * 16] = rbB[j][e];
+        B_lds[buf][(bcr + e) * LDT_B + bmr + j * (256 / TB)] = rbB[j][e];
   };
 
   stage_load(k_lo);
@@ -401,7 +420,8 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
                                        l4 * 8];
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni)
-      b_frag[ni] = *(const bf16x8*)&Bb[(wc * 32 + ni * 16 + l15) * LDT_B +
+      b_frag[ni] = *(const bf16x8*)&Bb[(wc * (BNW / 2) + ni * 16 + l15) *
+                                           LDT_B +
                                        l4 * 8];
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi)
@@ -417,7 +437,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
   for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
     for (int ni = 0; ni < NI; ++ni) {
-      int crs = n_blk + wc * 32 + ni * 16 + l15;
+      int crs = n_blk + wc * (BNW / 2) + ni * 16 + l15;
       if (crs >= Ncrs) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -479,6 +499,18 @@ extern "C" {
 void launch_splitk_reduce(const float* ws, float* C, const float* bias,
                           int M, int N, int ldc, int SK, int relu, void* s);
 int conv_bwd_weight_splitk(int Kout, int Ncrs, long Kdim);
+
+// bf16 split-K policy: target ~2304 blocks (9/CU queued) — the SK sweep
+// on MI355X showed the f32 path's 1024-block target leaves 10-20% on the
+// table for every ResNet layer shape (tests/perf/bwdw_micro)
+int conv_bwd_weight_bf16_splitk(int Kout, int Ncrs, long Kdim) {
+  long tiles = ((Kout + 63) / 64) * (long)((Ncrs + 63) / 64);
+  if (tiles >= 2304 || Kdim <= 2 * BKB) return 1;
+  long want = (2304 + tiles - 1) / tiles;
+  long max_chunks = (Kdim + BKB - 1) / BKB;
+  long sk = want < max_chunks ? want : max_chunks;
+  return (int)(sk < 1 ? 1 : (sk > 256 ? 256 : sk));
+}
 void launch_dwperm_rsc_crs(const float*, float*, int, int, int, void*);
 void launch_conv_db_stage2(const float*, float*, int, int, void*);
 int conv_db_chunks(long M, int Kout);
@@ -519,30 +551,51 @@ void launch_conv_bwd_data_bf16(const unsigned short* dy,
     conv_bwd_data_bf16_k<0><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
 }
 
-void launch_conv_bwd_weight_bf16(const unsigned short* dy,
-                                 const unsigned short* x, float* dw,
-                                 float* ws, int SK, int Nb, int C, int H,
-                                 int W, int Kout, int R, int S, int OH,
-                                 int OW, int stride, int pad, void* s) {
+// variant 0: 64-wide crs tile; variant 1: 128-wide (bwdw_micro compares)
+void launch_conv_bwd_weight_bf16_ex(const unsigned short* dy,
+                                    const unsigned short* x, float* dw,
+                                    float* ws, int SK, int variant, int Nb,
+                                    int C, int H, int W, int Kout, int R,
+                                    int S, int OH, int OW, int stride,
+                                    int pad, void* s) {
   ConvShapeB sh{Nb, C, H, W, Kout, R, S, OH, OW, stride, pad};
   int Ncrs = C * R * S;
   long Kdim = (long)Nb * OH * OW;
   long k_per_chunk =
       SK == 1 ? Kdim : (((Kdim + SK - 1) / SK + BKB - 1) / BKB) * BKB;
-  dim3 grid((Kout + 63) / 64, (Ncrs + 63) / 64, SK);
+  int bnw = variant == 1 ? 128 : 64;
+  dim3 grid((Kout + 63) / 64, (Ncrs + bnw - 1) / bnw, SK);
   hipStream_t st = (hipStream_t)s;
   float* slabs = ws;
   float* rsc = ws + (long)SK * Kout * Ncrs;
   float* target = SK == 1 ? rsc : slabs;
-  if (pad == 0)
-    conv_bwd_weight_bf16_k<true><<<grid, 256, 0, st>>>(
-        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
-  else
-    conv_bwd_weight_bf16_k<false><<<grid, 256, 0, st>>>(
-        dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  if (variant == 1) {
+    if (pad == 0)
+      conv_bwd_weight_bf16_k<true, 128><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+    else
+      conv_bwd_weight_bf16_k<false, 128><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  } else {
+    if (pad == 0)
+      conv_bwd_weight_bf16_k<true, 64><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+    else
+      conv_bwd_weight_bf16_k<false, 64><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  }
   if (SK > 1)
     launch_splitk_reduce(slabs, rsc, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
   launch_dwperm_rsc_crs(rsc, dw, Kout, C, R * S, s);
+}
+
+void launch_conv_bwd_weight_bf16(const unsigned short* dy,
+                                 const unsigned short* x, float* dw,
+                                 float* ws, int SK, int Nb, int C, int H,
+                                 int W, int Kout, int R, int S, int OH,
+                                 int OW, int stride, int pad, void* s) {
+  launch_conv_bwd_weight_bf16_ex(dy, x, dw, ws, SK, 0, Nb, C, H, W, Kout,
+                                 R, S, OH, OW, stride, pad, s);
 }
 
 void launch_wperm_rsc_ko_bf16(const float* w, unsigned short* out, int Kout,
