@@ -295,7 +295,7 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
 // BNW=64: 2x2 fragments (4 MFMA/wave/step, 19 KB LDS); BNW=128: 2x4
 // fragments (8 MFMA/wave/step, 31 KB LDS, halves dy re-reads) — variants
 // compared on-box by tests/perf/bwdw_micro.
-template <bool P0, int BNW>
+template <bool P0, int BNW, int DEPTH>
 __global__ __launch_bounds__(256)
 void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
                             const unsigned short* __restrict__ x,
@@ -337,7 +337,10 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     bc0_ = crs - rs * sh.C;
   }
 
-  unsigned short raA[8], rbB[BR][8];
+  // DEPTH-1 register sets ring between global load and LDS write: the
+  // global gather for k0 + DEPTH*BK is issued while k0 computes, giving
+  // (DEPTH-1) k-steps of latency budget for the scattered x-patch loads
+  unsigned short raA[DEPTH - 1][8], rbB[DEPTH - 1][BR][8];
   // B-side (n,oh,ow) coordinates are advanced INCREMENTALLY by BKB per
   // stage instead of div/mod per load: the k->(n,oh,ow) divisions cost
   // ~60 VALU cycles/step vs ~64 MFMA cycles/step — they were the
@@ -353,7 +356,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
     c_oh[j] = (int)((k / sh.OW) % sh.OH);
     c_nb[j] = k / ((long)sh.OW * sh.OH);
   }
-  auto stage_load = [&](long k0) {
+  auto stage_load = [&](int set, long k0) {
     {
       long k = k0 + amr;
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -367,7 +370,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
               ((unsigned short*)&q)[e] = dy[k * sh.Kout + m_blk + ako + e];
         }
       }
-      *(bf16x8*)raA = q;
+      *(bf16x8*)raA[set] = q;
     }
 #pragma unroll
     for (int j = 0; j < BR; ++j) {
@@ -380,7 +383,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
           q = *(const bf16x8*)(x + (c_nb[j] * sh.H * sh.W +
                                     (long)ih * sh.W + iw) * sh.C + bc0_);
       }
-      *(bf16x8*)rbB[j] = q;
+      *(bf16x8*)rbB[set][j] = q;
       // advance to this slot's next stage (k += BKB)
       c_k[j] += BKB;
       c_ow[j] += dow;
@@ -389,27 +392,29 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
       while (c_oh[j] >= sh.OH) { c_oh[j] -= sh.OH; ++c_nb[j]; }
     }
   };
-  auto stage_write = [&](int buf) {
+  auto stage_write = [&](int buf, int set) {
     int m = amr;
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      A_lds[buf][(ako + e) * LDA_B + m] = raA[e];
+      A_lds[buf][(ako + e) * LDA_B + m] = raA[set][e];
 #pragma unroll
     for (int j = 0; j < BR; ++j)
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        B_lds[buf][(bcr + e) * LDT_B + bmr + j * (256 / TB)] = rbB[j][e];
+        B_lds[buf][(bcr + e) * LDT_B + bmr + j * (256 / TB)] = rbB[set][j][e];
   };
 
-  stage_load(k_lo);
-  stage_write(0);
-  if (k_lo + BKB < k_hi) stage_load(k_lo + BKB);
+  stage_load(0, k_lo);
+  stage_write(0, 0);
+  if (k_lo + BKB < k_hi) stage_load(0, k_lo + BKB);
+  if (DEPTH > 2 && k_lo + 2 * BKB < k_hi) stage_load(1, k_lo + 2 * BKB);
   __syncthreads();
-  int buf = 0;
+  int buf = 0, set = 0;
   for (long k0 = k_lo; k0 < k_hi; k0 += BKB) {
     if (k0 + BKB < k_hi) {
-      stage_write(buf ^ 1);
-      if (k0 + 2 * BKB < k_hi) stage_load(k0 + 2 * BKB);
+      stage_write(buf ^ 1, set);
+      if (k0 + DEPTH * BKB < k_hi) stage_load(set, k0 + DEPTH * BKB);
+      if (DEPTH > 2) set ^= 1;
     }
     const unsigned short* Ab = A_lds[buf];
     const unsigned short* Bb = B_lds[buf];
@@ -551,7 +556,7 @@ void launch_conv_bwd_data_bf16(const unsigned short* dy,
     conv_bwd_data_bf16_k<0><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
 }
 
-// variant 0: 64-wide crs tile; variant 1: 128-wide (bwdw_micro compares)
+// variant 0: 64-wide tile depth-2; 1: 128-wide; 2: 64-wide depth-3 ring
 void launch_conv_bwd_weight_bf16_ex(const unsigned short* dy,
                                     const unsigned short* x, float* dw,
                                     float* ws, int SK, int variant, int Nb,
@@ -571,17 +576,24 @@ void launch_conv_bwd_weight_bf16_ex(const unsigned short* dy,
   float* target = SK == 1 ? rsc : slabs;
   if (variant == 1) {
     if (pad == 0)
-      conv_bwd_weight_bf16_k<true, 128><<<grid, 256, 0, st>>>(
+      conv_bwd_weight_bf16_k<true, 128, 2><<<grid, 256, 0, st>>>(
           dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
     else
-      conv_bwd_weight_bf16_k<false, 128><<<grid, 256, 0, st>>>(
+      conv_bwd_weight_bf16_k<false, 128, 2><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+  } else if (variant == 2) {
+    if (pad == 0)
+      conv_bwd_weight_bf16_k<true, 64, 3><<<grid, 256, 0, st>>>(
+          dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
+    else
+      conv_bwd_weight_bf16_k<false, 64, 3><<<grid, 256, 0, st>>>(
           dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
   } else {
     if (pad == 0)
-      conv_bwd_weight_bf16_k<true, 64><<<grid, 256, 0, st>>>(
+      conv_bwd_weight_bf16_k<true, 64, 2><<<grid, 256, 0, st>>>(
           dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
     else
-      conv_bwd_weight_bf16_k<false, 64><<<grid, 256, 0, st>>>(
+      conv_bwd_weight_bf16_k<false, 64, 2><<<grid, 256, 0, st>>>(
           dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
   }
   if (SK > 1)

@@ -79,7 +79,7 @@ int main(int argc, char** argv) {
     hipEvent_t e0, e1;
     CHK(hipEventCreate(&e0));
     CHK(hipEventCreate(&e1));
-    for (int variant = 0; variant <= 1; ++variant) {
+    for (int variant = 0; variant <= 2; ++variant) {
       for (int si = 0; si < 10; ++si) {
         int SK = sks[si];
         if (SK > maxc) continue;
@@ -124,6 +124,15 @@ int main(int argc, char** argv) {
     for (size_t i = 0; i < h0.size(); ++i)
       md = fmax(md, fabs((double)h0[i] - h1[i]));
     printf("%-20s v0-vs-v1 max|diff| = %g %s\n", sh.name, md,
+           md == 0 ? "OK" : "MISMATCH");
+    launch_conv_bwd_weight_bf16_ex(ddy, dx, dw1, ws, skp, 2, sh.Nb, sh.C,
+                                   sh.H, sh.W, sh.K, sh.R, sh.S, OH, OW,
+                                   sh.stride, sh.pad, 0);
+    CHK(hipMemcpy(h1.data(), dw1, h1.size() * 4, hipMemcpyDeviceToHost));
+    md = 0;
+    for (size_t i = 0; i < h0.size(); ++i)
+      md = fmax(md, fabs((double)h0[i] - h1[i]));
+    printf("%-20s v0-vs-v2 max|diff| = %g %s\n", sh.name, md,
            md == 0 ? "OK" : "MISMATCH");
     CHK(hipFree(dx)); CHK(hipFree(ddy)); CHK(hipFree(dw0));
     CHK(hipFree(dw1)); CHK(hipFree(ws));
