@@ -1105,12 +1105,14 @@ void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
 
 int conv_bwd_weight_splitk(int Kout, int Ncrs, long Kdim) {
   long tiles = ((Kout + 63) / 64) * (long)((Ncrs + BN - 1) / BN);
-  if (tiles >= 768 || Kdim <= 2 * BK) return 1;
-  // the 64x64 tile runs 4 blocks/CU (35 KB LDS): target ~1024 resident
-  long want = (1024 + tiles - 1) / tiles;
+  if (tiles >= 2304 || Kdim <= 2 * BK) return 1;
+  // target ~4608 blocks, cap SK at 192: the MI355X SK sweep
+  // (profiles/r01_bwdw_micro.md f32 section) put the optimum at
+  // 4.5k blocks for the CIFAR shapes and SK~192 for the FMNIST conv2
+  long want = (4608 + tiles - 1) / tiles;
   long max_chunks = (Kdim + BK - 1) / BK;
   long sk = want < max_chunks ? want : max_chunks;
-  return (int)(sk < 1 ? 1 : (sk > 256 ? 256 : sk));
+  return (int)(sk < 1 ? 1 : (sk > 192 ? 192 : sk));
 }
 
 // ws: SK*Kout*Ncrs + Kout*Ncrs floats (split-K slabs + rsc-ordered temp)

@@ -78,6 +78,20 @@ if [ "$WHAT" = "profr" ]; then
   grep -o '"ms_per_step": [0-9.]*' $GRAFT_REPO_ROOT/gpurun_out/profr.log
 fi
 
+if [ "$WHAT" = "dbacurves" ]; then
+  # BASELINE config 4 semantics at full reference scale (runner.sh:23-28):
+  # CIFAR10, 40 agents, 4 corrupt, DBA plus-pattern, 200 rounds
+  for CFG in "atk:--num_corrupt 4 --poison_frac 0.5" \
+             "rlr:--num_corrupt 4 --poison_frac 0.5 --robustLR_threshold 8"; do
+    TAG="${CFG%%:*}"; FLAGS="${CFG#*:}"
+    timeout 700 python -m rlr_amd.federated --data cifar10 --num_agents 40 \
+      --rounds 200 --snap 50 --pattern_type plus --synthetic --no_tb $FLAGS \
+      > gpurun_out/dba_$TAG.log 2>&1
+    echo "== $TAG rc=$?"
+    grep -E "Val_Loss|Poison Loss" gpurun_out/dba_$TAG.log | tail -2
+  done
+fi
+
 if [ "$WHAT" = "pmc" ]; then
   cd /tmp && export TMPDIR=/tmp
   timeout 300 rocprofv3 --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY SQ_ACTIVE_INST_ANY SQ_LDS_BANK_CONFLICT SQ_INSTS_VALU SQ_INSTS_MFMA SQ_INSTS_LDS \

@@ -21,6 +21,10 @@ void launch_conv_bwd_weight_bf16_ex(const unsigned short*,
                                     int, int, int, int, int, int, int, int,
                                     int, int, int, int, int, void*);
 int conv_bwd_weight_bf16_splitk(int, int, long);
+void launch_conv_bwd_weight(const float*, const float*, float*, float*,
+                            int, int, int, int, int, int, int, int, int,
+                            int, int, int, void*);
+int conv_bwd_weight_splitk(int, int, long);
 }
 
 #define CHK(x)                                                      \
@@ -41,8 +45,64 @@ static unsigned short f2bf(float f) {
 
 struct Shape { int Nb, C, H, W, K, R, S, stride, pad; const char* name; };
 
+// f32 bwd-weight SK sweep over the headline CNN shapes (fp32 path feeds
+// the FMNIST/CIFAR CNN bench) — same one-box method as the bf16 sweep.
+static void f32_sweep(int iters) {
+  Shape shapes[] = {
+      {256, 32, 26, 26, 64, 3, 3, 1, 0, "F2 fmnist 32->64"},
+      {256, 64, 30, 30, 128, 3, 3, 1, 0, "C2 cifar 64->128"},
+      {256, 128, 13, 13, 256, 3, 3, 1, 0, "C3 cifar 128->256"},
+  };
+  srand(11);
+  for (const Shape& sh : shapes) {
+    int OH = (sh.H - sh.R) / sh.stride + 1, OW = (sh.W - sh.S) / sh.stride + 1;
+    long M = (long)sh.Nb * OH * OW;
+    long nx = (long)sh.Nb * sh.H * sh.W * sh.C, ndy = M * sh.K;
+    int Ncrs = sh.C * sh.R * sh.S;
+    std::vector<float> hx(nx), hdy(ndy);
+    for (long i = 0; i < nx; ++i) hx[i] = (rand() % 1000 - 500) / 500.f;
+    for (long i = 0; i < ndy; ++i) hdy[i] = (rand() % 1000 - 500) / 500.f;
+    float *dx, *ddy, *dw, *ws;
+    CHK(hipMalloc(&dx, nx * 4));
+    CHK(hipMalloc(&ddy, ndy * 4));
+    CHK(hipMalloc(&dw, (long)sh.K * Ncrs * 4));
+    CHK(hipMalloc(&ws, 800L * 1024 * 1024));
+    CHK(hipMemcpy(dx, hx.data(), nx * 4, hipMemcpyHostToDevice));
+    CHK(hipMemcpy(ddy, hdy.data(), ndy * 4, hipMemcpyHostToDevice));
+    int skp = conv_bwd_weight_splitk(sh.K, Ncrs, M);
+    hipEvent_t e0, e1;
+    CHK(hipEventCreate(&e0));
+    CHK(hipEventCreate(&e1));
+    int sks[] = {skp, 64, 128, 192, 256};
+    for (int si = 0; si < 5; ++si) {
+      int SK = sks[si];
+      if (si > 0 && SK == skp) continue;
+      if ((long)(SK + 1) * sh.K * Ncrs * 4 > 800L * 1024 * 1024) continue;
+      for (int w = 0; w < 3; ++w)
+        launch_conv_bwd_weight(ddy, dx, dw, ws, SK, sh.Nb, sh.C, sh.H,
+                               sh.W, sh.K, sh.R, sh.S, OH, OW, sh.stride,
+                               sh.pad, 0);
+      CHK(hipDeviceSynchronize());
+      CHK(hipEventRecord(e0));
+      for (int it = 0; it < iters; ++it)
+        launch_conv_bwd_weight(ddy, dx, dw, ws, SK, sh.Nb, sh.C, sh.H,
+                               sh.W, sh.K, sh.R, sh.S, OH, OW, sh.stride,
+                               sh.pad, 0);
+      CHK(hipEventRecord(e1));
+      CHK(hipEventSynchronize(e1));
+      float ms;
+      CHK(hipEventElapsedTime(&ms, e0, e1));
+      printf("%-20s f32 SK=%-3d %8.1f us%s\n", sh.name, SK,
+             ms * 1000.f / iters, SK == skp ? "  (policy)" : "");
+    }
+    CHK(hipFree(dx)); CHK(hipFree(ddy)); CHK(hipFree(dw)); CHK(hipFree(ws));
+    CHK(hipEventDestroy(e0)); CHK(hipEventDestroy(e1));
+  }
+}
+
 int main(int argc, char** argv) {
   int iters = argc > 1 ? atoi(argv[1]) : 10;
+  if (argc > 2 && argv[2][0] == 'f') { f32_sweep(iters); return 0; }
   Shape shapes[] = {
       {256, 64, 32, 32, 64, 3, 3, 1, 1, "L1 64->64 32x32"},
       {256, 128, 16, 16, 128, 3, 3, 1, 1, "L2 128->128 16x16"},
