@@ -13,7 +13,16 @@
 // pass that bn_norm_k covers in 8 us.
 #include "common.h"
 
-constexpr int kBnChunks = 512;  // total row chunks (stage-1 partials)
+constexpr int kBnChunksMax = 512;  // partials allocation bound
+// row chunks scale with M: big activations want ~512-way stage-1
+// parallelism; small deep-layer tensors (M=4k) waste stage-2 iterations
+// on empty chunks.  Multiple of 16 so stage-1 slice blocks tile exactly.
+static __host__ __device__ inline int bn_chunks(long M) {
+  long c = M / 32;
+  if (c > kBnChunksMax) c = kBnChunksMax;
+  if (c < 64) c = 64;
+  return (int)(c & ~15L);
+}
 
 // ---- stage 1: partials[chunk][2C] = (sum x, sum x^2) over a row chunk ----
 // grid: x = kBnChunks / slices_per_block, y = ceil(C / C_blk);
@@ -26,8 +35,9 @@ __global__ void bn_stats1_k(const T* __restrict__ x,
   int c = blockIdx.y * C_blk + threadIdx.x % C_blk;
   int sub = threadIdx.x / C_blk;
   int chunk = blockIdx.x * sub_per + sub;
-  if (c >= C || chunk >= kBnChunks) return;
-  long per = (M + kBnChunks - 1) / kBnChunks;
+  const int chunks = bn_chunks(M);
+  if (c >= C || chunk >= chunks) return;
+  long per = (M + chunks - 1) / chunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float s = 0.f, ss = 0.f;
 #pragma unroll 4
@@ -45,7 +55,7 @@ __global__ void bn_stats1_k(const T* __restrict__ x,
 // strided loops), slices combined in LDS in fixed order (deterministic).
 // A single-block variant at 512 chunks measured 87 us — this is ~3 us.
 __global__ void bn_stats2_k(const float* __restrict__ partials,
-                            float* __restrict__ sums, int C,
+                            float* __restrict__ sums, int C, int chunks,
                             float* __restrict__ copy0,
                             float* __restrict__ copy1) {
   constexpr int CB = 16, SL = 16;  // CB*SL == blockDim.x == 256
@@ -55,7 +65,7 @@ __global__ void bn_stats2_k(const float* __restrict__ partials,
   int sub = threadIdx.x / CB;
   float s = 0.f, ss = 0.f;
   if (c < C) {
-    for (int ch = sub; ch < kBnChunks; ch += SL) {
+    for (int ch = sub; ch < chunks; ch += SL) {
       s += partials[((long)ch * 2) * C + c];
       ss += partials[((long)ch * 2 + 1) * C + c];
     }
@@ -145,8 +155,9 @@ __global__ void bn_bwd_stats1_k(const T* __restrict__ x,
   int c = blockIdx.y * C_blk + threadIdx.x % C_blk;
   int sub = threadIdx.x / C_blk;
   int chunk = blockIdx.x * sub_per + sub;
-  if (c >= C || chunk >= kBnChunks) return;
-  long per = (M + kBnChunks - 1) / kBnChunks;
+  const int chunks = bn_chunks(M);
+  if (c >= C || chunk >= chunks) return;
+  long per = (M + chunks - 1) / chunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float m_ = mean[c], rs = rstd[c];
   float sdy = 0.f, sdyx = 0.f;
@@ -224,14 +235,15 @@ static void bn_fwd_impl(const T* x, const float* w, const float* b,
                         hipStream_t st) {
   long M = (long)Nb * HW;
   float* partials = scratch;
-  float* sums = scratch + (long)kBnChunks * 2 * C;
+  float* sums = scratch + (long)kBnChunksMax * 2 * C;
   if (training) {
     int C_blk = bn_cblk(C);
     int sub_per = 256 / C_blk;
-    dim3 g1(kBnChunks / sub_per, (C + C_blk - 1) / C_blk);
+    int chunks = bn_chunks(M);
+    dim3 g1((chunks + sub_per - 1) / sub_per, (C + C_blk - 1) / C_blk);
     bn_stats1_k<T><<<g1, 256, 0, st>>>(x, partials, M, C, C_blk);
-    bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, sums, C, nullptr,
-                                               nullptr);
+    bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, sums, C, chunks,
+                                               nullptr, nullptr);
     bn_finalize_k<<<(C + 255) / 256, 256, 0, st>>>(
         sums, save_mean, save_rstd, running_mean, running_var, C, M,
         momentum, eps);
@@ -251,13 +263,15 @@ static void bn_bwd_impl(const T* x, const T* dy, const float* w,
                         int C, int HW, int training, hipStream_t st) {
   long M = (long)Nb * HW;
   float* partials = scratch;
-  float* stats = scratch + (long)kBnChunks * 2 * C;
+  float* stats = scratch + (long)kBnChunksMax * 2 * C;
   int C_blk = bn_cblk(C);
   int sub_per = 256 / C_blk;
-  dim3 g1(kBnChunks / sub_per, (C + C_blk - 1) / C_blk);
+  int chunks = bn_chunks(M);
+  dim3 g1((chunks + sub_per - 1) / sub_per, (C + C_blk - 1) / C_blk);
   bn_bwd_stats1_k<T><<<g1, 256, 0, st>>>(x, dy, save_mean, save_rstd,
                                          partials, M, C, C_blk);
-  bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, stats, C, db, dw);
+  bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, stats, C, chunks,
+                                             db, dw);
   long n = M * C;
   bn_bwd_dx_k<T><<<bn_grid(n, C), kBlock, 0, st>>>(
       x, dy, w, save_mean, save_rstd, stats, dx, n, C,
@@ -265,7 +279,7 @@ static void bn_bwd_impl(const T* x, const T* dy, const float* w,
 }
 
 extern "C" {
-int bn_scratch_floats(int C) { return (kBnChunks * 2 + 2) * C; }
+int bn_scratch_floats(int C) { return (kBnChunksMax * 2 + 2) * C; }
 
 void launch_bn_fwd(const float* x, const float* w, const float* b,
                    float* running_mean, float* running_var, float* save_mean,
