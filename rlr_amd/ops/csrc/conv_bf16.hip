@@ -487,15 +487,18 @@ __global__ void wperm_rsko_c_bf16_k(const float* __restrict__ w,
 }
 
 // bf16 NHWC conv-bias gradient: db[ko] = column sums of dy [M][KO]
+// (full-block slice mapping as in the f32 conv_db_stage1_k)
 __global__ void conv_db_bf16_stage1_k(const unsigned short* __restrict__ dy,
                                       float* __restrict__ partials, long M,
-                                      int Kout, int chunks) {
-  int chunk = blockIdx.x;
-  int ko = blockIdx.y * blockDim.x + threadIdx.x;
-  if (ko >= Kout) return;
+                                      int Kout, int chunks, int K_blk) {
+  int sub_per = blockDim.x / K_blk;
+  int ko = blockIdx.y * K_blk + threadIdx.x % K_blk;
+  int chunk = blockIdx.x * sub_per + threadIdx.x / K_blk;
+  if (ko >= Kout || chunk >= chunks) return;
   long per = (M + chunks - 1) / chunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float acc = 0.f;
+#pragma unroll 4
   for (long m = lo; m < hi; ++m) acc += bf2f_(dy[m * Kout + ko]);
   partials[(long)chunk * Kout + ko] = acc;
 }
@@ -627,9 +630,12 @@ void launch_conv_db_bf16(const unsigned short* dy, float* db,
   hipStream_t st = (hipStream_t)s;
   long M = (long)Nb * OHW;
   int chunks = conv_db_chunks(M, Kout);
-  dim3 g1(chunks, (Kout + kBlock - 1) / kBlock);
+  int K_blk = Kout >= kBlock ? kBlock
+                             : ((kBlock % Kout == 0) ? Kout : kBlock);
+  int sub_per = kBlock / K_blk;
+  dim3 g1((chunks + sub_per - 1) / sub_per, (Kout + K_blk - 1) / K_blk);
   conv_db_bf16_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout,
-                                               chunks);
+                                               chunks, K_blk);
   launch_conv_db_stage2(partials, db, Kout, chunks, s);
 }
 }

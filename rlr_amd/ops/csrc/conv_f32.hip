@@ -953,15 +953,20 @@ __global__ void dwperm_rsc_crs_k(const float* __restrict__ in,
 // ---- conv bias gradient (NHWC): db[ko] = column sum of dy [M][KO] ----
 // Dynamic chunk count keeps >=64k threads busy regardless of Kout; stage 2
 // is one wave per ko over the chunk partials (shuffle tree, deterministic).
+// thread -> (slice = tid / K_blk, ko = tid % K_blk): all four waves of a
+// block work even when Kout < 256 (the one-slice form left 7/8 of each
+// block idle at Kout=32 — 21.5 us for a 9.4 MB column sum)
 __global__ void conv_db_stage1_k(const float* __restrict__ dy,
                                  float* __restrict__ partials, long M,
-                                 int Kout, int chunks) {
-  int chunk = blockIdx.x;
-  int ko = blockIdx.y * blockDim.x + threadIdx.x;
-  if (ko >= Kout) return;
+                                 int Kout, int chunks, int K_blk) {
+  int sub_per = blockDim.x / K_blk;
+  int ko = blockIdx.y * K_blk + threadIdx.x % K_blk;
+  int chunk = blockIdx.x * sub_per + threadIdx.x / K_blk;
+  if (ko >= Kout || chunk >= chunks) return;
   long per = (M + chunks - 1) / chunks;
   long lo = (long)chunk * per, hi = min(M, lo + per);
   float acc = 0.f;
+#pragma unroll 4
   for (long m = lo; m < hi; ++m) acc += dy[m * Kout + ko];
   partials[(long)chunk * Kout + ko] = acc;
 }
@@ -1169,13 +1174,21 @@ int conv_db_chunks(long M, int Kout) {
   return (int)c;
 }
 
+static int conv_db_kblk(int Kout) {
+  if (Kout >= kBlock) return kBlock;
+  return (kBlock % Kout == 0) ? Kout : kBlock;
+}
+
 void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
                     int Kout, int OHW, void* s) {
   hipStream_t st = (hipStream_t)s;
   long M = (long)Nb * OHW;
   int chunks = conv_db_chunks(M, Kout);
-  dim3 g1(chunks, (Kout + kBlock - 1) / kBlock);
-  conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout, chunks);
+  int K_blk = conv_db_kblk(Kout);
+  int sub_per = kBlock / K_blk;
+  dim3 g1((chunks + sub_per - 1) / sub_per, (Kout + K_blk - 1) / K_blk);
+  conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout, chunks,
+                                          K_blk);
   conv_db_stage2_k<<<Kout, kBlock, 0, st>>>(partials, db, Kout, chunks);
 }
 
