@@ -817,7 +817,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   auto dw = torch::empty_like(w);
   // ws: SK slabs + one rsc-ordered temp; the small-Ncrs direct path uses
   // 512 chunk partials instead
-  long ws_mult = (Ncrs <= 32 && Kout <= 64) ? 513 : (long)SK + 1;
+  long ws_mult = (Ncrs <= 32 && Kout <= 64) ? 2049 : (long)SK + 1;
   auto ws = torch::empty({ws_mult * Kout * Ncrs}, w.options());
   launch_conv_bwd_weight(dy.data_ptr<float>(), x.data_ptr<float>(),
                          dw.data_ptr<float>(), ws.data_ptr<float>(), SK, Nb,

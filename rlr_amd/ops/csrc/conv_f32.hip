@@ -1133,7 +1133,11 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
   dim3 grid((Kout + 63) / 64, (Ncrs + BN - 1) / BN, SK);
   hipStream_t st = (hipStream_t)s;
   if (Ncrs <= 32 && Kout <= 64) {
-    int chunks = 512;
+    // one 128-row tile per block where possible: 512 fixed chunks left the
+    // chip at 2 blocks/CU (48 KB LDS, 4 waves) — finer chunking trades a
+    // longer fixed-order reduce for 3x the resident parallelism
+    int chunks = (int)min((Kdim + 127) / 128, (long)2048);
+    if (chunks < 64) chunks = 64;
     long kpc = ((Kdim + chunks - 1) / chunks + 127) / 128 * 128;
     conv_small_bwdw_k<128><<<chunks, 256, 0, st>>>(dy, x, ws, sh, Ncrs,
                                                    kpc);
