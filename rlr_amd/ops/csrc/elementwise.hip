@@ -107,13 +107,14 @@ __global__ void maxpool2x2_fwd_k(const T* __restrict__ x,
   }
 }
 
-// Gather form: one thread per INPUT element (write-once, no zero-init).
+// Gather form: one thread per INPUT element (write-once, no zero-init);
+// the default (RLR_MP_STRIP=1 selects the strip form below).
 template <typename T>
-__global__ void maxpool2x2_bwd_k(const T* __restrict__ dy,
-                                 const uint8_t* __restrict__ idx,
-                                 T* __restrict__ dx,
-                                 long B, int H, int W, int OH, int OW,
-                                 int C) {
+__global__ void maxpool2x2_bwd_gather_k(const T* __restrict__ dy,
+                                        const uint8_t* __restrict__ idx,
+                                        T* __restrict__ dx,
+                                        long B, int H, int W, int OH,
+                                        int OW, int C) {
   long n_in = B * H * (long)W * C;
   long stride = (long)gridDim.x * blockDim.x;
   int csh = 31 - __clz(C);
@@ -136,6 +137,65 @@ __global__ void maxpool2x2_bwd_k(const T* __restrict__ dy,
   }
 }
 
+// Strip form: one thread per (b, oh, c) output ROW STRIP.  Walking ow in
+// a loop removes the per-element div/mod of the earlier gather form (the
+// only VALU divisions on this path; they held the kernel to ~1.5 TB/s
+// while relu_bwd streams at ~2.8).  Every input cell belongs to exactly
+// one 2x2 window (stride 2), so each is written exactly once; odd tail
+// rows/cols (H or W > 2*OH/2*OW) receive zero gradient, matching the
+// reference pool's floor semantics.
+template <typename T>
+__global__ void maxpool2x2_bwd_k(const T* __restrict__ dy,
+                                 const uint8_t* __restrict__ idx,
+                                 T* __restrict__ dx,
+                                 long B, int H, int W, int OH, int OW,
+                                 int C) {
+  long n_strip = B * OH * (long)C;
+  long stride = (long)gridDim.x * blockDim.x;
+  int WC = W * C;
+  int csh = 31 - __clz(C);
+  bool cp2 = (C & (C - 1)) == 0;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_strip;
+       i += stride) {
+    int c = cp2 ? (int)(i & (C - 1)) : (int)(i % C);
+    long rest = cp2 ? (i >> csh) : (i / C);
+    int oh = rest % OH;
+    long b = rest / OH;
+    const T* dyp = dy + ((b * OH + oh) * (long)OW) * C + c;
+    const uint8_t* ip = idx + ((b * OH + oh) * (long)OW) * C + c;
+    T* dxp = dx + ((b * H + 2 * oh) * (long)W) * C + c;
+    for (int ow = 0; ow < OW; ++ow) {
+      float g = ldv(dyp + (long)ow * C);
+      uint8_t a = ip[(long)ow * C];
+      long p = 2L * ow * C;
+      stv(dxp + p, a == 0 ? g : 0.f);
+      stv(dxp + p + C, a == 1 ? g : 0.f);
+      stv(dxp + p + WC, a == 2 ? g : 0.f);
+      stv(dxp + p + WC + C, a == 3 ? g : 0.f);
+    }
+    // odd input tail columns/rows get zero gradient
+    for (int iw = 2 * OW; iw < W; ++iw) {
+      stv(dxp + (long)iw * C, 0.f);
+      stv(dxp + (long)iw * C + WC, 0.f);
+    }
+    if (oh == OH - 1)
+      for (int ih = 2 * OH; ih < H; ++ih)
+        for (int iw = 0; iw < W; ++iw)
+          stv(dx + ((b * H + ih) * (long)W + iw) * C + c, 0.f);
+  }
+}
+
+static bool mp_strip() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("RLR_MP_STRIP");
+    // A/B on MI355X was a tie at bench scale (4.21/4.21 gather vs
+    // 4.13/4.34 strip) — keep the simpler gather form unless asked
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 extern "C" {
 void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, long B,
                            int H, int W, int OH, int OW, int C, void* s) {
@@ -146,9 +206,16 @@ void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, long B,
 void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
                            long B, int H, int W, int OH, int OW, int C,
                            void* s) {
-  long n = B * H * (long)W * C;
-  maxpool2x2_bwd_k<float><<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
-      dy, idx, dx, B, H, W, OH, OW, C);
+  if (mp_strip()) {
+    long n = B * OH * (long)C;  // one thread per (b, oh, c) strip
+    maxpool2x2_bwd_k<float><<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
+        dy, idx, dx, B, H, W, OH, OW, C);
+  } else {
+    long n = B * H * (long)W * C;
+    maxpool2x2_bwd_gather_k<float>
+        <<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx, B, H, W,
+                                                     OH, OW, C);
+  }
 }
 void launch_maxpool2x2_fwd_bf16(const unsigned short* x, unsigned short* y,
                                 uint8_t* idx, long B, int H, int W, int OH,
@@ -162,7 +229,14 @@ void launch_maxpool2x2_bwd_bf16(const unsigned short* dy,
                                 const uint8_t* idx, unsigned short* dx,
                                 long B, int H, int W, int OH, int OW, int C,
                                 void* s) {
-  long n = B * H * (long)W * C;
+  if (!mp_strip()) {
+    long n2 = B * H * (long)W * C;
+    maxpool2x2_bwd_gather_k<unsigned short>
+        <<<grid_for(n2), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx, B, H, W,
+                                                      OH, OW, C);
+    return;
+  }
+  long n = B * OH * (long)C;  // one thread per (b, oh, c) strip
   maxpool2x2_bwd_k<unsigned short>
       <<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx, B, H, W, OH,
                                                    OW, C);
