@@ -125,3 +125,16 @@ def test_poison_deterministic_across_calls(tiny_sizes):
     p1 = poison_dataset(t1, args, agent_idx=2)
     p2 = poison_dataset(t2, args, agent_idx=2)
     assert p1 == p2
+
+
+def test_unsupported_pattern_raises():
+    """Explicit validation improvement over the reference (PARITY.md):
+    utils.py:187 silently relabels without writing a pattern for
+    unsupported (dataset, pattern_type) combos; this build raises."""
+    import pytest
+    with pytest.raises(ValueError):
+        pattern_spec('cifar10', 'square', agent_idx=-1)
+    with pytest.raises(ValueError):
+        pattern_spec('fmnist', 'nosuch')
+    with pytest.raises(ValueError):
+        pattern_spec('nosuchdata', 'plus')
