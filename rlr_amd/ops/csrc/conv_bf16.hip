@@ -404,18 +404,7 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
         B_lds[buf][(bcr + e) * LDT_B + bmr + j * (256 / TB)] = rbB[set][j][e];
   };
 
-  stage_load(0, k_lo);
-  stage_write(0, 0);
-  if (k_lo + BKB < k_hi) stage_load(0, k_lo + BKB);
-  if (DEPTH > 2 && k_lo + 2 * BKB < k_hi) stage_load(1, k_lo + 2 * BKB);
-  __syncthreads();
-  int buf = 0, set = 0;
-  for (long k0 = k_lo; k0 < k_hi; k0 += BKB) {
-    if (k0 + BKB < k_hi) {
-      stage_write(buf ^ 1, set);
-      if (k0 + DEPTH * BKB < k_hi) stage_load(set, k0 + DEPTH * BKB);
-      if (DEPTH > 2) set ^= 1;
-    }
+  auto mfma_step = [&](int buf) {
     const unsigned short* Ab = A_lds[buf];
     const unsigned short* Bb = B_lds[buf];
     bf16x8 a_frag[MI], b_frag[NI];
@@ -434,8 +423,48 @@ void conv_bwd_weight_bf16_k(const unsigned short* __restrict__ dy,
       for (int ni = 0; ni < NI; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
-    __syncthreads();
-    buf ^= 1;
+  };
+
+  stage_load(0, k_lo);
+  stage_write(0, 0);
+  if (k_lo + BKB < k_hi) stage_load(0, k_lo + BKB);
+  if (DEPTH > 2 && k_lo + 2 * BKB < k_hi) stage_load(1, k_lo + 2 * BKB);
+  __syncthreads();
+  int buf = 0;
+  if (DEPTH > 2) {
+    // 2x-unrolled so the register-ring set index is a LITERAL at every
+    // call site — a runtime index forces the ring into scratch (measured
+    // 2.5x regression, profiles/r01_bwdw_micro.md)
+    long k0 = k_lo;
+    while (k0 < k_hi) {
+      if (k0 + BKB < k_hi) {
+        stage_write(buf ^ 1, 0);
+        if (k0 + 3 * BKB < k_hi) stage_load(0, k0 + 3 * BKB);
+      }
+      mfma_step(buf);
+      __syncthreads();
+      buf ^= 1;
+      k0 += BKB;
+      if (k0 >= k_hi) break;
+      if (k0 + BKB < k_hi) {
+        stage_write(buf ^ 1, 1);
+        if (k0 + 3 * BKB < k_hi) stage_load(1, k0 + 3 * BKB);
+      }
+      mfma_step(buf);
+      __syncthreads();
+      buf ^= 1;
+      k0 += BKB;
+    }
+  } else {
+    for (long k0 = k_lo; k0 < k_hi; k0 += BKB) {
+      if (k0 + BKB < k_hi) {
+        stage_write(buf ^ 1, 0);
+        if (k0 + 2 * BKB < k_hi) stage_load(0, k0 + 2 * BKB);
+      }
+      mfma_step(buf);
+      __syncthreads();
+      buf ^= 1;
+    }
   }
 
 #pragma unroll
@@ -609,7 +638,9 @@ void launch_conv_bwd_weight_bf16(const unsigned short* dy,
                                  float* ws, int SK, int Nb, int C, int H,
                                  int W, int Kout, int R, int S, int OH,
                                  int OW, int stride, int pad, void* s) {
-  launch_conv_bwd_weight_bf16_ex(dy, x, dw, ws, SK, 0, Nb, C, H, W, Kout,
+  // depth-3 ring (variant 2) wins 3.5-4% on every 3x3 shape at the policy
+  // SK and ties on 1x1 (bwdw_micro run 4) — production default
+  launch_conv_bwd_weight_bf16_ex(dy, x, dw, ws, SK, 2, Nb, C, H, W, Kout,
                                  R, S, OH, OW, stride, pad, s);
 }
 
