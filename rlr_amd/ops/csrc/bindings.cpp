@@ -99,7 +99,6 @@ int gemm_f32_splitk(int, int, int);
 void launch_gemm_f32(const float*, const float*, float*, const float*,
                      float*, int, int, int, int, int, int, int, int, int,
                      void*);
-void launch_transpose_f32(const float*, float*, int, int, void*);
 void launch_colsum(const float*, float*, int, int, void*);
 // conv_f32.hip
 void launch_conv_fwd(const float*, const float*, const float*, float*, int,
@@ -645,13 +644,6 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
-torch::Tensor transpose2d(torch::Tensor A) {
-  CHK_CUDA(A);
-  auto out = torch::empty({A.size(1), A.size(0)}, A.options());
-  launch_transpose_f32(A.data_ptr<float>(), out.data_ptr<float>(),
-                       A.size(0), A.size(1), stream_of(A));
-  return out;
-}
 
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> b, bool relu) {
@@ -987,7 +979,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_noise", &add_noise);
   m.def("gemm", &gemm);
   m.def("gemm_bf16", &gemm_bf16);
-  m.def("transpose2d", &transpose2d);
   m.def("nhwc_flatten", &nhwc_flatten);
   m.def("nhwc_unflatten", &nhwc_unflatten);
   m.def("linear_fwd", &linear_fwd);

@@ -292,29 +292,6 @@ __global__ void splitk_reduce_k(const float* __restrict__ ws,
   }
 }
 
-// 32x32 LDS-tiled transpose: out[N,M] = in[M,N]^T
-__global__ void transpose_f32_k(const float* __restrict__ in,
-                                float* __restrict__ out, int M, int N) {
-  __shared__ float tile[32][33];
-  int bx = blockIdx.x * 32, by = blockIdx.y * 32;
-  int x = bx + (threadIdx.x & 31);
-  int y0 = by + (threadIdx.x >> 5) * 4;
-  for (int j = 0; j < 4; ++j) {
-    int y = y0 + j;
-    if (y < M && x < N)
-      tile[threadIdx.x & 31][(threadIdx.x >> 5) * 4 + j] =
-          in[(long)y * N + x];
-  }
-  __syncthreads();
-  int ox = by + (threadIdx.x & 31);
-  int oy0 = bx + (threadIdx.x >> 5) * 4;
-  for (int j = 0; j < 4; ++j) {
-    int oy = oy0 + j;
-    if (oy < N && ox < M)
-      out[(long)oy * M + ox] = tile[(threadIdx.x >> 5) * 4 + j]
-                                   [threadIdx.x & 31];
-  }
-}
 
 // column sum: db[n] = sum_m dY[m][n] (bias gradient).
 // One wave per column; lane l accumulates rows l, l+64, ... then a shuffle
@@ -405,11 +382,6 @@ void launch_splitk_reduce(const float* ws, float* C, const float* bias,
   }
 }
 
-void launch_transpose_f32(const float* in, float* out, int M, int N,
-                          void* s) {
-  dim3 grid((N + 31) / 32, (M + 31) / 32);
-  transpose_f32_k<<<grid, 256, 0, (hipStream_t)s>>>(in, out, M, N);
-}
 
 void launch_colsum(const float* dY, float* db, int M, int N, void* s) {
   int wpb = kBlock / kWave;
