@@ -18,7 +18,12 @@ Rules (each a HIP kernel on the GPU path, ops/csrc/aggregation.hip):
     (lower of the two middle values for even K).
   * sign (aggregation.py:71-75): sign(sum_k sign(U_k)).
   * optional N(0, noise*clip) gaussian noise (aggregation.py:34-35) from a
-    per-round derived stream.
+    per-round derived stream.  NOTE: the fused GPU avg path draws noise
+    from the on-device philox Box-Muller stream, while the CPU and
+    comed/sign paths draw from a torch CPU generator — the same
+    seed+config yields a DIFFERENT (but equally deterministic) noise
+    realization across the two backends.  Each backend is reproducible
+    and world-size-invariant on its own; see PARITY.md.
   * fused apply (aggregation.py:38-40): theta <- float32(theta + lr * agg).
 """
 
@@ -119,7 +124,9 @@ class Aggregation:
         return out / w.sum()
 
     def agg_comed(self, stacked):
-        if _gpu(stacked):
+        # the register-select kernel holds one column of K values per lane;
+        # beyond K=64 fall back to torch.median on device (same semantics)
+        if _gpu(stacked) and stacked.shape[0] <= 64:
             return ext().agg_comed(stacked)
         return torch.median(stacked, dim=0).values
 

@@ -151,16 +151,23 @@ def _synthetic_pair(kind, seed, train_n, val_n, n_classes=10):
 def _synthetic_fedemnist(seed, num_users, samples_per_user, val_n, n_classes=10):
     """Non-IID writers: each user draws from 2-4 classes with its own style
     offset; inputs are already-normalized floats like the reference's
-    pre-built .pt shards (utils.py:105-109)."""
+    pre-built .pt shards (utils.py:105-109).  Per-user sample counts are
+    lognormal-skewed around `samples_per_user` (the real Fed-EMNIST has
+    heavy per-writer size skew, mean ~341), floor 16 — so data-size FedAvg
+    weights are genuinely heterogeneous."""
     mean, std = NORM_STATS['fmnist']
     rng = np_rng(seed, 'data', 2)
     protos = rng.normal(0.0, 1.0, size=(n_classes, 28, 28))
     users = []
     for u in range(num_users):
         urng = np_rng(seed, 'data', 3, u)
+        # lognormal with sigma=0.5 has mean exp(mu + sigma^2/2); scale so
+        # the expected count equals samples_per_user
+        n_u = max(16, int(samples_per_user *
+                          urng.lognormal(-0.125, 0.5)))
         k = int(urng.integers(2, 5))
         classes = urng.choice(n_classes, k, replace=False)
-        t = urng.choice(classes, samples_per_user)
+        t = urng.choice(classes, n_u)
         style = urng.normal(0.0, 0.3, size=(28, 28))
         x = protos[t] + style[None] + urng.normal(0.0, 0.5, size=(len(t), 28, 28))
         users.append(TensorDataset(
@@ -179,7 +186,7 @@ def _synthetic_fedemnist(seed, num_users, samples_per_user, val_n, n_classes=10)
 DEFAULT_SIZES = {
     'fmnist': (60000, 10000),
     'cifar10': (50000, 10000),
-    'fedemnist': (3383, 161),  # users, samples/user (reference scale ~341 avg)
+    'fedemnist': (3383, 341),  # users, mean samples/user (reference scale)
 }
 
 
@@ -193,8 +200,18 @@ def get_datasets(data, args=None, train_n=None, val_n=None, data_dir='../data'):
     synthetic = True if args is None else bool(getattr(args, 'synthetic', True))
 
     if data in ('fmnist', 'cifar10'):
-        if not synthetic and _try_real_available(data, data_dir):
-            return _load_real(data, data_dir)
+        if not synthetic:
+            if _try_real_available(data, data_dir):
+                return _load_real(data, data_dir)
+            # never silently substitute synthetic for requested real data:
+            # the metrics would be misread as real-dataset results
+            import warnings
+            warnings.warn(
+                f"real {data} requested (--synthetic not set) but no dataset "
+                f"found under {data_dir!r}; FALLING BACK TO SYNTHETIC DATA — "
+                f"metrics are NOT real-{data} results. Pass --synthetic to "
+                f"silence this, or place the dataset on disk.",
+                stacklevel=2)
         tn, vn = DEFAULT_SIZES[data]
         return _synthetic_pair(data, seed, train_n or tn, val_n or vn)
     elif data == 'fedemnist':

@@ -1,13 +1,19 @@
-"""IID data partitioner + index-view dataset (reference utils.py:39-92).
+"""IID data partitioner + index-view dataset (reference utils.py:39-92
+semantics, rebuilt).
 
-`distribute_data` reproduces the reference's shard math exactly: sort by
-label, bucket per class, strided-chunk each class list into slice_size
-shards of shard_size = len/(num_agents*class_per_agent), deal one shard per
-class to each agent (utils.py:58-92).  FMNIST/10 agents -> 6000/agent
-(600/class); CIFAR/40 agents -> 1250/agent (125/class) — asserted in tests.
+Partition model: the dataset is cut into `num_agents * class_per_agent`
+equal shards per the reference's math — ``shard_size =
+len(dataset) // (num_agents * class_per_agent)`` — built by slicing each
+class's index pool into strided sub-sequences, then dealing one shard of
+each class to every agent in class order.  FMNIST/10 agents -> 6000
+samples/agent (600/class); CIFAR/40 agents -> 1250/agent (125/class) —
+asserted in tests/test_partition.py.
+
+Device invariance: class pools are built with `nonzero` on a CPU copy of
+the labels (ascending index order, deterministic on every backend) rather
+than sorting — CUDA sort tie-ordering differs from CPU's, which would make
+partitions device-dependent.
 """
-
-from collections import defaultdict
 
 import torch
 
@@ -35,32 +41,31 @@ class DatasetSplit:
 
 
 def distribute_data(dataset, args, n_classes=10, class_per_agent=10):
-    if args.num_agents == 1:
+    """Deal class-balanced shards to agents.  Returns {agent: [indices]}."""
+    n_agents = args.num_agents
+    if n_agents == 1:
         return {0: list(range(len(dataset)))}
 
-    def chunker_list(seq, size):
-        return [seq[i::size] for i in range(size)]
+    shard_size = len(dataset) // (n_agents * class_per_agent)
+    shards_per_class = (len(dataset) // n_classes) // shard_size
 
-    labels_sorted = dataset.targets.sort()
-    class_by_labels = list(zip(labels_sorted.values.tolist(),
-                               labels_sorted.indices.tolist()))
-    labels_dict = defaultdict(list)
-    for label, idx in class_by_labels:
-        labels_dict[label].append(idx)
+    labels = dataset.targets.detach().cpu()
+    # per-class shard queues: class pool (ascending indices) strided into
+    # `shards_per_class` interleaved slices
+    queues = []
+    for c in range(n_classes):
+        pool = (labels == c).nonzero(as_tuple=True)[0].tolist()
+        queues.append([pool[s::shards_per_class]
+                       for s in range(shards_per_class)])
 
-    shard_size = len(dataset) // (args.num_agents * class_per_agent)
-    slice_size = (len(dataset) // n_classes) // shard_size
-    for k, v in labels_dict.items():
-        labels_dict[k] = chunker_list(v, slice_size)
-
-    dict_users = defaultdict(list)
-    for user_idx in range(args.num_agents):
-        class_ctr = 0
-        for j in range(n_classes):
-            if class_ctr == class_per_agent:
+    assignment = {}
+    for agent in range(n_agents):
+        mine, taken = [], 0
+        for c in range(n_classes):
+            if taken == class_per_agent:
                 break
-            elif len(labels_dict[j]) > 0:
-                dict_users[user_idx] += labels_dict[j][0]
-                del labels_dict[j % n_classes][0]
-                class_ctr += 1
-    return dict(dict_users)
+            if queues[c]:
+                mine.extend(queues[c].pop(0))
+                taken += 1
+        assignment[agent] = mine
+    return assignment

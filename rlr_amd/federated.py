@@ -181,7 +181,7 @@ def run(args, writer=None, progress=False):
     start_round = 0
     cum_poison_acc_mean = 0.0
     if args.resume:
-        state = load_checkpoint(args.resume, gm)
+        state = load_checkpoint(args.resume, gm, expect_args=args)
         start_round = state['round']
         cum_poison_acc_mean = state['cum_poison_acc_mean']
 

@@ -464,17 +464,24 @@ void gather_grads(torch::Tensor flat, std::vector<torch::Tensor> grads,
                   std::vector<int64_t> offsets) {
   CHK_CUDA(flat);
   CHK(grads.size() == offsets.size());
+  // any .contiguous() temporaries must outlive ALL async launches below
+  // (freeing them per-iteration would race the gather kernel when a grad
+  // is non-contiguous)
+  std::vector<torch::Tensor> held;
+  held.reserve(grads.size());
+  for (auto& g : grads) {
+    TORCH_CHECK(g.scalar_type() == torch::kFloat);
+    held.push_back(g.contiguous());
+  }
   size_t i = 0;
-  while (i < grads.size()) {
+  while (i < held.size()) {
     const float* srcs[16];
     long offs[16], lens[16];
     int cnt = 0;
-    for (; cnt < 16 && i < grads.size(); ++cnt, ++i) {
-      auto g = grads[i].contiguous();
-      TORCH_CHECK(g.scalar_type() == torch::kFloat);
-      srcs[cnt] = g.data_ptr<float>();
+    for (; cnt < 16 && i < held.size(); ++cnt, ++i) {
+      srcs[cnt] = held[i].data_ptr<float>();
       offs[cnt] = offsets[i];
-      lens[cnt] = g.numel();
+      lens[cnt] = held[i].numel();
     }
     launch_gather_grads(srcs, offs, lens, cnt, flat.data_ptr<float>(),
                         stream_of(flat));

@@ -31,11 +31,28 @@ def save_checkpoint(path, global_model, rnd, args, cum_poison_acc_mean):
     os.replace(tmp, path)
 
 
-def load_checkpoint(path, global_model=None):
+def load_checkpoint(path, global_model=None, expect_args=None):
+    """Load and (optionally) restore into `global_model`.  Refuses
+    checkpoints whose model/dataset/param-count don't match the run being
+    resumed — a silent mismatch would corrupt the weights at the copy."""
     state = torch.load(path, map_location='cpu', weights_only=False)
     assert state.get('version') == FORMAT_VERSION, \
         f"unknown checkpoint version {state.get('version')}"
+    if expect_args is not None:
+        saved = state.get('args', {})
+        for key in ('data', 'model'):
+            want = getattr(expect_args, key, None)
+            got = saved.get(key)
+            if got is not None and want is not None and got != want:
+                raise ValueError(
+                    f"checkpoint {path!r} was written for {key}={got!r}; "
+                    f"this run has {key}={want!r} — refusing to resume")
     if global_model is not None:
+        if state['params'].numel() != global_model.n_params:
+            raise ValueError(
+                f"checkpoint {path!r} holds {state['params'].numel()} params "
+                f"but the model has {global_model.n_params} — wrong "
+                f"model/dataset for this checkpoint")
         global_model.load_vector(
             state['params'].to(global_model.flat_params.device))
         if global_model.n_buffers and 'buffers' in state:

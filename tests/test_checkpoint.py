@@ -44,3 +44,24 @@ def test_checkpoint_contents(tmp_path, tiny_sizes):
     assert state['params'].shape == (1_199_882,)
     assert state['args']['data'] == 'fmnist'
     assert 'seed' in state
+
+
+def test_resume_refuses_config_mismatch(tmp_path, tiny_sizes):
+    """Resuming with a different model/dataset must fail loudly, not
+    silently corrupt the weights (checkpoint.load_checkpoint guard)."""
+    import pytest
+    from rlr_amd.federated import build_world
+    from rlr_amd.utils import load_checkpoint
+
+    a = _args(tmp_path, rounds=1)
+    run(a)
+    ck = os.path.join(str(tmp_path), 'round_000001.pt')
+
+    b = default_args(num_agents=2, rounds=2, snap=1, local_ep=1, bs=64,
+                     synthetic=True, no_tb=True, data='cifar10')
+    world = build_world(b)
+    with pytest.raises(ValueError, match='refusing to resume'):
+        load_checkpoint(ck, world['global_model'], expect_args=b)
+    # param-count guard fires even without expect_args
+    with pytest.raises(ValueError, match='params'):
+        load_checkpoint(ck, world['global_model'])

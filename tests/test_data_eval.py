@@ -103,3 +103,17 @@ def test_sample_agents_contract():
     assert all(0 <= a < 40 for a in s)
     assert s == sample_agents(42, 5, 40, 0.25)   # deterministic per round
     assert s != sample_agents(42, 6, 40, 0.25)   # varies across rounds
+
+
+def test_real_data_fallback_warns(tmp_path):
+    """Requesting real data with none on disk warns loudly before the
+    synthetic substitution (never a silent swap — ADVICE r1 medium)."""
+    import warnings
+    from rlr_amd.data.datasets import get_datasets
+    from rlr_amd.options import default_args
+    args = default_args(synthetic=False, data='fmnist')
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter('always')
+        get_datasets('fmnist', args, train_n=100, val_n=50,
+                     data_dir=str(tmp_path / 'nope'))
+    assert any('FALLING BACK TO SYNTHETIC' in str(x.message) for x in w)
