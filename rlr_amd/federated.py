@@ -109,10 +109,26 @@ def _replica_pool(world, args, gm):
     return world['replicas']
 
 
-def run_round(args, world, rnd, rank, world_size):
-    """One FL round; returns the stacked update matrix's agent id list."""
+def run_round(args, world, rnd, rank, world_size, phase_times=None):
+    """One FL round; returns the stacked update matrix's agent id list.
+
+    phase_times (optional dict): per-phase wall times (train/gather/
+    aggregate) are measured with a device sync at each boundary and
+    accumulated into it — only passed on snap rounds, so the syncs never
+    touch steady-state throughput."""
     gm = world['global_model']
     agents = world['agents']
+
+    def _mark(name, t0):
+        if phase_times is None:
+            return None
+        if gm.device.type == 'cuda':
+            torch.cuda.synchronize(gm.device)
+        t1 = time.perf_counter()
+        phase_times[name] = phase_times.get(name, 0.0) + (t1 - t0)
+        return t1
+
+    t0 = time.perf_counter() if phase_times is not None else None
 
     sampled = sample_agents(args.seed, rnd, args.num_agents, args.agent_frac)
     lo, hi, chunk = pdist.chunk_bounds(len(sampled), world_size, rank)
@@ -151,18 +167,22 @@ def run_round(args, world, rnd, rank, world_size):
             if local_buf is not None:
                 local_buf[slot].copy_(agents[agent_id].buffer_delta)
 
+    t0 = _mark('train', t0)
+
     n_valid = []
     for r in range(world_size):
         rlo, rhi, _ = pdist.chunk_bounds(len(sampled), world_size, r)
         n_valid.append(rhi - rlo)
     with trace_range('gather_updates'):
         stacked = pdist.all_gather_updates(local, n_valid, chunk)
+    t0 = _mark('gather', t0)
     with trace_range('aggregate'):
         world['aggregator'].aggregate_updates(gm, stacked, rnd,
                                               agent_ids=sampled)
     if local_buf is not None:
         buf_stacked = pdist.all_gather_updates(local_buf, n_valid, chunk)
         world['aggregator'].aggregate_buffers(gm, buf_stacked, sampled)
+    _mark('aggregate', t0)
     return sampled
 
 
@@ -196,19 +216,22 @@ def run(args, writer=None, progress=False):
         except ImportError:
             pass
 
+    shard = (rank, world_size) if world_size > 1 else None
     t_prev = time.perf_counter()
     for rnd in rounds_iter:
-        run_round(args, world, rnd, rank, world_size)
+        phases = {} if rnd % args.snap == 0 else None
+        run_round(args, world, rnd, rank, world_size, phase_times=phases)
 
         if rnd % args.snap == 0:
             t_now = time.perf_counter()
             rps = args.snap / (t_now - t_prev)
-            t_prev = t_now
             with torch.no_grad():
                 val_loss, (val_acc, val_pc) = get_loss_n_accuracy(
-                    gm, world['X_val'], world['Y_val'], args)
+                    gm, world['X_val'], world['Y_val'], args, shard=shard)
                 poison_loss, (poison_acc, _) = get_loss_n_accuracy(
-                    gm, world['X_pv'], world['Y_pv'], args)
+                    gm, world['X_pv'], world['Y_pv'], args, shard=shard)
+            phases['eval'] = time.perf_counter() - t_now
+            t_prev = time.perf_counter()
             cum_poison_acc_mean += poison_acc
             history['round'].append(rnd)
             history['val_acc'].append(val_acc)
@@ -227,6 +250,11 @@ def run(args, writer=None, progress=False):
                     writer.add_scalar('Poison/Cumulative_Poison_Accuracy_Mean',
                                       cum_poison_acc_mean / rnd, rnd)
                     writer.add_scalar('Perf/Rounds_Per_Sec', rps, rnd)
+                    # per-phase wall times, sampled on the snap round
+                    # (train/gather/aggregate from run_round, plus eval)
+                    for ph, secs in phases.items():
+                        writer.add_scalar(f'Perf/{ph.capitalize()}_ms',
+                                          secs * 1e3, rnd)
                 print(f'| Val_Loss/Val_Acc: {val_loss:.3f} / {val_acc:.3f} |')
                 print(f'| Val_Per_Class_Acc: {val_pc} ')
                 print(f'| Poison Loss/Poison Acc: '

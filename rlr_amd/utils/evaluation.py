@@ -13,19 +13,30 @@ from ..ops import ext, force_eager
 
 
 @torch.no_grad()
-def get_loss_n_accuracy(model, X, Y, args, num_classes=10, bs=None):
+def get_loss_n_accuracy(model, X, Y, args, num_classes=10, bs=None,
+                        shard=None):
     """Returns (avg_loss, (accuracy, per_class_accuracy)) — reference
-    signature parity, tensor inputs."""
+    signature parity, tensor inputs.
+
+    shard=(rank, world_size): batches are strided across ranks and the
+    (loss_sum, confusion) accumulators all-reduced, so eval wall time is
+    ~1/world_size (SURVEY.md §2c call-site 4).  Counts are exact integer
+    sums — accuracy/confusion are bitwise world-size-invariant; loss_sum
+    fp64 summation order differs across world sizes by at most an ulp."""
     model.eval()
     bs = bs or args.bs
     n = X.shape[0]
     device = X.device
     use_hip = X.is_cuda and not force_eager()
 
+    starts = list(range(0, n, bs))
+    if shard is not None and shard[1] > 1:
+        starts = starts[shard[0]::shard[1]]
+
     conf = torch.zeros(num_classes * num_classes, dtype=torch.float32,
                        device=device)
     loss_sum = torch.zeros(1, dtype=torch.float64, device=device)
-    for lo in range(0, n, bs):
+    for lo in starts:
         inputs, labels = X[lo:lo + bs], Y[lo:lo + bs]
         outputs = model(inputs)
         if outputs.dtype != torch.float32:
@@ -39,6 +50,10 @@ def get_loss_n_accuracy(model, X, Y, args, num_classes=10, bs=None):
             conf += torch.bincount(labels * num_classes + pred,
                                    minlength=num_classes * num_classes
                                    ).float()
+    if shard is not None and shard[1] > 1:
+        from ..parallel import dist as pdist
+        pdist.all_reduce_(conf)
+        pdist.all_reduce_(loss_sum)
     conf = conf.view(num_classes, num_classes).cpu()
     avg_loss = (loss_sum.item() / n) if n else 0.0
     correct = conf.diag().sum().item()

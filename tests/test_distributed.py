@@ -31,7 +31,8 @@ def _worker(rank, world_size, port, out_q):
         if rank == 0:
             # numpy copy: pickled by value (a torch tensor would ship a
             # shared-memory fd that dies with this process)
-            out_q.put(h['final_params'].numpy().copy())
+            out_q.put((h['final_params'].numpy().copy(),
+                       list(h['val_acc']), list(h['poison_acc'])))
     finally:
         pdist.teardown()
 
@@ -44,24 +45,28 @@ def _run_world(world_size):
              for r in range(world_size)]
     for p in procs:
         p.start()
-    result = q.get(timeout=300)
+    params, val_acc, poison_acc = q.get(timeout=300)
     for p in procs:
         p.join(timeout=120)
         assert p.exitcode == 0
-    return torch.from_numpy(result)
+    return torch.from_numpy(params), val_acc, poison_acc
 
 
 @pytest.mark.timeout(600)
 def test_world_size_invariance():
-    p1 = _run_world(1)
-    p2 = _run_world(2)
+    p1, v1, pa1 = _run_world(1)
+    p2, v2, pa2 = _run_world(2)
     assert torch.equal(p1, p2), (p1 - p2).abs().max()
+    # sharded eval (strided batches + all-reduced counts) must report the
+    # exact same accuracies as the single-process full pass
+    assert v1 == v2 and pa1 == pa2
 
 
 @pytest.mark.timeout(600)
 def test_three_ranks_uneven_chunks():
     """4 sampled agents over 3 ranks (2/2/0 after ceil-chunking — one rank
     idle) must still match the single-process result."""
-    p1 = _run_world(1)
-    p3 = _run_world(3)
+    p1, v1, _ = _run_world(1)
+    p3, v3, _ = _run_world(3)
     assert torch.equal(p1, p3)
+    assert v1 == v3
