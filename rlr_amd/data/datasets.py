@@ -227,14 +227,59 @@ def _try_real_available(data, data_dir):
     return os.path.isdir(os.path.join(data_dir, sub))
 
 
+# Native on-disk readers (torchvision layouts, no torchvision dependency —
+# it is not installed in the target image).  FashionMNIST ships as
+# idx-ubyte files (possibly still .gz); CIFAR-10 as python-pickled batches.
+
+def _read_idx(path):
+    """Parse an idx1/idx3 ubyte file (the (Fashion)MNIST on-disk format)."""
+    import gzip
+    opener = gzip.open if not os.path.exists(path) and \
+        os.path.exists(path + '.gz') else open
+    if opener is gzip.open:
+        path = path + '.gz'
+    with opener(path, 'rb') as f:
+        buf = f.read()
+    magic = int.from_bytes(buf[0:4], 'big')
+    ndim = magic & 0xFF
+    dims = [int.from_bytes(buf[4 + 4 * i: 8 + 4 * i], 'big')
+            for i in range(ndim)]
+    data = np.frombuffer(buf, dtype=np.uint8, offset=4 + 4 * ndim)
+    return torch.from_numpy(data.reshape(dims).copy())
+
+
+def _load_fmnist_raw(data_dir):
+    raw = os.path.join(data_dir, 'FashionMNIST', 'raw')
+    out = []
+    for split in ('train', 't10k'):
+        imgs = _read_idx(os.path.join(raw, f'{split}-images-idx3-ubyte'))
+        lbls = _read_idx(os.path.join(raw, f'{split}-labels-idx1-ubyte'))
+        out.append(ArrayDataset(imgs, lbls.long(), 'fmnist'))
+    return tuple(out)
+
+
+def _load_cifar10_raw(data_dir):
+    import pickle
+    base = os.path.join(data_dir, 'cifar-10-batches-py')
+
+    def batches(names):
+        xs, ys = [], []
+        for name in names:
+            with open(os.path.join(base, name), 'rb') as f:
+                d = pickle.load(f, encoding='bytes')
+            xs.append(np.asarray(d[b'data'], dtype=np.uint8))
+            ys.extend(d[b'labels'])
+        x = np.concatenate(xs).reshape(-1, 3, 32, 32).transpose(0, 2, 3, 1)
+        return (ArrayDataset(torch.from_numpy(np.ascontiguousarray(x)),
+                             torch.tensor(ys), 'cifar10'))
+
+    train = batches([f'data_batch_{i}' for i in range(1, 6)
+                     if os.path.exists(os.path.join(base, f'data_batch_{i}'))])
+    val = batches(['test_batch'])
+    return train, val
+
+
 def _load_real(data, data_dir):
-    from torchvision import datasets as tvd
     if data == 'fmnist':
-        tr = tvd.FashionMNIST(data_dir, train=True, download=False)
-        va = tvd.FashionMNIST(data_dir, train=False, download=False)
-        return (ArrayDataset(tr.data, tr.targets, 'fmnist'),
-                ArrayDataset(va.data, va.targets, 'fmnist'))
-    tr = tvd.CIFAR10(data_dir, train=True, download=False)
-    va = tvd.CIFAR10(data_dir, train=False, download=False)
-    return (ArrayDataset(torch.from_numpy(tr.data), torch.tensor(tr.targets), 'cifar10'),
-            ArrayDataset(torch.from_numpy(va.data), torch.tensor(va.targets), 'cifar10'))
+        return _load_fmnist_raw(data_dir)
+    return _load_cifar10_raw(data_dir)

@@ -117,3 +117,80 @@ def test_real_data_fallback_warns(tmp_path):
         get_datasets('fmnist', args, train_n=100, val_n=50,
                      data_dir=str(tmp_path / 'nope'))
     assert any('FALLING BACK TO SYNTHETIC' in str(x.message) for x in w)
+
+
+def _write_idx(path, arr):
+    import numpy as np
+    arr = np.asarray(arr, dtype=np.uint8)
+    with open(path, 'wb') as f:
+        f.write((0x800 | arr.ndim).to_bytes(4, 'big'))
+        for d in arr.shape:
+            f.write(int(d).to_bytes(4, 'big'))
+        f.write(arr.tobytes())
+
+
+def test_load_real_fmnist_fixture(tmp_path):
+    """Drive _load_real end-to-end against a tiny on-disk fixture in the
+    torchvision FashionMNIST raw layout (idx-ubyte files)."""
+    import numpy as np
+    from rlr_amd.data.datasets import get_datasets
+    from rlr_amd.options import default_args
+    raw = tmp_path / 'FashionMNIST' / 'raw'
+    raw.mkdir(parents=True)
+    rng = np.random.default_rng(0)
+    tr_x = rng.integers(0, 256, (40, 28, 28)).astype('uint8')
+    tr_y = (np.arange(40) % 10).astype('uint8')
+    va_x = rng.integers(0, 256, (20, 28, 28)).astype('uint8')
+    va_y = (np.arange(20) % 10).astype('uint8')
+    _write_idx(raw / 'train-images-idx3-ubyte', tr_x)
+    _write_idx(raw / 'train-labels-idx1-ubyte', tr_y)
+    _write_idx(raw / 't10k-images-idx3-ubyte', va_x)
+    _write_idx(raw / 't10k-labels-idx1-ubyte', va_y)
+
+    args = default_args(synthetic=False, data='fmnist')
+    train, val = get_datasets('fmnist', args, data_dir=str(tmp_path))
+    assert len(train) == 40 and len(val) == 20
+    assert (train.data.numpy() == tr_x).all()
+    assert train.targets.tolist() == tr_y.tolist()
+    x, t = train[3]
+    assert x.shape == (1, 28, 28) and t == 3
+    # gz-only files load too (torchvision sometimes leaves only .gz)
+    import gzip, os
+    plain = raw / 't10k-images-idx3-ubyte'
+    data = plain.read_bytes()
+    os.remove(plain)
+    with gzip.open(str(plain) + '.gz', 'wb') as f:
+        f.write(data)
+    _, val2 = get_datasets('fmnist', args, data_dir=str(tmp_path))
+    assert (val2.data.numpy() == va_x).all()
+
+
+def test_load_real_cifar10_fixture(tmp_path):
+    """CIFAR-10 pickled-batch layout fixture through _load_real."""
+    import numpy as np, pickle
+    from rlr_amd.data.datasets import get_datasets
+    from rlr_amd.options import default_args
+    base = tmp_path / 'cifar-10-batches-py'
+    base.mkdir(parents=True)
+    rng = np.random.default_rng(1)
+
+    def write(name, n):
+        x = rng.integers(0, 256, (n, 3072)).astype('uint8')
+        y = (np.arange(n) % 10).tolist()
+        with open(base / name, 'wb') as f:
+            pickle.dump({b'data': x, b'labels': y}, f)
+        return x, y
+
+    tx1, ty1 = write('data_batch_1', 30)
+    tx2, ty2 = write('data_batch_2', 30)
+    vx, vy = write('test_batch', 20)
+
+    args = default_args(synthetic=False, data='cifar10')
+    train, val = get_datasets('cifar10', args, data_dir=str(tmp_path))
+    assert len(train) == 60 and len(val) == 20
+    assert train.data.shape == (60, 32, 32, 3)   # HWC raw storage
+    want = np.concatenate([tx1, tx2]).reshape(-1, 3, 32, 32).transpose(0, 2, 3, 1)
+    assert (train.data.numpy() == want).all()
+    assert train.targets.tolist() == ty1 + ty2
+    x, t = val[5]
+    assert x.shape == (3, 32, 32) and t == vy[5]
