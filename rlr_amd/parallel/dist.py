@@ -23,8 +23,10 @@ def setup(args=None):
     Returns (rank, world_size)."""
     if dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
-    if 'RANK' in os.environ and 'WORLD_SIZE' in os.environ \
-            and int(os.environ['WORLD_SIZE']) > 1:
+    # any torchrun launch (RANK/WORLD_SIZE in env) initializes the group —
+    # including world_size 1, so a single-GPU lease exercises the real
+    # RCCL init + collectives rather than the uninitialized shortcut
+    if 'RANK' in os.environ and 'WORLD_SIZE' in os.environ:
         backend = 'nccl' if torch.cuda.is_available() else 'gloo'
         if torch.cuda.is_available():
             torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
@@ -69,7 +71,10 @@ def all_gather_updates(local: torch.Tensor, n_valid_per_rank, chunk: int):
     valid counts.  Single-process: returns local's valid rows."""
     rank, world = rank_world()
     if world == 1:
-        return local[:n_valid_per_rank[0]]
+        if not dist.is_initialized():
+            return local[:n_valid_per_rank[0]]
+        # ws=1 under an initialized group still issues the real collective
+        # (a self-copy) so single-GPU runs exercise the RCCL path end-to-end
     out = torch.empty(world * chunk, local.shape[1], dtype=local.dtype,
                       device=local.device)
     dist.all_gather_into_tensor(out, local)
