@@ -115,3 +115,26 @@ if [ "$WHAT" = "curves" ]; then
     grep -E "Val_Loss|Poison Loss" gpurun_out/curve_$TAG.log | tail -4
   done
 fi
+
+if [ "$WHAT" = "san" ]; then
+  # sanitizer-class pass: the full GPU kernel suite with
+  # serialized kernel launches + blocking copies, so any async fault
+  # (OOB, bad address) surfaces at the exact offending launch instead of
+  # a later sync point.  Kept log: gpurun_out/ci_san.log
+  AMD_SERIALIZE_KERNEL=3 AMD_SERIALIZE_COPY=3 HIP_LAUNCH_BLOCKING=1 \
+    timeout 900 python -m pytest tests/test_kernels_gpu.py \
+    tests/test_bf16_gpu.py tests/test_bf16_conv_gpu.py -q \
+    > gpurun_out/ci_san.log 2>&1
+  echo "san rc=$?" >> gpurun_out/ci_san.log
+  tail -4 gpurun_out/ci_san.log
+fi
+
+if [ "$WHAT" = "defense" ]; then
+  # the full-scale FMNIST defense curves as an
+  # ASSERTED CI target with tight bands (attack >=0.95 backdoor without
+  # defense, <=0.05 with RLR theta=4; val accs within 0.05), written to
+  # gpurun_out/defense_ci.json for tracking under profiles/.
+  timeout 1500 python scripts/defense_ci.py > gpurun_out/defense_ci.log 2>&1
+  echo "defense rc=$?"
+  tail -4 gpurun_out/defense_ci.log
+fi
