@@ -90,7 +90,7 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
 
   // per-thread m -> (nb, oh, ow), once
   int ow0[BM / 32], oh0[BM / 32];
-  long base[BM / 32];  // nb * H*W*C
+  long base[BM / 32];  // V4: patch base incl. tap 0 + ak; else nb*H*W*C
   bool mval[BM / 32];
 #pragma unroll
   for (int j = 0; j < BM / 32; ++j) {
@@ -103,25 +103,46 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
     ow0[j] = ow * sh.stride - sh.pad;
     oh0[j] = oh * sh.stride - sh.pad;
     base[j] = nb * (long)sh.H * sh.W * sh.C;
+    if (V4)  // fold the per-thread patch origin into the base: the gather
+             // address becomes base[j] + tap_off (ONE add per load)
+      base[j] += ((long)oh0[j] * sh.W + ow0[j]) * sh.C + ak;
   }
+
+  // V4 staged-tap state: stage_load is called with strictly ascending k0
+  // (0, BK, 2BK, ...), so the (r,s,c)->offset decomposition is carried
+  // incrementally — no div/mod in the loop.  tap_off = (r*W+s)*C + cb.
+  int tr = 0, ts = 0, tcb = 0;
+  long tap_off = 0;
+  const bool wt_v4 = (sh.Kout % 4) == 0 && n_blk + bn + 3 < sh.Kout;
+  const bool wt_any = n_blk + bn < sh.Kout;
+  long wt_base = (long)bk * sh.Kout + n_blk + bn;  // advances BK*Kout/call
+  const long wt_j16 = 16L * sh.Kout;
 
   float ra[BM / 32][4];
   float4 rb[2];
   auto stage_load = [&](int k0) {
     if (V4) {
-      // one tap per 32-wide K-tile: decompose once per tile
-      int rs = k0 / sh.C;
-      int r = rs / sh.S, s = rs % sh.S;
-      int c0 = k0 - rs * sh.C + ak;
+      const int r = tr, s = ts;
+      const long toff = tap_off;
 #pragma unroll
       for (int j = 0; j < BM / 32; ++j) {
-        int ih = oh0[j] + r, iw = ow0[j] + s;
         float4 q = {0.f, 0.f, 0.f, 0.f};
-        if (mval[j] &&
-            (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
-          q = *(const float4*)(x + base[j] +
-                               ((long)ih * sh.W + iw) * sh.C + c0);
+        bool ok = mval[j];
+        if (!P0) {
+          int ih = oh0[j] + r, iw = ow0[j] + s;
+          ok = ok && (unsigned)ih < (unsigned)sh.H &&
+               (unsigned)iw < (unsigned)sh.W;
+        }
+        if (ok) q = *(const float4*)(x + base[j] + toff);
         ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      }
+      // advance tap state to the next K-tile (C % 32 == 0 in V4: a c-
+      // rollover lands exactly on tile boundaries)
+      tcb += BK; tap_off += BK;
+      if (tcb == sh.C) {
+        tcb = 0;
+        if (++ts == sh.S) { ts = 0; ++tr;
+                            tap_off += (long)(sh.W - sh.S) * sh.C; }
       }
     } else {
 #pragma unroll
@@ -149,10 +170,10 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
       int gk = k0 + bk + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
-        const float* src = wt + (long)gk * sh.Kout + n_blk + bn;
-        if (n_blk + bn + 3 < sh.Kout && (sh.Kout % 4) == 0)
+        const float* src = wt + wt_base + j * wt_j16;
+        if (wt_v4)
           q = *(const float4*)src;
-        else {
+        else if (wt_any) {
           if (n_blk + bn + 0 < sh.Kout) q.x = src[0];
           if (n_blk + bn + 1 < sh.Kout) q.y = src[1];
           if (n_blk + bn + 2 < sh.Kout) q.z = src[2];
@@ -161,6 +182,7 @@ void conv_fwd_k(const float* __restrict__ x, const float* __restrict__ wt,
       }
       rb[j] = q;
     }
+    wt_base += (long)BK * sh.Kout;
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
@@ -274,6 +296,10 @@ void conv_fwd_res_k(const float* __restrict__ x,
 
   // ---- B staging (T14 split, double-buffered) ----
   const int bk = t >> 4, bn = (t & 15) * 4;
+  const bool wt_v4 = (sh.Kout % 4) == 0 && n_blk + bn + 3 < sh.Kout;
+  const bool wt_any = n_blk + bn < sh.Kout;
+  long wt_base = (long)bk * sh.Kout + n_blk + bn;
+  const long wt_j16 = 16L * sh.Kout;
   float4 rb[2];
   auto stage_loadB = [&](int k0) {
 #pragma unroll
@@ -281,10 +307,10 @@ void conv_fwd_res_k(const float* __restrict__ x,
       int gk = k0 + bk + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
-        const float* srcp = wt + (long)gk * sh.Kout + n_blk + bn;
-        if (n_blk + bn + 3 < sh.Kout && (sh.Kout % 4) == 0)
+        const float* srcp = wt + wt_base + j * wt_j16;
+        if (wt_v4)
           q = *(const float4*)srcp;
-        else {
+        else if (wt_any) {
           if (n_blk + bn + 0 < sh.Kout) q.x = srcp[0];
           if (n_blk + bn + 1 < sh.Kout) q.y = srcp[1];
           if (n_blk + bn + 2 < sh.Kout) q.z = srcp[2];
@@ -293,6 +319,7 @@ void conv_fwd_res_k(const float* __restrict__ x,
       }
       rb[j] = q;
     }
+    wt_base += (long)BK * sh.Kout;
   };
   auto stage_writeB = [&](int buf) {
 #pragma unroll
@@ -306,16 +333,14 @@ void conv_fwd_res_k(const float* __restrict__ x,
   if (BK < Kdim) stage_loadB(BK);
   __syncthreads();
   int buf = 0;
+  // incremental tap state (one tap per 32-wide K-tile; C % 32 == 0 here):
+  // tap = (r*W + s)*Cp + c0 + l4, carried across iterations — no div/mod
+  int ts = 0, tcb = 0, tap = l4;
   for (int k0 = 0; k0 < Kdim; k0 += BK) {
     if (k0 + BK < Kdim) {
       stage_writeB(buf ^ 1);
       if (k0 + 2 * BK < Kdim) stage_loadB(k0 + 2 * BK);
     }
-    // one tap per 32-wide K-tile
-    int rs = k0 / sh.C;
-    int r = rs / sh.S, s = rs % sh.S;
-    int c0 = k0 - rs * sh.C;
-    int tap = (r * sh.W + s) * Cp + c0 + l4;
     const float* Bbuf = &B_lds0[buf * BK * LDB_S];
 #pragma unroll
     for (int kk = 0; kk < BK / 4; ++kk) {
@@ -333,6 +358,11 @@ void conv_fwd_res_k(const float* __restrict__ x,
         for (int ni = 0; ni < NI; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
               a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    tcb += BK; tap += BK;
+    if (tcb == sh.C) {
+      tcb = 0; tap += Cp - sh.C;  // +1: next s column in the padded row
+      if (++ts == sh.S) { ts = 0; tap += (sh.W - sh.S) * Cp; }
     }
     __syncthreads();
     buf ^= 1;
@@ -385,7 +415,7 @@ void conv_bwd_data_k(const float* __restrict__ dy,
   const int stride = ST > 0 ? ST : sh.stride;
 
   int iwp[BM / 32], ihp[BM / 32];
-  long base[BM / 32];  // nb * OH*OW*KO
+  long base[BM / 32];  // ST==1&&V4: folded patch base; else nb*OH*OW*KO
   bool mval[BM / 32];
 #pragma unroll
   for (int j = 0; j < BM / 32; ++j) {
@@ -395,27 +425,57 @@ void conv_bwd_data_k(const float* __restrict__ dy,
     iwp[j] = (int)(gmc % sh.W) + sh.pad;
     ihp[j] = (int)((gmc / sh.W) % sh.H) + sh.pad;
     base[j] = (gmc / ((long)sh.W * sh.H)) * (long)sh.OH * sh.OW * sh.Kout;
+    if (ST == 1 && V4)  // gather address = base[j] - tile_off (one sub)
+      base[j] += ((long)ihp[j] * sh.OW + iwp[j]) * sh.Kout + ak;
   }
+
+  // staged-tap state, advanced per stage_load call (ascending k0; KO%32==0
+  // in V4 so ko-rollover lands on tile boundaries): (r,s) of the tap plus,
+  // for ST==1, the folded offset noff = (r*OW+s)*KO - kb.
+  int tr = 0, ts = 0, tkb = 0;
+  long noff = 0;
+  const bool wp_v4 = (sh.C % 4) == 0 && n_blk + bn + 3 < sh.C;
+  const bool wp_any = n_blk + bn < sh.C;
+  long wp_base = (long)bk * sh.C + n_blk + bn;
+  const long wp_j16 = 16L * sh.C;
 
   float ra[BM / 32][4];
   float4 rb[2];
   auto stage_load = [&](int k0) {
     if (V4) {
-      int rs = k0 / sh.Kout;
-      int r = rs / sh.S, s = rs % sh.S;
-      int ko0 = k0 - rs * sh.Kout + ak;
+      const int r = tr, s = ts;
+      if (ST == 1) {
+        const long toff = noff;
 #pragma unroll
-      for (int j = 0; j < BM / 32; ++j) {
-        int ohn = ihp[j] - r, own = iwp[j] - s;
-        float4 q = {0.f, 0.f, 0.f, 0.f};
-        if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
-            own % stride == 0) {
-          int oh = ohn / stride, ow = own / stride;
-          if (oh < sh.OH && ow < sh.OW)
-            q = *(const float4*)(dy + base[j] +
-                                 ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+        for (int j = 0; j < BM / 32; ++j) {
+          int ohn = ihp[j] - r, own = iwp[j] - s;
+          float4 q = {0.f, 0.f, 0.f, 0.f};
+          if (mval[j] && (unsigned)ohn < (unsigned)sh.OH &&
+              (unsigned)own < (unsigned)sh.OW)
+            q = *(const float4*)(dy + base[j] - toff);
+          ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
         }
-        ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      } else {
+        const int ko0 = tkb + ak;
+#pragma unroll
+        for (int j = 0; j < BM / 32; ++j) {
+          int ohn = ihp[j] - r, own = iwp[j] - s;
+          float4 q = {0.f, 0.f, 0.f, 0.f};
+          if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+              own % stride == 0) {
+            int oh = ohn / stride, ow = own / stride;
+            if (oh < sh.OH && ow < sh.OW)
+              q = *(const float4*)(dy + base[j] +
+                                   ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+          }
+          ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+        }
+      }
+      tkb += BK; noff -= BK;
+      if (tkb == sh.Kout) {
+        tkb = 0; noff += 2L * sh.Kout;
+        if (++ts == sh.S) { ts = 0; ++tr;
+                            noff += (long)(sh.OW - sh.S) * sh.Kout; }
       }
     } else {
 #pragma unroll
@@ -447,10 +507,10 @@ void conv_bwd_data_k(const float* __restrict__ dy,
       int gk = k0 + bk + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
-        const float* src = wp + (long)gk * sh.C + n_blk + bn;
-        if (n_blk + bn + 3 < sh.C && (sh.C % 4) == 0)
+        const float* src = wp + wp_base + j * wp_j16;
+        if (wp_v4)
           q = *(const float4*)src;
-        else {
+        else if (wp_any) {
           if (n_blk + bn + 0 < sh.C) q.x = src[0];
           if (n_blk + bn + 1 < sh.C) q.y = src[1];
           if (n_blk + bn + 2 < sh.C) q.z = src[2];
@@ -459,6 +519,7 @@ void conv_bwd_data_k(const float* __restrict__ dy,
       }
       rb[j] = q;
     }
+    wp_base += (long)BK * sh.C;
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
@@ -532,7 +593,7 @@ void conv_bwd_data32_k(const float* __restrict__ dy,
   const int stride = ST > 0 ? ST : sh.stride;
 
   int iwp[BM / 32], ihp[BM / 32];
-  long base[BM / 32];
+  long base[BM / 32];  // ST==1&&V4: folded patch base (see bwd_data_k)
   bool mval[BM / 32];
 #pragma unroll
   for (int j = 0; j < BM / 32; ++j) {
@@ -542,27 +603,50 @@ void conv_bwd_data32_k(const float* __restrict__ dy,
     iwp[j] = (int)(gmc % sh.W) + sh.pad;
     ihp[j] = (int)((gmc / sh.W) % sh.H) + sh.pad;
     base[j] = (gmc / ((long)sh.W * sh.H)) * (long)sh.OH * sh.OW * sh.Kout;
+    if (ST == 1 && V4)
+      base[j] += ((long)ihp[j] * sh.OW + iwp[j]) * sh.Kout + ak;
   }
+
+  int tr = 0, ts = 0, tkb = 0;
+  long noff = 0;
 
   float ra[BM / 32][4];
   float4 rb;
   auto stage_load = [&](int k0) {
     if (V4) {
-      int rs = k0 / sh.Kout;
-      int r = rs / sh.S, s = rs % sh.S;
-      int ko0 = k0 - rs * sh.Kout + ak;
+      const int r = tr, s = ts;
+      if (ST == 1) {
+        const long toff = noff;
 #pragma unroll
-      for (int j = 0; j < BM / 32; ++j) {
-        int ohn = ihp[j] - r, own = iwp[j] - s;
-        float4 q = {0.f, 0.f, 0.f, 0.f};
-        if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
-            own % stride == 0) {
-          int oh = ohn / stride, ow = own / stride;
-          if (oh < sh.OH && ow < sh.OW)
-            q = *(const float4*)(dy + base[j] +
-                                 ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+        for (int j = 0; j < BM / 32; ++j) {
+          int ohn = ihp[j] - r, own = iwp[j] - s;
+          float4 q = {0.f, 0.f, 0.f, 0.f};
+          if (mval[j] && (unsigned)ohn < (unsigned)sh.OH &&
+              (unsigned)own < (unsigned)sh.OW)
+            q = *(const float4*)(dy + base[j] - toff);
+          ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
         }
-        ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+      } else {
+        const int ko0 = tkb + ak;
+#pragma unroll
+        for (int j = 0; j < BM / 32; ++j) {
+          int ohn = ihp[j] - r, own = iwp[j] - s;
+          float4 q = {0.f, 0.f, 0.f, 0.f};
+          if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+              own % stride == 0) {
+            int oh = ohn / stride, ow = own / stride;
+            if (oh < sh.OH && ow < sh.OW)
+              q = *(const float4*)(dy + base[j] +
+                                   ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+          }
+          ra[j][0] = q.x; ra[j][1] = q.y; ra[j][2] = q.z; ra[j][3] = q.w;
+        }
+      }
+      tkb += BK; noff -= BK;
+      if (tkb == sh.Kout) {
+        tkb = 0; noff += 2L * sh.Kout;
+        if (++ts == sh.S) { ts = 0; ++tr;
+                            noff += (long)(sh.OW - sh.S) * sh.Kout; }
       }
     } else {
 #pragma unroll
@@ -594,9 +678,9 @@ void conv_bwd_data32_k(const float* __restrict__ dy,
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (gk < Kdim) {
         const float* src = wp + (long)gk * sh.C + n_blk + bn;
-        if (n_blk + bn + 3 < sh.C && (sh.C % 4) == 0)
+        if ((sh.C % 4) == 0 && n_blk + bn + 3 < sh.C)
           q = *(const float4*)src;
-        else {
+        else if (n_blk + bn < sh.C) {
           if (n_blk + bn + 0 < sh.C) q.x = src[0];
           if (n_blk + bn + 1 < sh.C) q.y = src[1];
           if (n_blk + bn + 2 < sh.C) q.z = src[2];
@@ -698,6 +782,49 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
     bc0_ = crs - rs * sh.C;
   }
 
+  // hoisted thread-constant bounds + A-row base (advances BK*Kout/call)
+  const bool a_v4 = (sh.Kout % 4) == 0 && m_blk + ako + 3 < sh.Kout;
+  const bool b_v4q = n_blk + bn4 + 3 < Ncrs;
+  const bool b_any = n_blk + bn4 < Ncrs;
+  long a_base = (k_lo + amr) * (long)sh.Kout + m_blk + ako;
+  const long a_j16 = 16L * sh.Kout;
+
+  // per-j B-row pixel state, carried incrementally (+BK rows per call):
+  // the x gather address is LINEAR in (nb, oh, ow), so one carry loop
+  // replaces the two 64-bit div/mod chains per row per tile.
+  const long strideC = (long)sh.stride * sh.C;
+  const long strideWC = (long)sh.stride * sh.W * sh.C;
+  const long HWC = (long)sh.H * sh.W * sh.C;
+  int bow[2], boh[2];
+  long baddr[2];  // x + baddr[j] = this row's (ih,iw) patch element
+  {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      long k = k_lo + bmr + j * 16;
+      bow[j] = (int)(k % sh.OW);
+      boh[j] = (int)((k / sh.OW) % sh.OH);
+      long nb = k / ((long)sh.OW * sh.OH);
+      baddr[j] = nb * HWC +
+                 ((long)(boh[j] * sh.stride - sh.pad + br_) * sh.W +
+                  (bow[j] * sh.stride - sh.pad + bs_)) * sh.C + bc0_;
+    }
+  }
+  auto b_advance = [&]() {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      bow[j] += BK;
+      baddr[j] += BK * strideC;
+      while (bow[j] >= sh.OW) {
+        bow[j] -= sh.OW;
+        baddr[j] += strideWC - (long)sh.OW * strideC;
+        if (++boh[j] == sh.OH) {
+          boh[j] = 0;
+          baddr[j] += HWC - (long)sh.OH * strideWC;
+        }
+      }
+    }
+  };
+
   float raA[2][4];
   float4 rbB[2];
   auto stage_load = [&](long k0) {
@@ -706,41 +833,43 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
       long k = k0 + amr + j * 16;  // the m index
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (k < k_hi) {
-        if ((sh.Kout % 4) == 0 && m_blk + ako + 3 < sh.Kout) {
-          q = *(const float4*)(dy + k * sh.Kout + m_blk + ako);
+        const float* src = dy + a_base + j * a_j16;
+        if (a_v4) {
+          q = *(const float4*)src;
         } else {
 #pragma unroll
           for (int e = 0; e < 4; ++e)
-            if (m_blk + ako + e < sh.Kout)
-              (&q.x)[e] = dy[k * sh.Kout + m_blk + ako + e];
+            if (m_blk + ako + e < sh.Kout) (&q.x)[e] = src[e];
         }
       }
       raA[j][0] = q.x; raA[j][1] = q.y; raA[j][2] = q.z; raA[j][3] = q.w;
     }
+    a_base += (long)BK * sh.Kout;
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       long k = k0 + bmr + j * 16;
       float4 q = {0.f, 0.f, 0.f, 0.f};
       if (k < k_hi) {
-        int ow = k % sh.OW;
-        int oh = (k / sh.OW) % sh.OH;
-        long nb = k / ((long)sh.OW * sh.OH);
-        long xb = nb * (long)sh.H * sh.W * sh.C;
         if (V4) {
-          int ih = oh * sh.stride - sh.pad + br_;
-          int iw = ow * sh.stride - sh.pad + bs_;
-          if (n_blk + bn4 + 3 < Ncrs &&
-              (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
-            q = *(const float4*)(x + xb + ((long)ih * sh.W + iw) * sh.C +
-                                 bc0_);
-          else if (n_blk + bn4 < Ncrs &&
-                   (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W))) {
+          bool inb = P0;
+          if (!P0) {
+            int ih = boh[j] * sh.stride - sh.pad + br_;
+            int iw = bow[j] * sh.stride - sh.pad + bs_;
+            inb = (unsigned)ih < (unsigned)sh.H &&
+                  (unsigned)iw < (unsigned)sh.W;
+          }
+          if (b_v4q && inb)
+            q = *(const float4*)(x + baddr[j]);
+          else if (b_any && inb) {
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              if (n_blk + bn4 + e < Ncrs)
-                (&q.x)[e] = x[xb + ((long)ih * sh.W + iw) * sh.C + bc0_ + e];
+              if (n_blk + bn4 + e < Ncrs) (&q.x)[e] = x[baddr[j] + e];
           }
         } else {
+          int ow = bow[j], oh = boh[j];
+          long xb = baddr[j];  // recover nb*HWC from the linear address
+          xb -= ((long)(oh * sh.stride - sh.pad + br_) * sh.W +
+                 (ow * sh.stride - sh.pad + bs_)) * sh.C + bc0_;
 #pragma unroll
           for (int e = 0; e < 4; ++e) {
             int crs = n_blk + bn4 + e;
@@ -758,6 +887,7 @@ void conv_bwd_weight_k(const float* __restrict__ dy,
       }
       rbB[j] = q;
     }
+    b_advance();
   };
   auto stage_write = [&](int buf) {
     // A transposed: element (m, ko+i) -> A_lds[ko+i][m]
