@@ -106,10 +106,16 @@ class Agent:
                     engine.step(self._X, self._Y, sel)
                     continue
                 inputs, labels = self._X[sel], self._Y[sel]
-                gm.zero_grad()
-                outputs = gm(inputs)
-                loss = Fo.cross_entropy(outputs, labels)
-                loss.backward()
+                model = gm.model
+                if (self.device.type == 'cuda' and not Fo.force_eager()
+                        and getattr(model, 'manual_step', None) is not None
+                        and model.compute_dtype is None):
+                    model.manual_step(inputs, labels, gm.dloss_ones())
+                else:
+                    gm.zero_grad()
+                    outputs = gm(inputs)
+                    loss = Fo.cross_entropy(outputs, labels)
+                    loss.backward()
                 flat_ops.clipped_sgd_step_(gm.flat_params, gm.flat_grads,
                                            gm.momentum, args.client_lr,
                                            args.client_moment, 10.0)

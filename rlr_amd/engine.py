@@ -37,19 +37,22 @@ class TrainEngine:
     # ------------------------------------------------------------- step
 
     def _step_body(self, static_x, static_y):
-        # NOTE on a rejected optimization: detaching p.grad so autograd
-        # ADOPTS fresh gradients (then gathering them with ONE kernel,
-        # ops ext.gather_grads) removes zero_grad + the 8..62 per-param
-        # accumulate-adds (~7% of a CNN step), but torch's allocator
-        # aborts the hipGraph capture when autograd allocates adopted
-        # grads inside the capture window on ROCm 7.0 — revisit with a
-        # newer torch.  The preset-view + accumulate scheme below is
-        # capture-stable.
+        # Preferred path: the model's hand-rolled fwd+bwd (manual_step)
+        # writes each weight/bias grad DIRECTLY into its flat_grads view —
+        # no zero-grad fill and no autograd accumulate-add kernels, with a
+        # bitwise-identical kernel stream otherwise (assignment ==
+        # accumulate-into-zero).  The autograd fallback below remains for
+        # models without a manual tape (bf16 ResNet).
         gm, args = self.gm, self.args
-        gm.flat_grads.zero_()
-        out = gm(static_x)
-        loss = Fo.cross_entropy(out, static_y)
-        loss.backward()
+        model = gm.model
+        if (getattr(model, 'manual_step', None) is not None
+                and model.compute_dtype is None):
+            model.manual_step(static_x, static_y, gm.dloss_ones())
+        else:
+            gm.flat_grads.zero_()
+            out = gm(static_x)
+            loss = Fo.cross_entropy(out, static_y)
+            loss.backward()
         flat_ops.clipped_sgd_step_(gm.flat_params, gm.flat_grads,
                                    gm.momentum, args.client_lr,
                                    args.client_moment, 10.0)

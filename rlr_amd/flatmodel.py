@@ -100,6 +100,12 @@ class FlatParamModel:
     def set_dropout_seed(self, seed):
         self.model.set_dropout_seed(seed)
 
+    def dloss_ones(self):
+        """Static d(loss) seed tensor for the manual-tape backward."""
+        if not hasattr(self, '_dloss'):
+            self._dloss = torch.ones(1, device=self.device)
+        return self._dloss
+
     def get_engine(self, args):
         """Cached hipGraph TrainEngine (engine.py) for this model."""
         if not hasattr(self, '_engine') or self._engine is None:

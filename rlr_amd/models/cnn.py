@@ -56,6 +56,43 @@ class CNN_MNIST(_OpsModel):
         x = Fo.linear(x, self.fc2.weight, self.fc2.bias)
         return x
 
+    def manual_step(self, x, labels, dloss):
+        """Hand-rolled fwd+bwd (fp32 GPU): identical kernel sequence to
+        the autograd path, but weight/bias grads are written DIRECTLY into
+        the preset p.grad flat-views (assignment == accumulate-into-zero,
+        bitwise) — no zero-grad fill, no autograd accumulate-adds."""
+        from ..ops import ext
+        E = ext()
+        p = self.p_drop
+        st = self.rng.gpu_state(x.device)
+        a1 = E.conv2d_fwd(x, self.conv1.weight, self.conv1.bias, 1, 0, True)
+        a2 = E.conv2d_fwd(a1, self.conv2.weight, self.conv2.bias, 1, 0, True)
+        pl, idx = E.maxpool2x2_fwd(a2)
+        fl = E.nhwc_flatten(pl)
+        d1, m1 = E.dropout_fwd_dev(fl, p, st, 0)
+        h1 = E.linear_fwd(d1, self.fc1.weight, self.fc1.bias, True)
+        d2, m2 = E.dropout_fwd_dev(h1, p, st, 1)
+        out = E.linear_fwd(d2, self.fc2.weight, self.fc2.bias, False)
+        loss, softmax = E.cross_entropy_fwd(out, labels)
+
+        g = E.cross_entropy_bwd(softmax, labels, dloss)
+        g = E.linear_bwd_into(d2, self.fc2.weight, g,
+                              self.fc2.weight.grad, self.fc2.bias.grad, True)
+        g = E.dropout_bwd(g, m2, p)
+        g = E.relu_bwd(h1, g)
+        g = E.linear_bwd_into(d1, self.fc1.weight, g,
+                              self.fc1.weight.grad, self.fc1.bias.grad, True)
+        g = E.dropout_bwd(g, m1, p)
+        g = E.nhwc_unflatten(g, pl.shape[1], pl.shape[2], pl.shape[3])
+        g = E.maxpool2x2_bwd(g, idx, list(a2.shape))
+        g = E.relu_bwd(a2, g)
+        g = E.conv2d_bwd_into(a1, self.conv2.weight, g, 1, 0, True,
+                              self.conv2.weight.grad, self.conv2.bias.grad)
+        g = E.relu_bwd(a1, g)
+        E.conv2d_bwd_into(x, self.conv1.weight, g, 1, 0, False,
+                          self.conv1.weight.grad, self.conv1.bias.grad)
+        return loss
+
 
 class CNN_CIFAR(_OpsModel):
     """32x32x3: 3 x [conv3x3 (3->64->128->256)+relu+maxpool2], flatten 1024,
@@ -85,3 +122,51 @@ class CNN_CIFAR(_OpsModel):
         x = Fo.dropout(x, self.p_drop, self.training, self.rng)
         x = Fo.linear(x, self.fc3.weight, self.fc3.bias)
         return x
+
+    def manual_step(self, x, labels, dloss):
+        """Hand-rolled fwd+bwd — see CNN_MNIST.manual_step."""
+        from ..ops import ext
+        E = ext()
+        p = self.p_drop
+        st = self.rng.gpu_state(x.device)
+        a1 = E.conv2d_fwd(x, self.conv1.weight, self.conv1.bias, 1, 0, True)
+        p1, i1 = E.maxpool2x2_fwd(a1)
+        a2 = E.conv2d_fwd(p1, self.conv2.weight, self.conv2.bias, 1, 0, True)
+        p2, i2 = E.maxpool2x2_fwd(a2)
+        a3 = E.conv2d_fwd(p2, self.conv3.weight, self.conv3.bias, 1, 0, True)
+        p3, i3 = E.maxpool2x2_fwd(a3)
+        fl = E.nhwc_flatten(p3)
+        d1, m1 = E.dropout_fwd_dev(fl, p, st, 0)
+        h1 = E.linear_fwd(d1, self.fc1.weight, self.fc1.bias, True)
+        d2, m2 = E.dropout_fwd_dev(h1, p, st, 1)
+        h2 = E.linear_fwd(d2, self.fc2.weight, self.fc2.bias, True)
+        d3, m3 = E.dropout_fwd_dev(h2, p, st, 2)
+        out = E.linear_fwd(d3, self.fc3.weight, self.fc3.bias, False)
+        loss, softmax = E.cross_entropy_fwd(out, labels)
+
+        g = E.cross_entropy_bwd(softmax, labels, dloss)
+        g = E.linear_bwd_into(d3, self.fc3.weight, g,
+                              self.fc3.weight.grad, self.fc3.bias.grad, True)
+        g = E.dropout_bwd(g, m3, p)
+        g = E.relu_bwd(h2, g)
+        g = E.linear_bwd_into(d2, self.fc2.weight, g,
+                              self.fc2.weight.grad, self.fc2.bias.grad, True)
+        g = E.dropout_bwd(g, m2, p)
+        g = E.relu_bwd(h1, g)
+        g = E.linear_bwd_into(d1, self.fc1.weight, g,
+                              self.fc1.weight.grad, self.fc1.bias.grad, True)
+        g = E.dropout_bwd(g, m1, p)
+        g = E.nhwc_unflatten(g, p3.shape[1], p3.shape[2], p3.shape[3])
+        g = E.maxpool2x2_bwd(g, i3, list(a3.shape))
+        g = E.relu_bwd(a3, g)
+        g = E.conv2d_bwd_into(p2, self.conv3.weight, g, 1, 0, True,
+                              self.conv3.weight.grad, self.conv3.bias.grad)
+        g = E.maxpool2x2_bwd(g, i2, list(a2.shape))
+        g = E.relu_bwd(a2, g)
+        g = E.conv2d_bwd_into(p1, self.conv2.weight, g, 1, 0, True,
+                              self.conv2.weight.grad, self.conv2.bias.grad)
+        g = E.maxpool2x2_bwd(g, i1, list(a1.shape))
+        g = E.relu_bwd(a1, g)
+        E.conv2d_bwd_into(x, self.conv1.weight, g, 1, 0, False,
+                          self.conv1.weight.grad, self.conv1.bias.grad)
+        return loss

@@ -690,6 +690,42 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> linear_bwd(
   return {dx, dw, db};
 }
 
+// manual-tape variant: dw/db written into caller-owned views (see
+// conv2d_bwd_into below for the rationale); fp32 only.
+torch::Tensor linear_bwd_into(torch::Tensor x, torch::Tensor w,
+                              torch::Tensor dy, torch::Tensor dw_out,
+                              c10::optional<torch::Tensor> db_out,
+                              bool need_dx) {
+  TORCH_CHECK(x.is_cuda() && !is_bf16(x));
+  dy = dy.contiguous();
+  x = x.contiguous();
+  w = w.contiguous();
+  int bM = x.size(0), in = x.size(1), out = w.size(0);
+  TORCH_CHECK(dw_out.is_contiguous() &&
+              dw_out.numel() == (long)out * in);
+  torch::Tensor dx;
+  if (need_dx)
+    dx = gemm_layout(dy, w, c10::nullopt, false, bM, in, out, out, in, 0);
+  {  // dw = dY^T @ X straight into the view
+    int SK = gemm_f32_splitk(out, in, bM);
+    torch::Tensor ws;
+    float* wsp = nullptr;
+    if (SK > 1) {
+      ws = torch::empty({(long)SK * out * in}, x.options());
+      wsp = ws.data_ptr<float>();
+    }
+    launch_gemm_f32(dy.data_ptr<float>(), x.data_ptr<float>(),
+                    dw_out.data_ptr<float>(), nullptr, wsp, out, in, bM,
+                    out, in, in, SK, 0, 1, stream_of(dy));
+  }
+  if (db_out) {
+    TORCH_CHECK(db_out->is_contiguous() && db_out->numel() == out);
+    launch_colsum(dy.data_ptr<float>(), db_out->data_ptr<float>(), bM, out,
+                  stream_of(dy));
+  }
+  return dx;
+}
+
 // ------------------------------------------------------------------ conv
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
@@ -833,6 +869,58 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
     db = torch::Tensor();
   }
   return {dx, dw, db};
+}
+
+// ---- manual-tape backward: grads land DIRECTLY in caller-owned views ----
+// Used by engine.ManualTape (the hand-rolled training step): with the
+// weight/bias gradients written straight into their flat_grads slices,
+// autograd's per-param accumulate-add and the zero-grad fill disappear.
+// Assignment into the view is bitwise-identical to autograd's
+// accumulate-into-zeroed-grad.  fp32 only (the bf16 ResNet path keeps
+// autograd this round).
+
+torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
+                              torch::Tensor dy, int64_t stride, int64_t pad,
+                              bool need_dx, torch::Tensor dw_out,
+                              c10::optional<torch::Tensor> db_out) {
+  TORCH_CHECK(x.is_cuda() && !is_bf16(x));
+  TORCH_CHECK(dw_out.is_contiguous() && dw_out.numel() == w.numel());
+  x = cl(x, "conv_bwd.x");
+  w = w.contiguous();
+  dy = cl(dy, "conv_bwd.dy");
+  int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kout = w.size(0), R = w.size(2), S = w.size(3);
+  int OH = dy.size(2), OW = dy.size(3);
+  auto st = stream_of(x);
+
+  torch::Tensor dx;
+  if (need_dx) {
+    auto wp = torch::empty({(long)Kout * R * S, C}, w.options());
+    launch_wperm_kors_c(w.data_ptr<float>(), wp.data_ptr<float>(), Kout, C,
+                        R * S, st);
+    dx = empty_cl({Nb, C, H, W}, x.options());
+    launch_conv_bwd_data(dy.data_ptr<float>(), wp.data_ptr<float>(),
+                         dx.data_ptr<float>(), Nb, C, H, W, Kout, R, S, OH,
+                         OW, (int)stride, (int)pad, st);
+  }
+
+  int Ncrs = C * R * S;
+  long Kdim = (long)Nb * OH * OW;
+  int SK = conv_bwd_weight_splitk(Kout, Ncrs, Kdim);
+  long ws_mult = (Ncrs <= 32 && Kout <= 64) ? 2049 : (long)SK + 1;
+  auto ws = torch::empty({ws_mult * Kout * Ncrs}, w.options());
+  launch_conv_bwd_weight(dy.data_ptr<float>(), x.data_ptr<float>(),
+                         dw_out.data_ptr<float>(), ws.data_ptr<float>(), SK,
+                         Nb, C, H, W, Kout, R, S, OH, OW, (int)stride,
+                         (int)pad, st);
+
+  if (db_out) {
+    TORCH_CHECK(db_out->is_contiguous() && db_out->numel() == Kout);
+    auto parts = torch::empty({(long)Kout * 4096}, dy.options());
+    launch_conv_db(dy.data_ptr<float>(), db_out->data_ptr<float>(),
+                   parts.data_ptr<float>(), Nb, Kout, OH * OW, st);
+  }
+  return dx;
 }
 
 // ------------------------------------------------------------- batchnorm
@@ -992,6 +1080,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_bwd", &linear_bwd);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_bwd", &conv2d_bwd);
+  m.def("conv2d_bwd_into", &conv2d_bwd_into);
+  m.def("linear_bwd_into", &linear_bwd_into);
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("poison_set_u8", &poison_set_u8);
