@@ -55,7 +55,7 @@ void conv_fwd_bf16_k(const unsigned short* __restrict__ x,
   const int bko = (t & 7) * 8, bkr = t >> 3;
 
   int ow0[2], oh0[2];
-  long base[2];
+  long base[2];  // patch base incl. tap 0 + ak: gather = base[j] + tap_off
   bool mval[2];
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
@@ -64,39 +64,55 @@ void conv_fwd_bf16_k(const unsigned short* __restrict__ x,
     long gmc = mval[j] ? gm : 0;
     ow0[j] = (int)(gmc % sh.OW) * sh.stride - sh.pad;
     oh0[j] = (int)((gmc / sh.OW) % sh.OH) * sh.stride - sh.pad;
-    base[j] = (gmc / ((long)sh.OW * sh.OH)) * (long)sh.H * sh.W * sh.C;
+    base[j] = (gmc / ((long)sh.OW * sh.OH)) * (long)sh.H * sh.W * sh.C +
+              ((long)oh0[j] * sh.W + ow0[j]) * sh.C + ak;
   }
+
+  // staged-tap state carried incrementally across stage_load calls
+  // (ascending k0; C % 32 == 0 on this path) — no div/mod in the loop
+  int tr = 0, ts = 0, tcb = 0;
+  long tap_off = 0;
+  const bool wt_v8 = (sh.Kout % 8) == 0 && n_blk + bko + 7 < sh.Kout;
+  long wt_base = (long)bkr * sh.Kout + n_blk + bko;
 
   unsigned short ra[2][8], rb[8];
   auto stage_load = [&](int k0) {
-    int rs = k0 / sh.C;
-    int r = rs / sh.S, s = rs % sh.S;
-    int c0 = k0 - rs * sh.C + ak;
+    const int r = tr, s = ts;
+    const long toff = tap_off;
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      int ih = oh0[j] + r, iw = ow0[j] + s;
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (mval[j] && (P0 || (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)))
-        q = *(const bf16x8*)(x + base[j] + ((long)ih * sh.W + iw) * sh.C +
-                             c0);
+      bool ok = mval[j];
+      if (!P0) {
+        int ih = oh0[j] + r, iw = ow0[j] + s;
+        ok = ok && (unsigned)ih < (unsigned)sh.H &&
+             (unsigned)iw < (unsigned)sh.W;
+      }
+      if (ok) q = *(const bf16x8*)(x + base[j] + toff);
       *(bf16x8*)ra[j] = q;
+    }
+    tcb += BKB; tap_off += BKB;
+    if (tcb == sh.C) {
+      tcb = 0;
+      if (++ts == sh.S) { ts = 0; ++tr;
+                          tap_off += (long)(sh.W - sh.S) * sh.C; }
     }
     {
       int gk = k0 + bkr;
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (gk < Kdim && n_blk + bko + 7 < sh.Kout + 7) {
-        if (n_blk + bko + 7 < sh.Kout && (sh.Kout % 8) == 0)
-          q = *(const bf16x8*)(wt + (long)gk * sh.Kout + n_blk + bko);
+      if (gk < Kdim) {
+        if (wt_v8)
+          q = *(const bf16x8*)(wt + wt_base);
         else {
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             if (n_blk + bko + e < sh.Kout)
-              ((unsigned short*)&q)[e] =
-                  wt[(long)gk * sh.Kout + n_blk + bko + e];
+              ((unsigned short*)&q)[e] = wt[wt_base + e];
         }
       }
       *(bf16x8*)rb = q;
     }
+    wt_base += (long)BKB * sh.Kout;
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
@@ -185,7 +201,7 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
   const int stride = ST > 0 ? ST : sh.stride;
 
   int iwp[2], ihp[2];
-  long base[2];
+  long base[2];  // ST==1: folded patch base (gather = base[j] - tile_off)
   bool mval[2];
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
@@ -195,42 +211,68 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
     iwp[j] = (int)(gmc % sh.W) + sh.pad;
     ihp[j] = (int)((gmc / sh.W) % sh.H) + sh.pad;
     base[j] = (gmc / ((long)sh.W * sh.H)) * (long)sh.OH * sh.OW * sh.Kout;
+    if (ST == 1)
+      base[j] += ((long)ihp[j] * sh.OW + iwp[j]) * sh.Kout + ak;
   }
+
+  // staged-tap state (ascending k0; KO % 32 == 0 on this path)
+  int tr = 0, ts = 0, tkb = 0;
+  long noff = 0;
+  const bool wp_v8 = (sh.C % 8) == 0 && n_blk + bc8 + 7 < sh.C;
+  long wp_base = (long)bkr * sh.C + n_blk + bc8;
 
   unsigned short ra[2][8], rb[8];
   auto stage_load = [&](int k0) {
-    int rs = k0 / sh.Kout;
-    int r = rs / sh.S, s = rs % sh.S;
-    int ko0 = k0 - rs * sh.Kout + ak;
+    const int r = tr, s = ts;
+    if (ST == 1) {
+      const long toff = noff;
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      int ohn = ihp[j] - r, own = iwp[j] - s;
-      bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
-          own % stride == 0) {
-        int oh = ohn / stride, ow = own / stride;
-        if (oh < sh.OH && ow < sh.OW)
-          q = *(const bf16x8*)(dy + base[j] +
-                               ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+      for (int j = 0; j < 2; ++j) {
+        int ohn = ihp[j] - r, own = iwp[j] - s;
+        bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (mval[j] && (unsigned)ohn < (unsigned)sh.OH &&
+            (unsigned)own < (unsigned)sh.OW)
+          q = *(const bf16x8*)(dy + base[j] - toff);
+        *(bf16x8*)ra[j] = q;
       }
-      *(bf16x8*)ra[j] = q;
+    } else {
+      const int ko0 = tkb + ak;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int ohn = ihp[j] - r, own = iwp[j] - s;
+        bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (mval[j] && ohn >= 0 && own >= 0 && ohn % stride == 0 &&
+            own % stride == 0) {
+          int oh = ohn / stride, ow = own / stride;
+          if (oh < sh.OH && ow < sh.OW)
+            q = *(const bf16x8*)(dy + base[j] +
+                                 ((long)oh * sh.OW + ow) * sh.Kout + ko0);
+        }
+        *(bf16x8*)ra[j] = q;
+      }
+    }
+    tkb += BKB; noff -= BKB;
+    if (tkb == sh.Kout) {
+      tkb = 0; noff += 2L * sh.Kout;
+      if (++ts == sh.S) { ts = 0; ++tr;
+                          noff += (long)(sh.OW - sh.S) * sh.Kout; }
     }
     {
       int gk = k0 + bkr;
       bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
       if (gk < Kdim) {
-        if (n_blk + bc8 + 7 < sh.C && (sh.C % 8) == 0)
-          q = *(const bf16x8*)(wp + (long)gk * sh.C + n_blk + bc8);
+        if (wp_v8)
+          q = *(const bf16x8*)(wp + wp_base);
         else {
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             if (n_blk + bc8 + e < sh.C)
-              ((unsigned short*)&q)[e] = wp[(long)gk * sh.C + n_blk + bc8 +
-                                            e];
+              ((unsigned short*)&q)[e] = wp[wp_base + e];
         }
       }
       *(bf16x8*)rb = q;
     }
+    wp_base += (long)BKB * sh.C;
   };
   auto stage_write = [&](int buf) {
 #pragma unroll
