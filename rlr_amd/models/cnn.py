@@ -75,22 +75,28 @@ class CNN_MNIST(_OpsModel):
         out = E.linear_fwd(d2, self.fc2.weight, self.fc2.bias, False)
         loss, softmax = E.cross_entropy_fwd(out, labels)
 
+        # Backward with the relu masks FUSED away (each bitwise-identical
+        # to the separate relu_bwd: see maxpool2x2_bwd_gather4_k /
+        # dropout_relu_bwd_k / conv_bwd_data relu_y notes in ops/csrc):
+        #   * fc1's relu folds into the dropout-2 backward;
+        #   * conv2's relu folds into the maxpool gather via the pooled
+        #     values (window max == pooled value for post-relu inputs);
+        #   * conv1's relu folds into conv2's bwd-data epilogue.
         g = E.cross_entropy_bwd(softmax, labels, dloss)
         g = E.linear_bwd_into(d2, self.fc2.weight, g,
                               self.fc2.weight.grad, self.fc2.bias.grad, True)
-        g = E.dropout_bwd(g, m2, p)
-        g = E.relu_bwd(h1, g)
+        g = E.dropout_relu_bwd(g, m2, h1, p)
         g = E.linear_bwd_into(d1, self.fc1.weight, g,
                               self.fc1.weight.grad, self.fc1.bias.grad, True)
         g = E.dropout_bwd(g, m1, p)
         g = E.nhwc_unflatten(g, pl.shape[1], pl.shape[2], pl.shape[3])
-        g = E.maxpool2x2_bwd(g, idx, list(a2.shape))
-        g = E.relu_bwd(a2, g)
+        g = E.maxpool2x2_bwd_relu(g, idx, pl, list(a2.shape))
         g = E.conv2d_bwd_into(a1, self.conv2.weight, g, 1, 0, True,
-                              self.conv2.weight.grad, self.conv2.bias.grad)
-        g = E.relu_bwd(a1, g)
+                              self.conv2.weight.grad, self.conv2.bias.grad,
+                              a1)
         E.conv2d_bwd_into(x, self.conv1.weight, g, 1, 0, False,
-                          self.conv1.weight.grad, self.conv1.bias.grad)
+                          self.conv1.weight.grad, self.conv1.bias.grad,
+                          None)
         return loss
 
 
@@ -144,29 +150,30 @@ class CNN_CIFAR(_OpsModel):
         out = E.linear_fwd(d3, self.fc3.weight, self.fc3.bias, False)
         loss, softmax = E.cross_entropy_fwd(out, labels)
 
+        # relu masks fused (see CNN_MNIST.manual_step): fc relus fold
+        # into the dropout backwards; every conv relu folds into the
+        # following maxpool gather via the pooled values.
         g = E.cross_entropy_bwd(softmax, labels, dloss)
         g = E.linear_bwd_into(d3, self.fc3.weight, g,
                               self.fc3.weight.grad, self.fc3.bias.grad, True)
-        g = E.dropout_bwd(g, m3, p)
-        g = E.relu_bwd(h2, g)
+        g = E.dropout_relu_bwd(g, m3, h2, p)
         g = E.linear_bwd_into(d2, self.fc2.weight, g,
                               self.fc2.weight.grad, self.fc2.bias.grad, True)
-        g = E.dropout_bwd(g, m2, p)
-        g = E.relu_bwd(h1, g)
+        g = E.dropout_relu_bwd(g, m2, h1, p)
         g = E.linear_bwd_into(d1, self.fc1.weight, g,
                               self.fc1.weight.grad, self.fc1.bias.grad, True)
         g = E.dropout_bwd(g, m1, p)
         g = E.nhwc_unflatten(g, p3.shape[1], p3.shape[2], p3.shape[3])
-        g = E.maxpool2x2_bwd(g, i3, list(a3.shape))
-        g = E.relu_bwd(a3, g)
+        g = E.maxpool2x2_bwd_relu(g, i3, p3, list(a3.shape))
         g = E.conv2d_bwd_into(p2, self.conv3.weight, g, 1, 0, True,
-                              self.conv3.weight.grad, self.conv3.bias.grad)
-        g = E.maxpool2x2_bwd(g, i2, list(a2.shape))
-        g = E.relu_bwd(a2, g)
+                              self.conv3.weight.grad, self.conv3.bias.grad,
+                              None)
+        g = E.maxpool2x2_bwd_relu(g, i2, p2, list(a2.shape))
         g = E.conv2d_bwd_into(p1, self.conv2.weight, g, 1, 0, True,
-                              self.conv2.weight.grad, self.conv2.bias.grad)
-        g = E.maxpool2x2_bwd(g, i1, list(a1.shape))
-        g = E.relu_bwd(a1, g)
+                              self.conv2.weight.grad, self.conv2.bias.grad,
+                              None)
+        g = E.maxpool2x2_bwd_relu(g, i1, p1, list(a1.shape))
         E.conv2d_bwd_into(x, self.conv1.weight, g, 1, 0, False,
-                          self.conv1.weight.grad, self.conv1.bias.grad)
+                          self.conv1.weight.grad, self.conv1.bias.grad,
+                          None)
         return loss

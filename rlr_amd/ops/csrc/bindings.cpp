@@ -66,6 +66,14 @@ void launch_dropout_fwd_dev(const float*, float*, uint8_t*, long, float,
                             const unsigned long long*, int, void*);
 void launch_dropout_bwd(const float*, const uint8_t*, float*, long, float,
                         void*);
+void launch_dropout_relu_bwd(const float*, const uint8_t*, const float*,
+                             float*, long, float, void*);
+void launch_maxpool2x2_bwd_relu(const float*, const uint8_t*, const float*,
+                                float*, long, int, int, int, int, int,
+                                void*);
+void launch_conv_bwd_data_relu(const float*, const float*, float*,
+                               const float*, int, int, int, int, int, int,
+                               int, int, int, int, int, void*);
 void launch_gap_fwd(const float*, float*, long, int, int, void*);
 void launch_gap_bwd(const float*, float*, long, int, int, void*);
 void launch_ce_fwd(const float*, const long*, float*, float*, float*, int,
@@ -289,6 +297,24 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
   return dx;
 }
 
+// fused relu-mask maxpool backward (manual tape; C % 4 == 0)
+torch::Tensor maxpool2x2_bwd_relu(torch::Tensor dy, torch::Tensor idx,
+                                  torch::Tensor relu_pooled,
+                                  std::vector<int64_t> in_shape) {
+  TORCH_CHECK(dy.is_cuda() && !is_bf16(dy));
+  dy = cl(dy, "mp_bwd.dy");
+  relu_pooled = cl(relu_pooled, "mp_bwd.relu");
+  int Nb = in_shape[0], C = in_shape[1], H = in_shape[2], W = in_shape[3];
+  TORCH_CHECK((C % 4) == 0);
+  int OH = dy.size(2), OW = dy.size(3);
+  auto dx = empty_cl({Nb, C, H, W}, dy.options());
+  launch_maxpool2x2_bwd_relu(dy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
+                             relu_pooled.data_ptr<float>(),
+                             dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
+                             stream_of(dy));
+  return dx;
+}
+
 std::tuple<torch::Tensor, torch::Tensor> dropout_fwd(torch::Tensor x,
                                                      double p, int64_t seed,
                                                      int64_t offset) {
@@ -337,6 +363,19 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
     launch_dropout_bwd(dy.data_ptr<float>(), mask.data_ptr<uint8_t>(),
                        dx.data_ptr<float>(), dy.numel(), (float)p,
                        stream_of(dy));
+  return dx;
+}
+
+// fused dropout+relu backward (manual tape): (y>0) * dropout_bwd(dy)
+torch::Tensor dropout_relu_bwd(torch::Tensor dy, torch::Tensor mask,
+                               torch::Tensor y, double p) {
+  CHK_CUDA(dy);
+  TORCH_CHECK(!is_bf16(dy));
+  dy = dy.contiguous();
+  auto dx = torch::empty_like(dy);
+  launch_dropout_relu_bwd(dy.data_ptr<float>(), mask.data_ptr<uint8_t>(),
+                          y.data_ptr<float>(), dx.data_ptr<float>(),
+                          dy.numel(), (float)p, stream_of(dy));
   return dx;
 }
 
@@ -882,7 +921,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
 torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
                               torch::Tensor dy, int64_t stride, int64_t pad,
                               bool need_dx, torch::Tensor dw_out,
-                              c10::optional<torch::Tensor> db_out) {
+                              c10::optional<torch::Tensor> db_out,
+                              c10::optional<torch::Tensor> relu_y) {
   TORCH_CHECK(x.is_cuda() && !is_bf16(x));
   TORCH_CHECK(dw_out.is_contiguous() && dw_out.numel() == w.numel());
   x = cl(x, "conv_bwd.x");
@@ -899,9 +939,15 @@ torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
     launch_wperm_kors_c(w.data_ptr<float>(), wp.data_ptr<float>(), Kout, C,
                         R * S, st);
     dx = empty_cl({Nb, C, H, W}, x.options());
-    launch_conv_bwd_data(dy.data_ptr<float>(), wp.data_ptr<float>(),
-                         dx.data_ptr<float>(), Nb, C, H, W, Kout, R, S, OH,
-                         OW, (int)stride, (int)pad, st);
+    const float* ry = nullptr;
+    torch::Tensor ryt;
+    if (relu_y) {
+      ryt = cl(*relu_y, "conv_bwd.relu_y");
+      ry = ryt.data_ptr<float>();
+    }
+    launch_conv_bwd_data_relu(dy.data_ptr<float>(), wp.data_ptr<float>(),
+                              dx.data_ptr<float>(), ry, Nb, C, H, W, Kout,
+                              R, S, OH, OW, (int)stride, (int)pad, st);
   }
 
   int Ncrs = C * R * S;
@@ -1082,6 +1128,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd", &conv2d_bwd);
   m.def("conv2d_bwd_into", &conv2d_bwd_into);
   m.def("linear_bwd_into", &linear_bwd_into);
+  m.def("dropout_relu_bwd", &dropout_relu_bwd);
+  m.def("maxpool2x2_bwd_relu", &maxpool2x2_bwd_relu);
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("poison_set_u8", &poison_set_u8);

@@ -393,7 +393,8 @@ template <int ST, bool V4>
 __global__ __launch_bounds__(256)
 void conv_bwd_data_k(const float* __restrict__ dy,
                      const float* __restrict__ wp,  // [(r,s,ko)][C]
-                     float* __restrict__ dx, ConvShape sh, int Kdim) {
+                     float* __restrict__ dx, ConvShape sh, int Kdim,
+                     const float* __restrict__ relu_y) {
   constexpr int BM = 128, MI = 4, NI = 2;
   __shared__ float A_lds[2][BM * LDA_S];
   __shared__ float B_lds[2][BK * LDB_S];
@@ -558,7 +559,9 @@ void conv_bwd_data_k(const float* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         long m = m_blk + wr * 64 + mi * 16 + l4 * 4 + r;
         if (m >= M) continue;
-        dx[m * sh.C + c] = acc[mi][ni][r];
+        float v = acc[mi][ni][r];
+        if (relu_y && relu_y[m * sh.C + c] <= 0.f) v = 0.f;
+        dx[m * sh.C + c] = v;
       }
     }
 }
@@ -571,7 +574,8 @@ template <int ST, bool V4>
 __global__ __launch_bounds__(256)
 void conv_bwd_data32_k(const float* __restrict__ dy,
                        const float* __restrict__ wp,  // [(r,s,ko)][C]
-                       float* __restrict__ dx, ConvShape sh, int Kdim) {
+                       float* __restrict__ dx, ConvShape sh, int Kdim,
+                       const float* __restrict__ relu_y) {
   constexpr int BM = 128, MI = 2, NI = 2, BN32 = 32;
   __shared__ float A_lds[2][BM * LDA_S];
   __shared__ float B_lds[2][BK * LDB32_S];
@@ -726,7 +730,9 @@ void conv_bwd_data32_k(const float* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         long m = m_blk + wr * 32 + mi * 16 + l4 * 4 + r;
         if (m >= M) continue;
-        dx[m * sh.C + c] = acc[mi][ni][r];
+        float v = acc[mi][ni][r];
+        if (relu_y && relu_y[m * sh.C + c] <= 0.f) v = 0.f;
+        dx[m * sh.C + c] = v;
       }
     }
 }
@@ -1202,10 +1208,11 @@ void launch_conv_fwd(const float* x, const float* wt, const float* bias,
                                                    relu);
 }
 
-void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
-                          int Nb, int C, int H, int W, int Kout, int R,
-                          int S, int OH, int OW, int stride, int pad,
-                          void* s) {
+void launch_conv_bwd_data_relu(const float* dy, const float* wp,
+                               float* dx, const float* relu_y, int Nb,
+                               int C, int H, int W, int Kout, int R, int S,
+                               int OH, int OW, int stride, int pad,
+                               void* s) {
   ConvShape sh{Nb, C, H, W, Kout, R, S, OH, OW, stride, pad};
   int Kdim = Kout * R * S;
   long M = (long)Nb * H * W;
@@ -1215,27 +1222,35 @@ void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
   if (C <= 32) {
     dim3 g32((M + 127) / 128, (C + 31) / 32, 1);
     if (stride == 1 && v4)
-      conv_bwd_data32_k<1, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+      conv_bwd_data32_k<1, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
     else if (stride == 2 && v4)
-      conv_bwd_data32_k<2, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+      conv_bwd_data32_k<2, true><<<g32, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
     else if (stride == 1)
       conv_bwd_data32_k<1, false><<<g32, 256, 0, st>>>(dy, wp, dx, sh,
-                                                       Kdim);
+                                                       Kdim, relu_y);
     else
       conv_bwd_data32_k<0, false><<<g32, 256, 0, st>>>(dy, wp, dx, sh,
-                                                       Kdim);
+                                                       Kdim, relu_y);
     return;
   }
   if (stride == 1 && v4)
-    conv_bwd_data_k<1, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<1, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
   else if (stride == 2 && v4)
-    conv_bwd_data_k<2, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<2, true><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
   else if (stride == 1)
-    conv_bwd_data_k<1, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<1, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
   else if (stride == 2)
-    conv_bwd_data_k<2, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<2, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
   else
-    conv_bwd_data_k<0, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_k<0, false><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim, relu_y);
+}
+
+void launch_conv_bwd_data(const float* dy, const float* wp, float* dx,
+                          int Nb, int C, int H, int W, int Kout, int R,
+                          int S, int OH, int OW, int stride, int pad,
+                          void* s) {
+  launch_conv_bwd_data_relu(dy, wp, dx, nullptr, Nb, C, H, W, Kout, R, S,
+                            OH, OW, stride, pad, s);
 }
 
 int conv_bwd_weight_splitk(int Kout, int Ncrs, long Kdim) {

@@ -137,6 +137,54 @@ __global__ void maxpool2x2_bwd_gather_k(const T* __restrict__ dy,
   }
 }
 
+// float4 gather (C % 4 == 0 — every pooled activation here): 4 channels
+// per thread share one (ih,iw)->(oh,ow) decomposition; idx is a uchar4,
+// dy/dx are float4 — a quarter of the address math and store issues of
+// the scalar form.  `relu_pooled` (optional) fuses the upstream relu
+// mask: for post-relu inputs the window max equals the pooled value, so
+// masking dy by pooled>0 is bitwise-identical to a separate relu_bwd on
+// the pre-pool tensor (manual tape, models/cnn.py).
+__global__ void maxpool2x2_bwd_gather4_k(const float* __restrict__ dy,
+                                         const uint8_t* __restrict__ idx,
+                                         const float* __restrict__ relu_pooled,
+                                         float* __restrict__ dx, long B,
+                                         int H, int W, int OH, int OW,
+                                         int C) {
+  long n4 = B * H * (long)W * C / 4;
+  long stride = (long)gridDim.x * blockDim.x;
+  int c4 = C / 4;
+  int csh = 31 - __clz(c4);
+  bool cp2 = (c4 & (c4 - 1)) == 0;
+  for (long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
+       i4 += stride) {
+    int c = (cp2 ? (int)(i4 & (c4 - 1)) : (int)(i4 % c4)) * 4;
+    long rest = cp2 ? (i4 >> csh) : (i4 / c4);
+    int iw = rest % W;
+    int ih = (rest / W) % H;
+    long b = rest / ((long)W * H);
+    int oh = ih >> 1, ow = iw >> 1;
+    float4 g = {0.f, 0.f, 0.f, 0.f};
+    if (oh < OH && ow < OW) {
+      long o = ((b * OH + oh) * (long)OW + ow) * C + c;
+      const uint8_t* ap = idx + o;
+      float4 d4 = *(const float4*)(dy + o);
+      uint8_t a = ((ih & 1) << 1) | (iw & 1);
+      g.x = ap[0] == a ? d4.x : 0.f;
+      g.y = ap[1] == a ? d4.y : 0.f;
+      g.z = ap[2] == a ? d4.z : 0.f;
+      g.w = ap[3] == a ? d4.w : 0.f;
+      if (relu_pooled) {
+        float4 p4 = *(const float4*)(relu_pooled + o);
+        g.x = p4.x > 0.f ? g.x : 0.f;
+        g.y = p4.y > 0.f ? g.y : 0.f;
+        g.z = p4.z > 0.f ? g.z : 0.f;
+        g.w = p4.w > 0.f ? g.w : 0.f;
+      }
+    }
+    *(float4*)(dx + i4 * 4) = g;
+  }
+}
+
 // Strip form: one thread per (b, oh, c) output ROW STRIP.  Walking ow in
 // a loop removes the per-element div/mod of the earlier gather form (the
 // only VALU divisions on this path; they held the kernel to ~1.5 TB/s
@@ -206,6 +254,12 @@ void launch_maxpool2x2_fwd(const float* x, float* y, uint8_t* idx, long B,
 void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
                            long B, int H, int W, int OH, int OW, int C,
                            void* s) {
+  if ((C % 4) == 0) {
+    long n4 = B * H * (long)W * C / 4;
+    maxpool2x2_bwd_gather4_k<<<grid_for(n4), kBlock, 0, (hipStream_t)s>>>(
+        dy, idx, nullptr, dx, B, H, W, OH, OW, C);
+    return;
+  }
   if (mp_strip()) {
     long n = B * OH * (long)C;  // one thread per (b, oh, c) strip
     maxpool2x2_bwd_k<float><<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
@@ -216,6 +270,16 @@ void launch_maxpool2x2_bwd(const float* dy, const uint8_t* idx, float* dx,
         <<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, idx, dx, B, H, W,
                                                      OH, OW, C);
   }
+}
+
+// fused relu-mask variant (C % 4 == 0 required; manual tape only)
+void launch_maxpool2x2_bwd_relu(const float* dy, const uint8_t* idx,
+                                const float* relu_pooled, float* dx, long B,
+                                int H, int W, int OH, int OW, int C,
+                                void* s) {
+  long n4 = B * H * (long)W * C / 4;
+  maxpool2x2_bwd_gather4_k<<<grid_for(n4), kBlock, 0, (hipStream_t)s>>>(
+      dy, idx, relu_pooled, dx, B, H, W, OH, OW, C);
 }
 void launch_maxpool2x2_fwd_bf16(const unsigned short* x, unsigned short* y,
                                 uint8_t* idx, long B, int H, int W, int OH,
@@ -352,6 +416,19 @@ __global__ void dropout_bwd_k(const float* __restrict__ dy,
     dx[i] = mask[i] ? dy[i] * scale : 0.f;
 }
 
+// fused dropout+relu backward (manual tape): out = (y>0) * dropout_bwd —
+// bitwise-identical to dropout_bwd_k followed by relu_bwd_k
+__global__ void dropout_relu_bwd_k(const float* __restrict__ dy,
+                                   const uint8_t* __restrict__ mask,
+                                   const float* __restrict__ y,
+                                   float* __restrict__ dx, long n, float p) {
+  float scale = 1.0f / (1.0f - p);
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dx[i] = (y[i] > 0.f && mask[i]) ? dy[i] * scale : 0.f;
+}
+
 extern "C" {
 void launch_dropout_fwd(const float* x, float* y, uint8_t* mask, long n,
                         float p, uint64_t seed, uint64_t offset, void* s) {
@@ -368,6 +445,12 @@ void launch_dropout_bwd(const float* dy, const uint8_t* mask, float* dx,
                         long n, float p, void* s) {
   dropout_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(dy, mask, dx, n,
                                                             p);
+}
+void launch_dropout_relu_bwd(const float* dy, const uint8_t* mask,
+                             const float* y, float* dx, long n, float p,
+                             void* s) {
+  dropout_relu_bwd_k<<<grid_for(n), kBlock, 0, (hipStream_t)s>>>(
+      dy, mask, y, dx, n, p);
 }
 }
 
