@@ -995,56 +995,115 @@ void conv_small_fwd_k(const float* __restrict__ x,
 
 // dw partials for small Ncrs: block handles an m-chunk; per-thread (ko,crs)
 // pairs accumulate from LDS-staged dy/x tiles; fixed-order combine follows.
+// LDS is TRANSPOSED ([ko][m], [crs][m]) so the pairs loop reads float4
+// over m — 4x fewer LDS instructions than the [m][ko] form; dy staging is
+// a linear coalesced float4 copy (dy[m0*KO:] IS [m][ko] flat); x staging
+// exploits that one m's (r, s, c) taps are R contiguous rows of S*C
+// floats, with the m->(nb,oh,ow) decomposition carried incrementally.
 template <int MC>
 __global__ __launch_bounds__(256)
 void conv_small_bwdw_k(const float* __restrict__ dy,
                        const float* __restrict__ x,
                        float* __restrict__ partials,  // [chunk][KO*Ncrs]
                        ConvShape sh, int Ncrs, long k_per_chunk) {
-  __shared__ float dy_lds[MC * 64];   // [m][ko] (KO <= 64)
-  __shared__ float xp_lds[MC * 32];   // [m][crs] (Ncrs <= 32)
+  constexpr int LDR = MC + 4;          // 16-B aligned row stride
+  __shared__ float dy_lds[64][LDR];    // [ko][m] (KO <= 64)
+  __shared__ float xp_lds[32][LDR];    // [crs][m] (Ncrs <= 32)
   const long Kdim = (long)sh.Nb * sh.OH * sh.OW;
-  const long k_lo = (long)blockIdx.x * k_per_chunk;
+  const long k_lo = (long)blockIdx.x * k_per_chunk;  // multiple of MC
   const long k_hi = min(Kdim, k_lo + k_per_chunk);
   const int KO = sh.Kout;
   const int pairs = KO * Ncrs;
+  const int t = threadIdx.x;
+  const int SC = sh.S * sh.C;
+  const bool ko_p2 = (KO & (KO - 1)) == 0;
+  const int kosh = 31 - __clz(KO);
   float acc[8];  // up to 8 pairs per thread (KO*Ncrs <= 2048)
 #pragma unroll
   for (int i = 0; i < 8; ++i) acc[i] = 0.f;
 
+  // this thread's m-row state (m = m0 + t), advanced by MC per chunk
+  const bool mown = t < MC;
+  int ow_t = 0, oh_t = 0;
+  long nb_t = 0;
+  if (mown) {
+    long m = k_lo + t;
+    ow_t = (int)(m % sh.OW);
+    oh_t = (int)((m / sh.OW) % sh.OH);
+    nb_t = m / ((long)sh.OW * sh.OH);
+  }
+
   for (long m0 = k_lo; m0 < k_hi; m0 += MC) {
     int mc = (int)min((long)MC, k_hi - m0);
-    // stage dy[m][ko]
-    for (int i = threadIdx.x; i < mc * KO; i += blockDim.x)
-      dy_lds[i] = dy[(m0 + i / KO) * KO + (i % KO)];
-    // stage x patches [m][crs]
-    for (int i = threadIdx.x; i < mc * Ncrs; i += blockDim.x) {
-      int m = i / Ncrs, crs = i % Ncrs;
-      long gm = m0 + m;
-      int ow = gm % sh.OW;
-      int oh = (gm / sh.OW) % sh.OH;
-      long nb = gm / ((long)sh.OW * sh.OH);
-      int c = crs % sh.C;
-      int rs = crs / sh.C;
-      int r = rs / sh.S, s = rs % sh.S;
-      int ih = oh * sh.stride - sh.pad + r;
-      int iw = ow * sh.stride - sh.pad + s;
-      float v = 0.f;
-      if (ih >= 0 && ih < sh.H && iw >= 0 && iw < sh.W)
-        v = x[(nb * (long)sh.H * sh.W + (long)ih * sh.W + iw) * sh.C + c];
-      xp_lds[i] = v;
+    {  // dy: linear float4 read, transposed scalar LDS writes
+      const float* src = dy + m0 * KO;
+      int t4 = mc * KO / 4;  // KO % 4 == 0
+      for (int i = t; i < t4; i += blockDim.x) {
+        float4 v = *(const float4*)(src + (long)i * 4);
+        int e = i * 4;
+        int m = ko_p2 ? (e >> kosh) : (e / KO);
+        int ko = e - m * KO;
+        dy_lds[ko + 0][m] = v.x;
+        dy_lds[ko + 1][m] = v.y;
+        dy_lds[ko + 2][m] = v.z;
+        dy_lds[ko + 3][m] = v.w;
+      }
+      if (mc < MC)  // zero the tail rows so the full-MC pairs loop is exact
+        for (int i = t; i < (MC - mc) * KO; i += blockDim.x)
+          dy_lds[i % KO][mc + i / KO] = 0.f;
+    }
+    if (mown) {  // x patches for row m = m0 + t
+      if (m0 + t < k_hi) {
+        int ih0 = oh_t * sh.stride - sh.pad;
+        int iw0 = ow_t * sh.stride - sh.pad;
+        const float* xb = x + nb_t * (long)sh.H * sh.W * sh.C;
+        int crs = 0;
+        for (int r = 0; r < sh.R; ++r) {
+          int ih = ih0 + r;
+          const float* row = xb + ((long)ih * sh.W + iw0) * sh.C;
+          bool rok = (unsigned)ih < (unsigned)sh.H;
+          if (rok && sh.pad == 0) {
+            for (int j = 0; j < SC; ++j) xp_lds[crs + j][t] = row[j];
+          } else {
+            for (int s2 = 0; s2 < sh.S; ++s2) {
+              int iw = iw0 + s2;
+              bool ok = rok && (unsigned)iw < (unsigned)sh.W;
+              for (int c = 0; c < sh.C; ++c)
+                xp_lds[crs + s2 * sh.C + c][t] =
+                    ok ? row[s2 * sh.C + c] : 0.f;
+            }
+          }
+          crs += SC;
+        }
+      } else {
+        for (int j = 0; j < Ncrs; ++j) xp_lds[j][t] = 0.f;
+      }
+      ow_t += MC;
+      while (ow_t >= sh.OW) {
+        ow_t -= sh.OW;
+        if (++oh_t == sh.OH) { oh_t = 0; ++nb_t; }
+      }
     }
     __syncthreads();
-    for (int p = threadIdx.x, pi = 0; p < pairs; p += blockDim.x, ++pi) {
-      int ko = p / Ncrs, crs = p % Ncrs;
+    for (int pp = t, pi = 0; pp < pairs; pp += blockDim.x, ++pi) {
+      int ko = pp / Ncrs, crs = pp - (pp / Ncrs) * Ncrs;
+      const float4* dp = (const float4*)dy_lds[ko];
+      const float4* xp = (const float4*)xp_lds[crs];
       float a = acc[pi];
-      for (int m = 0; m < mc; ++m)
-        a = fmaf(dy_lds[m * KO + ko], xp_lds[m * Ncrs + crs], a);
+#pragma unroll 4
+      for (int m4 = 0; m4 < MC / 4; ++m4) {
+        float4 d = dp[m4];
+        float4 xv = xp[m4];
+        a = fmaf(d.x, xv.x, a);
+        a = fmaf(d.y, xv.y, a);
+        a = fmaf(d.z, xv.z, a);
+        a = fmaf(d.w, xv.w, a);
+      }
       acc[pi] = a;
     }
     __syncthreads();
   }
-  for (int p = threadIdx.x, pi = 0; p < pairs; p += blockDim.x, ++pi)
+  for (int p = t, pi = 0; p < pairs; p += blockDim.x, ++pi)
     partials[(long)blockIdx.x * pairs + p] = acc[pi];
 }
 
