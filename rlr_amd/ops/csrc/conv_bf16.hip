@@ -670,9 +670,10 @@ void launch_conv_bwd_weight_bf16_ex(const unsigned short* dy,
       conv_bwd_weight_bf16_k<false, 64, 2><<<grid, 256, 0, st>>>(
           dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
   }
-  if (SK > 1)
-    launch_splitk_reduce(slabs, rsc, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
-  launch_dwperm_rsc_crs(rsc, dw, Kout, C, R * S, s);
+  extern void launch_splitk_reduce_dwperm(const float*, float*, int, int,
+                                          int, int, void*);
+  launch_splitk_reduce_dwperm(SK == 1 ? rsc : slabs, dw, Kout, C, R * S, SK,
+                              s);
 }
 
 void launch_conv_bwd_weight_bf16(const unsigned short* dy,

@@ -1129,6 +1129,35 @@ __global__ void conv_small_bwdw_reduce_k(const float* __restrict__ partials,
   }
 }
 
+// fused split-K combine + dw permute: sums the SK rsc-ordered slabs and
+// writes straight into the torch (ko,(c,r,s)) layout — one kernel instead
+// of splitk_reduce + dwperm (the intermediate rsc tensor disappears)
+__global__ void splitk_reduce_dwperm_k(const float* __restrict__ ws,
+                                       float* __restrict__ dw, int Kout,
+                                       int C, int RS, int SK) {
+  long n_out = (long)Kout * C * RS;
+  long stride = (long)gridDim.x * blockDim.x;
+  int CRS = C * RS;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += stride) {
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int z = 0;
+    for (; z + 3 < SK; z += 4) {
+      a0 += ws[(long)z * n_out + i];
+      a1 += ws[(long)(z + 1) * n_out + i];
+      a2 += ws[(long)(z + 2) * n_out + i];
+      a3 += ws[(long)(z + 3) * n_out + i];
+    }
+    for (; z < SK; ++z) a0 += ws[(long)z * n_out + i];
+    float acc = (a0 + a1) + (a2 + a3);
+    int col = i % CRS;  // (r,s,c) flat: rs*C + c
+    long ko = i / CRS;
+    int c = col % C;
+    int rs = col / C;
+    dw[(ko * C + c) * RS + rs] = acc;
+  }
+}
+
 // dw column permute: [(ko)][(r,s,c)] -> torch layout [(ko)][(c,r,s)]
 __global__ void dwperm_rsc_crs_k(const float* __restrict__ in,
                                  float* __restrict__ out, int Kout, int C,
@@ -1367,10 +1396,10 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
   else
     conv_bwd_weight_k<false, false><<<grid, 256, 0, st>>>(
         dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
-  if (SK > 1)
-    launch_splitk_reduce(slabs, rsc, nullptr, Kout, Ncrs, Ncrs, SK, 0, s);
-  dwperm_rsc_crs_k<<<grid_for((long)Kout * Ncrs), kBlock, 0, st>>>(
-      rsc, dw, Kout, C, R * S);
+  // one fused combine+permute pass (for SK==1 the kernel degenerates to
+  // the permute of the directly-written rsc tensor)
+  splitk_reduce_dwperm_k<<<grid_for((long)Kout * Ncrs), kBlock, 0, st>>>(
+      SK == 1 ? rsc : slabs, dw, Kout, C, R * S, SK);
 }
 
 int conv_db_chunks(long M, int Kout) {
@@ -1398,6 +1427,12 @@ void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
   conv_db_stage1_k<<<g1, kBlock, 0, st>>>(dy, partials, M, Kout, chunks,
                                           K_blk);
   conv_db_stage2_k<<<Kout, kBlock, 0, st>>>(partials, db, Kout, chunks);
+}
+
+void launch_splitk_reduce_dwperm(const float* ws, float* dw, int Kout,
+                                 int C, int RS, int SK, void* s) {
+  splitk_reduce_dwperm_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
+                           (hipStream_t)s>>>(ws, dw, Kout, C, RS, SK);
 }
 
 void launch_dwperm_rsc_crs(const float* in, float* out, int Kout, int C,
