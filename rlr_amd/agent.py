@@ -109,7 +109,8 @@ class Agent:
                 model = gm.model
                 if (self.device.type == 'cuda' and not Fo.force_eager()
                         and getattr(model, 'manual_step', None) is not None
-                        and model.compute_dtype is None):
+                        and (model.compute_dtype is None
+                             or getattr(model, 'manual_bf16_ok', False))):
                     model.manual_step(inputs, labels, gm.dloss_ones())
                 else:
                     gm.zero_grad()

@@ -46,7 +46,8 @@ class TrainEngine:
         gm, args = self.gm, self.args
         model = gm.model
         if (getattr(model, 'manual_step', None) is not None
-                and model.compute_dtype is None):
+                and (model.compute_dtype is None
+                     or getattr(model, 'manual_bf16_ok', False))):
             model.manual_step(static_x, static_y, gm.dloss_ones())
         else:
             gm.flat_grads.zero_()

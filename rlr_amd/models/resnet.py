@@ -53,6 +53,51 @@ class BasicBlock(nn.Module):
             sc = self.down_bn(sc)
         return Fo.add_relu(out, sc)
 
+    # ---- manual tape (ResNet18.manual_step) ----
+
+    def _bn_fwd(self, E, bn, x):
+        return E.batchnorm_fwd(x, bn.weight, bn.bias, bn.running_mean,
+                               bn.running_var, bn.momentum, bn.eps, True)
+
+    def _tape_fwd(self, E, x):
+        s = self.conv1.stride[0]
+        o1 = E.conv2d_fwd(x, self.conv1.weight, None, s, 1, False)
+        b1, sm1, sr1 = self._bn_fwd(E, self.bn1, o1)
+        a1 = E.relu_fwd(b1)
+        o2 = E.conv2d_fwd(a1, self.conv2.weight, None, 1, 1, False)
+        b2, sm2, sr2 = self._bn_fwd(E, self.bn2, o2)
+        if self.down_conv is not None:
+            sc = E.conv2d_fwd(x, self.down_conv.weight, None, s, 0, False)
+            sb, smd, srd = self._bn_fwd(E, self.down_bn, sc)
+        else:
+            sc = smd = srd = None
+            sb = x
+        y = E.add_relu_fwd(b2, sb)
+        return y, (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd, y)
+
+    def _tape_bwd(self, E, dy, saved, need_dx):
+        (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd, y) = saved
+        s = self.conv1.stride[0]
+        d = E.relu_bwd(y, dy)
+        g = E.batchnorm_bwd_into(o2, self.bn2.weight, sm2, sr2, d,
+                                 self.bn2.weight.grad, self.bn2.bias.grad)
+        g = E.conv2d_bwd_wdx_into(a1, self.conv2.weight, g, 1, 1, True,
+                                  self.conv2.weight.grad)
+        g = E.relu_bwd(a1, g)
+        g = E.batchnorm_bwd_into(o1, self.bn1.weight, sm1, sr1, g,
+                                 self.bn1.weight.grad, self.bn1.bias.grad)
+        gmain = E.conv2d_bwd_wdx_into(x, self.conv1.weight, g, s, 1,
+                                      need_dx, self.conv1.weight.grad)
+        if self.down_conv is not None:
+            gs = E.batchnorm_bwd_into(sc, self.down_bn.weight, smd, srd, d,
+                                      self.down_bn.weight.grad,
+                                      self.down_bn.bias.grad)
+            gs = E.conv2d_bwd_wdx_into(x, self.down_conv.weight, gs, s, 0,
+                                       need_dx, self.down_conv.weight.grad)
+            return E.add_(gmain, gs) if need_dx else None
+        # identity shortcut: the block-input grad is main + through-grad
+        return E.add_(gmain, d) if need_dx else None
+
 
 class ResNet18(_OpsModel):
     def __init__(self, num_classes=10):
@@ -70,6 +115,8 @@ class ResNet18(_OpsModel):
         self.layers = nn.Sequential(*layers)
         self.fc = nn.Linear(512, num_classes)
 
+    manual_bf16_ok = True  # the tape below handles the bf16 compute path
+
     def forward(self, x):
         x = self._cast_in(x)
         x = Fo.conv2d(x, self.conv1.weight, None, 1, 1)
@@ -77,3 +124,47 @@ class ResNet18(_OpsModel):
         x = self.layers(x)
         x = Fo.global_avg_pool(x)
         return Fo.linear(x, self.fc.weight, self.fc.bias)
+
+    def manual_step(self, x, labels, dloss):
+        """Hand-rolled fwd+bwd (see CNN_MNIST.manual_step): conv/BN grads
+        land directly in their flat_grads views — no autograd accumulate
+        kernels (~62 per step for this model).  The residual joins sum the
+        two incoming gradients with one fused add (commutative: bitwise-
+        equal to autograd's accumulation)."""
+        import torch
+        from ..ops import ext
+        E = ext()
+        if self.compute_dtype is not None and x.dtype != self.compute_dtype:
+            x = x.to(self.compute_dtype)
+        a0 = E.conv2d_fwd(x, self.conv1.weight, None, 1, 1, False)
+        b0, sm0, sr0 = E.batchnorm_fwd(
+            a0, self.bn1.weight, self.bn1.bias, self.bn1.running_mean,
+            self.bn1.running_var, self.bn1.momentum, self.bn1.eps, True)
+        r0 = E.relu_fwd(b0)
+        h = r0
+        saves = []
+        for blk in self.layers:
+            h, sv = blk._tape_fwd(E, h)
+            saves.append(sv)
+        gp = E.gap_fwd(h)
+        out = E.linear_fwd(gp, self.fc.weight, self.fc.bias, False)
+        logits = out.float() if out.dtype != torch.float32 else out
+        loss, softmax = E.cross_entropy_fwd(logits.contiguous(), labels)
+
+        g = E.cross_entropy_bwd(softmax, labels, dloss)
+        if out.dtype != torch.float32:
+            g = g.to(out.dtype)
+        dx_fc, dw_fc, db_fc = E.linear_bwd(gp, self.fc.weight, g)
+        # the fc is 512x10: a copy into the views is cheaper than a
+        # dedicated bf16 out-variant of the gemm
+        self.fc.weight.grad.copy_(dw_fc.view_as(self.fc.weight))
+        self.fc.bias.grad.copy_(db_fc)
+        g = E.gap_bwd(dx_fc.contiguous(), list(h.shape))
+        for blk, sv in zip(reversed(self.layers), reversed(saves)):
+            g = blk._tape_bwd(E, g, sv, True)
+        g = E.relu_bwd(r0, g)
+        g = E.batchnorm_bwd_into(a0, self.bn1.weight, sm0, sr0, g,
+                                 self.bn1.weight.grad, self.bn1.bias.grad)
+        E.conv2d_bwd_wdx_into(x, self.conv1.weight, g, 1, 1, False,
+                              self.conv1.weight.grad)
+        return loss

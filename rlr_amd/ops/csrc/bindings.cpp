@@ -56,6 +56,9 @@ extern "C" {
 void launch_relu_fwd(const float*, float*, long, void*);
 void launch_relu_bwd(const float*, const float*, float*, long, void*);
 void launch_add_relu(const float*, const float*, float*, long, void*);
+void launch_add_inplace(float*, const float*, long, void*);
+void launch_add_inplace_bf16(unsigned short*, const unsigned short*, long,
+                             void*);
 void launch_maxpool2x2_fwd(const float*, float*, uint8_t*, long, int, int,
                            int, int, int, void*);
 void launch_maxpool2x2_bwd(const float*, const uint8_t*, float*, long, int,
@@ -295,6 +298,20 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
                           dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
                           stream_of(dy));
   return dx;
+}
+
+// in-place elementwise add (residual gradient join in the manual tape)
+torch::Tensor add_inplace(torch::Tensor a, torch::Tensor b) {
+  CHK_CUDA(a);
+  TORCH_CHECK(a.numel() == b.numel() && a.scalar_type() == b.scalar_type());
+  if (is_bf16(a))
+    launch_add_inplace_bf16((unsigned short*)a.data_ptr(),
+                            (const unsigned short*)b.data_ptr(), a.numel(),
+                            stream_of(a));
+  else
+    launch_add_inplace(a.data_ptr<float>(), b.data_ptr<float>(), a.numel(),
+                       stream_of(a));
+  return a;
 }
 
 // fused relu-mask maxpool backward (manual tape; C % 4 == 0)
@@ -969,6 +986,59 @@ torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
   return dx;
 }
 
+// bf16/fp32 conv backward with dw written into a caller-owned view and
+// no bias (the ResNet tape: conv layers are bias-free).  Falls back to
+// the fp32 kernels with casts for the thin first layer, like conv2d_bwd.
+torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor dy, int64_t stride,
+                                  int64_t pad, bool need_dx,
+                                  torch::Tensor dw_out) {
+  TORCH_CHECK(x.is_cuda());
+  TORCH_CHECK(dw_out.is_contiguous() && dw_out.numel() == w.numel());
+  if (!is_bf16(x)) {
+    return conv2d_bwd_into(x, w, dy, stride, pad, need_dx, dw_out,
+                           c10::nullopt, c10::nullopt);
+  }
+  x = cl(x, "conv_bwd.x");
+  w = w.contiguous();
+  dy = cl(dy, "conv_bwd.dy");
+  int Nb = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Kout = w.size(0), R = w.size(2), S = w.size(3);
+  int OH = dy.size(2), OW = dy.size(3);
+  auto st = stream_of(x);
+  bool fast = (Kout % 32) == 0 && (C % 8) == 0;
+  if (!fast) {
+    auto dx32 = conv2d_bwd_into(x.to(torch::kFloat), w,
+                                dy.to(torch::kFloat), stride, pad, need_dx,
+                                dw_out, c10::nullopt, c10::nullopt);
+    return dx32.defined() ? dx32.to(torch::kBFloat16) : dx32;
+  }
+  torch::Tensor dxb;
+  if (need_dx) {
+    auto wp = torch::empty({(long)Kout * R * S, C},
+                           w.options().dtype(torch::kBFloat16));
+    launch_wperm_rsko_c_bf16(w.data_ptr<float>(),
+                             (unsigned short*)wp.data_ptr(), Kout, C, R * S,
+                             st);
+    dxb = empty_cl({Nb, C, H, W}, x.options());
+    launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
+                              (const unsigned short*)wp.data_ptr(),
+                              (unsigned short*)dxb.data_ptr(), Nb, C, H, W,
+                              Kout, R, S, OH, OW, (int)stride, (int)pad,
+                              st);
+  }
+  int Ncrs = C * R * S;
+  long Kdim = (long)Nb * OH * OW;
+  int SK = conv_bwd_weight_bf16_splitk(Kout, Ncrs, Kdim);
+  auto ws = torch::empty({((long)SK + 1) * Kout * Ncrs}, w.options());
+  launch_conv_bwd_weight_bf16((const unsigned short*)dy.data_ptr(),
+                              (const unsigned short*)x.data_ptr(),
+                              dw_out.data_ptr<float>(),
+                              ws.data_ptr<float>(), SK, Nb, C, H, W, Kout,
+                              R, S, OH, OW, (int)stride, (int)pad, st);
+  return dxb;
+}
+
 // ------------------------------------------------------------- batchnorm
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
@@ -1035,6 +1105,39 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_bwd(
                   dx.data_ptr<float>(), dw.data_ptr<float>(),
                   db.data_ptr<float>(), Nb, C, HW, 1, stream_of(x));
   return {dx, dw, db};
+}
+
+// manual-tape variant: dw/db written into caller-owned flat-grad views
+torch::Tensor batchnorm_bwd_into(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor save_mean,
+                                 torch::Tensor save_rstd, torch::Tensor dy,
+                                 torch::Tensor dw_out, torch::Tensor db_out) {
+  TORCH_CHECK(x.is_cuda());
+  x = cl(x, "bn_bwd.x");
+  dy = cl(dy, "bn_bwd.dy");
+  int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(dw_out.is_contiguous() && dw_out.numel() == C);
+  TORCH_CHECK(db_out.is_contiguous() && db_out.numel() == C);
+  auto dx = empty_cl({x.size(0), x.size(1), x.size(2), x.size(3)},
+                     x.options());
+  auto scratch = torch::empty({(long)bn_scratch_floats(C)},
+                              x.options().dtype(torch::kFloat));
+  if (is_bf16(x))
+    launch_bn_bwd_bf16((const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)dy.data_ptr(),
+                       w.data_ptr<float>(), save_mean.data_ptr<float>(),
+                       save_rstd.data_ptr<float>(),
+                       scratch.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(),
+                       dw_out.data_ptr<float>(), db_out.data_ptr<float>(),
+                       Nb, C, HW, 1, stream_of(x));
+  else
+    launch_bn_bwd(x.data_ptr<float>(), dy.data_ptr<float>(),
+                  w.data_ptr<float>(), save_mean.data_ptr<float>(),
+                  save_rstd.data_ptr<float>(), scratch.data_ptr<float>(),
+                  dx.data_ptr<float>(), dw_out.data_ptr<float>(),
+                  db_out.data_ptr<float>(), Nb, C, HW, 1, stream_of(x));
+  return dx;
 }
 
 // ---------------------------------------------------------------- poison
@@ -1129,6 +1232,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_into", &conv2d_bwd_into);
   m.def("linear_bwd_into", &linear_bwd_into);
   m.def("dropout_relu_bwd", &dropout_relu_bwd);
+  m.def("batchnorm_bwd_into", &batchnorm_bwd_into);
+  m.def("conv2d_bwd_wdx_into", &conv2d_bwd_wdx_into);
+  m.def("add_", &add_inplace);
   m.def("maxpool2x2_bwd_relu", &maxpool2x2_bwd_relu);
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);

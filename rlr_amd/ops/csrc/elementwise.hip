@@ -72,6 +72,53 @@ void launch_add_relu(const float* a, const float* b, float* y, long n,
 }
 }
 
+// in-place a += b (residual gradient join in the manual tape)
+__global__ void add_inplace_k(float* __restrict__ a,
+                              const float* __restrict__ b, long n) {
+  long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long i = i4; i < n4; i += stride) {
+    float4 va = ((const float4*)a)[i];
+    float4 vb = ((const float4*)b)[i];
+    va.x += vb.x; va.y += vb.y; va.z += vb.z; va.w += vb.w;
+    ((float4*)a)[i] = va;
+  }
+  for (long i = n4 * 4 + i4; i < n; i += stride) a[i] += b[i];
+}
+
+typedef unsigned short ush8 __attribute__((ext_vector_type(8)));
+
+__global__ void add_inplace_bf16_k(unsigned short* __restrict__ a,
+                                   const unsigned short* __restrict__ b,
+                                   long n) {
+  long i1 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n8 = n / 8;
+  for (long i = i1; i < n8; i += stride) {
+    ush8 va = ((const ush8*)a)[i];
+    ush8 vb = ((const ush8*)b)[i];
+    ush8 out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = f2bf_(bf2f_(va[e]) + bf2f_(vb[e]));
+    ((ush8*)a)[i] = out;
+  }
+  for (long i = n8 * 8 + i1; i < n; i += stride)
+    a[i] = f2bf_(bf2f_(a[i]) + bf2f_(b[i]));
+}
+
+extern "C" {
+void launch_add_inplace(float* a, const float* b, long n, void* s) {
+  add_inplace_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(a, b, n);
+}
+void launch_add_inplace_bf16(unsigned short* a, const unsigned short* b,
+                             long n, void* s) {
+  add_inplace_bf16_k<<<grid_for(n / 8 + 1), kBlock, 0, (hipStream_t)s>>>(
+      a, b, n);
+}
+}
+
 // ------------------------------------------------------- maxpool 2x2 NHWC
 
 // channels_last: c is the fastest axis, so threads over the flat index are

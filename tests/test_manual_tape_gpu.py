@@ -11,13 +11,16 @@ import torch
 pytestmark = pytest.mark.gpu
 
 
-def _run_steps(model_cls, data_shape, use_manual, steps=5):
+def _run_steps(model_cls, data_shape, use_manual, steps=5, dtype=None):
     from rlr_amd.flatmodel import FlatParamModel
     from rlr_amd.ops import flat as flat_ops
     from rlr_amd.ops import functional as Fo
 
     torch.manual_seed(7)
-    gm = FlatParamModel(model_cls(), 'cuda:0')
+    model = model_cls()
+    if dtype is not None:
+        model.set_compute_dtype(dtype)
+    gm = FlatParamModel(model, 'cuda:0')
     gm.train()
     gm.set_dropout_seed(99)
     gm.ensure_grad_views()
@@ -50,6 +53,19 @@ def test_manual_tape_bitwise(which):
                   else (CNN_CIFAR, (3, 32, 32)))
     g_auto, p_auto = _run_steps(cls, shape, use_manual=False)
     g_man, p_man = _run_steps(cls, shape, use_manual=True)
+    assert torch.equal(g_auto, g_man), \
+        (g_auto - g_man).abs().max().item()
+    assert torch.equal(p_auto, p_man)
+
+
+@pytest.mark.parametrize('dtype', [None, torch.bfloat16])
+def test_manual_tape_resnet_bitwise(dtype):
+    """ResNet18 tape (residual joins, BN, bf16 path) vs autograd."""
+    from rlr_amd.models import ResNet18
+    g_auto, p_auto = _run_steps(ResNet18, (3, 32, 32), use_manual=False,
+                                steps=3, dtype=dtype)
+    g_man, p_man = _run_steps(ResNet18, (3, 32, 32), use_manual=True,
+                              steps=3, dtype=dtype)
     assert torch.equal(g_auto, g_man), \
         (g_auto - g_man).abs().max().item()
     assert torch.equal(p_auto, p_man)
