@@ -55,15 +55,15 @@ class BasicBlock(nn.Module):
 
     # ---- manual tape (ResNet18.manual_step) ----
 
-    def _bn_fwd(self, E, bn, x):
+    def _bn_fwd(self, E, bn, x, relu=False):
         return E.batchnorm_fwd(x, bn.weight, bn.bias, bn.running_mean,
-                               bn.running_var, bn.momentum, bn.eps, True)
+                               bn.running_var, bn.momentum, bn.eps, True,
+                               relu)
 
     def _tape_fwd(self, E, x):
         s = self.conv1.stride[0]
         o1 = E.conv2d_fwd(x, self.conv1.weight, None, s, 1, False)
-        b1, sm1, sr1 = self._bn_fwd(E, self.bn1, o1)
-        a1 = E.relu_fwd(b1)
+        a1, sm1, sr1 = self._bn_fwd(E, self.bn1, o1, relu=True)  # bn+relu
         o2 = E.conv2d_fwd(a1, self.conv2.weight, None, 1, 1, False)
         b2, sm2, sr2 = self._bn_fwd(E, self.bn2, o2)
         if self.down_conv is not None:
@@ -73,30 +73,33 @@ class BasicBlock(nn.Module):
             sc = smd = srd = None
             sb = x
         y = E.add_relu_fwd(b2, sb)
-        return y, (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd, y)
+        return y, (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd)
 
-    def _tape_bwd(self, E, dy, saved, need_dx):
-        (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd, y) = saved
+    def _tape_bwd(self, E, dy, saved):
+        """dy arrives PRE-MASKED by this block's output relu (the
+        downstream consumer fuses the mask — add_relu_bwd_ or
+        gap_bwd_relu); the returned dx is masked by this block's INPUT
+        relu, so the producer upstream never runs a separate relu_bwd."""
+        (x, o1, sm1, sr1, a1, o2, sm2, sr2, sc, smd, srd) = saved
         s = self.conv1.stride[0]
-        d = E.relu_bwd(y, dy)
+        d = dy
         g = E.batchnorm_bwd_into(o2, self.bn2.weight, sm2, sr2, d,
                                  self.bn2.weight.grad, self.bn2.bias.grad)
         g = E.conv2d_bwd_wdx_into(a1, self.conv2.weight, g, 1, 1, True,
-                                  self.conv2.weight.grad)
-        g = E.relu_bwd(a1, g)
+                                  self.conv2.weight.grad, a1)  # relu fused
         g = E.batchnorm_bwd_into(o1, self.bn1.weight, sm1, sr1, g,
                                  self.bn1.weight.grad, self.bn1.bias.grad)
         gmain = E.conv2d_bwd_wdx_into(x, self.conv1.weight, g, s, 1,
-                                      need_dx, self.conv1.weight.grad)
+                                      True, self.conv1.weight.grad, None)
         if self.down_conv is not None:
             gs = E.batchnorm_bwd_into(sc, self.down_bn.weight, smd, srd, d,
                                       self.down_bn.weight.grad,
                                       self.down_bn.bias.grad)
             gs = E.conv2d_bwd_wdx_into(x, self.down_conv.weight, gs, s, 0,
-                                       need_dx, self.down_conv.weight.grad)
-            return E.add_(gmain, gs) if need_dx else None
-        # identity shortcut: the block-input grad is main + through-grad
-        return E.add_(gmain, d) if need_dx else None
+                                       True, self.down_conv.weight.grad,
+                                       None)
+            return E.add_relu_bwd_(gmain, gs, x)  # join + input-relu mask
+        return E.add_relu_bwd_(gmain, d, x)
 
 
 class ResNet18(_OpsModel):
@@ -137,10 +140,10 @@ class ResNet18(_OpsModel):
         if self.compute_dtype is not None and x.dtype != self.compute_dtype:
             x = x.to(self.compute_dtype)
         a0 = E.conv2d_fwd(x, self.conv1.weight, None, 1, 1, False)
-        b0, sm0, sr0 = E.batchnorm_fwd(
+        r0, sm0, sr0 = E.batchnorm_fwd(
             a0, self.bn1.weight, self.bn1.bias, self.bn1.running_mean,
-            self.bn1.running_var, self.bn1.momentum, self.bn1.eps, True)
-        r0 = E.relu_fwd(b0)
+            self.bn1.running_var, self.bn1.momentum, self.bn1.eps, True,
+            True)  # bn+relu fused
         h = r0
         saves = []
         for blk in self.layers:
@@ -159,12 +162,13 @@ class ResNet18(_OpsModel):
         # dedicated bf16 out-variant of the gemm
         self.fc.weight.grad.copy_(dw_fc.view_as(self.fc.weight))
         self.fc.bias.grad.copy_(db_fc)
-        g = E.gap_bwd(dx_fc.contiguous(), list(h.shape))
+        # last block's output-relu mask fused into the gap gradient
+        g = E.gap_bwd_relu(dx_fc.contiguous(), h, list(h.shape))
         for blk, sv in zip(reversed(self.layers), reversed(saves)):
-            g = blk._tape_bwd(E, g, sv, True)
-        g = E.relu_bwd(r0, g)
+            g = blk._tape_bwd(E, g, sv)
+        # g is already masked by r0 (block 0's fused join)
         g = E.batchnorm_bwd_into(a0, self.bn1.weight, sm0, sr0, g,
                                  self.bn1.weight.grad, self.bn1.bias.grad)
         E.conv2d_bwd_wdx_into(x, self.conv1.weight, g, 1, 1, False,
-                              self.conv1.weight.grad)
+                              self.conv1.weight.grad, None)
         return loss

@@ -79,6 +79,18 @@ void launch_conv_bwd_data_relu(const float*, const float*, float*,
                                int, int, int, int, int, void*);
 void launch_gap_fwd(const float*, float*, long, int, int, void*);
 void launch_gap_bwd(const float*, float*, long, int, int, void*);
+void launch_gap_bwd_relu(const float*, const float*, float*, long, int,
+                         int, void*);
+void launch_gap_bwd_relu_bf16(const unsigned short*, const unsigned short*,
+                              unsigned short*, long, int, int, void*);
+void launch_add_relu_bwd(float*, const float*, const float*, long, void*);
+void launch_add_relu_bwd_bf16(unsigned short*, const unsigned short*,
+                              const unsigned short*, long, void*);
+void launch_conv_bwd_data_bf16_relu(const unsigned short*,
+                                    const unsigned short*, unsigned short*,
+                                    const unsigned short*, int, int, int,
+                                    int, int, int, int, int, int, int, int,
+                                    void*);
 void launch_ce_fwd(const float*, const long*, float*, float*, float*, int,
                    int, void*);
 void launch_ce_bwd(const float*, const long*, const float*, float*, int, int,
@@ -181,7 +193,8 @@ void launch_nhwc_unflatten_bf16(const unsigned short*, unsigned short*,
 // batchnorm bf16
 void launch_bn_fwd_bf16(const unsigned short*, const float*, const float*,
                         float*, float*, float*, float*, unsigned short*,
-                        float*, int, int, int, float, float, int, void*);
+                        float*, int, int, int, float, float, int, int,
+                        void*);
 void launch_bn_bwd_bf16(const unsigned short*, const unsigned short*,
                         const float*, const float*, const float*, float*,
                         unsigned short*, float*, float*, int, int, int, int,
@@ -190,7 +203,7 @@ void launch_bn_bwd_bf16(const unsigned short*, const unsigned short*,
 int bn_scratch_floats(int);
 void launch_bn_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, float*, float*, int, int, int, float,
-                   float, int, void*);
+                   float, int, int, void*);
 void launch_bn_bwd(const float*, const float*, const float*, const float*,
                    const float*, float*, float*, float*, float*, int, int,
                    int, int, void*);
@@ -297,6 +310,42 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx,
     launch_maxpool2x2_bwd(dy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
                           dx.data_ptr<float>(), Nb, H, W, OH, OW, C,
                           stream_of(dy));
+  return dx;
+}
+
+// masked join: a = (relu_y>0) ? a+b : 0 (manual tape — see
+// add_relu_bwd_* kernels)
+torch::Tensor add_relu_bwd_(torch::Tensor a, torch::Tensor b,
+                            torch::Tensor relu_y) {
+  CHK_CUDA(a);
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() == relu_y.numel());
+  if (is_bf16(a))
+    launch_add_relu_bwd_bf16((unsigned short*)a.data_ptr(),
+                             (const unsigned short*)b.data_ptr(),
+                             (const unsigned short*)relu_y.data_ptr(),
+                             a.numel(), stream_of(a));
+  else
+    launch_add_relu_bwd(a.data_ptr<float>(), b.data_ptr<float>(),
+                        relu_y.data_ptr<float>(), a.numel(), stream_of(a));
+  return a;
+}
+
+// masked gap backward (manual tape)
+torch::Tensor gap_bwd_relu(torch::Tensor dy, torch::Tensor relu_y,
+                           std::vector<int64_t> in_shape) {
+  CHK_CUDA(dy);
+  int Nb = in_shape[0], C = in_shape[1];
+  int HW = in_shape[2] * in_shape[3];
+  auto dx = empty_cl(in_shape, relu_y.options());
+  relu_y = cl(relu_y, "gap_bwd.relu");
+  if (is_bf16(relu_y))
+    launch_gap_bwd_relu_bf16((const unsigned short*)dy.data_ptr(),
+                             (const unsigned short*)relu_y.data_ptr(),
+                             (unsigned short*)dx.data_ptr(), Nb, HW, C,
+                             stream_of(dy));
+  else
+    launch_gap_bwd_relu(dy.data_ptr<float>(), relu_y.data_ptr<float>(),
+                        dx.data_ptr<float>(), Nb, HW, C, stream_of(dy));
   return dx;
 }
 
@@ -992,12 +1041,13 @@ torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
                                   torch::Tensor dy, int64_t stride,
                                   int64_t pad, bool need_dx,
-                                  torch::Tensor dw_out) {
+                                  torch::Tensor dw_out,
+                                  c10::optional<torch::Tensor> relu_y) {
   TORCH_CHECK(x.is_cuda());
   TORCH_CHECK(dw_out.is_contiguous() && dw_out.numel() == w.numel());
   if (!is_bf16(x)) {
     return conv2d_bwd_into(x, w, dy, stride, pad, need_dx, dw_out,
-                           c10::nullopt, c10::nullopt);
+                           c10::nullopt, relu_y);
   }
   x = cl(x, "conv_bwd.x");
   w = w.contiguous();
@@ -1008,9 +1058,11 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
   auto st = stream_of(x);
   bool fast = (Kout % 32) == 0 && (C % 8) == 0;
   if (!fast) {
-    auto dx32 = conv2d_bwd_into(x.to(torch::kFloat), w,
-                                dy.to(torch::kFloat), stride, pad, need_dx,
-                                dw_out, c10::nullopt, c10::nullopt);
+    auto dx32 = conv2d_bwd_into(
+        x.to(torch::kFloat), w, dy.to(torch::kFloat), stride, pad, need_dx,
+        dw_out, c10::nullopt,
+        relu_y ? c10::optional<torch::Tensor>(relu_y->to(torch::kFloat))
+               : c10::nullopt);
     return dx32.defined() ? dx32.to(torch::kBFloat16) : dx32;
   }
   torch::Tensor dxb;
@@ -1021,11 +1073,17 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
                              (unsigned short*)wp.data_ptr(), Kout, C, R * S,
                              st);
     dxb = empty_cl({Nb, C, H, W}, x.options());
-    launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
-                              (const unsigned short*)wp.data_ptr(),
-                              (unsigned short*)dxb.data_ptr(), Nb, C, H, W,
-                              Kout, R, S, OH, OW, (int)stride, (int)pad,
-                              st);
+    const unsigned short* ry = nullptr;
+    torch::Tensor ryt;
+    if (relu_y) {
+      ryt = cl(*relu_y, "conv_bwd.relu_y");
+      ry = (const unsigned short*)ryt.data_ptr();
+    }
+    launch_conv_bwd_data_bf16_relu((const unsigned short*)dy.data_ptr(),
+                                   (const unsigned short*)wp.data_ptr(),
+                                   (unsigned short*)dxb.data_ptr(), ry, Nb,
+                                   C, H, W, Kout, R, S, OH, OW, (int)stride,
+                                   (int)pad, st);
   }
   int Ncrs = C * R * S;
   long Kdim = (long)Nb * OH * OW;
@@ -1044,7 +1102,7 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor b,
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
-    double eps, bool training) {
+    double eps, bool training, bool relu) {
   TORCH_CHECK(x.is_cuda());
   x = cl(x, "bn_fwd.x");
   int Nb = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
@@ -1064,7 +1122,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                        (unsigned short*)y.data_ptr(),
                        scratch.data_ptr<float>(), Nb, C, HW,
                        (float)momentum, (float)eps, training ? 1 : 0,
-                       stream_of(x));
+                       relu ? 1 : 0, stream_of(x));
   else
     launch_bn_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
                   b.data_ptr<float>(), running_mean.data_ptr<float>(),
@@ -1072,7 +1130,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> batchnorm_fwd(
                   save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                   y.data_ptr<float>(), scratch.data_ptr<float>(), Nb, C,
                   HW, (float)momentum, (float)eps, training ? 1 : 0,
-                  stream_of(x));
+                  relu ? 1 : 0, stream_of(x));
   return {y, save_mean, save_rstd};
 }
 
@@ -1235,6 +1293,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batchnorm_bwd_into", &batchnorm_bwd_into);
   m.def("conv2d_bwd_wdx_into", &conv2d_bwd_wdx_into);
   m.def("add_", &add_inplace);
+  m.def("add_relu_bwd_", &add_relu_bwd_);
+  m.def("gap_bwd_relu", &gap_bwd_relu);
   m.def("maxpool2x2_bwd_relu", &maxpool2x2_bwd_relu);
   m.def("batchnorm_fwd", &batchnorm_fwd);
   m.def("batchnorm_bwd", &batchnorm_bwd);

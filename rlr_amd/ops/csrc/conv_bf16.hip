@@ -179,7 +179,8 @@ __global__ __launch_bounds__(256)
 void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
                           const unsigned short* __restrict__ wp,  // [(r,s,ko)][C]
                           unsigned short* __restrict__ dx, ConvShapeB sh,
-                          int Kdim) {
+                          int Kdim,
+                          const unsigned short* __restrict__ relu_y) {
   constexpr int BM = 128, BN = 64, MI = 4, NI = 2;
   __shared__ unsigned short A_lds[2][BM * LDA_B];
   __shared__ unsigned short B_lds[2][BN * LDT_B];  // transposed [c][k]
@@ -324,7 +325,9 @@ void conv_bwd_data_bf16_k(const unsigned short* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         long m = m_blk + wr * 64 + mi * 16 + l4 * 4 + r;
         if (m >= M) continue;
-        dx[m * sh.C + c] = f2bf_(acc[mi][ni][r]);
+        float v = acc[mi][ni][r];
+        if (relu_y && bf2f_(relu_y[m * sh.C + c]) <= 0.f) v = 0.f;
+        dx[m * sh.C + c] = f2bf_(v);
       }
     }
 }
@@ -612,22 +615,36 @@ void launch_conv_fwd_bf16(const unsigned short* x, const unsigned short* wt,
                                                  relu);
 }
 
-void launch_conv_bwd_data_bf16(const unsigned short* dy,
-                               const unsigned short* wp, unsigned short* dx,
-                               int Nb, int C, int H, int W, int Kout, int R,
-                               int S, int OH, int OW, int stride, int pad,
-                               void* s) {
+void launch_conv_bwd_data_bf16_relu(const unsigned short* dy,
+                                    const unsigned short* wp,
+                                    unsigned short* dx,
+                                    const unsigned short* relu_y, int Nb,
+                                    int C, int H, int W, int Kout, int R,
+                                    int S, int OH, int OW, int stride,
+                                    int pad, void* s) {
   ConvShapeB sh{Nb, C, H, W, Kout, R, S, OH, OW, stride, pad};
   int Kdim = Kout * R * S;
   long M = (long)Nb * H * W;
   dim3 grid((M + 127) / 128, (C + 63) / 64, 1);
   hipStream_t st = (hipStream_t)s;
   if (stride == 1)
-    conv_bwd_data_bf16_k<1><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_bf16_k<1><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim,
+                                                  relu_y);
   else if (stride == 2)
-    conv_bwd_data_bf16_k<2><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_bf16_k<2><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim,
+                                                  relu_y);
   else
-    conv_bwd_data_bf16_k<0><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim);
+    conv_bwd_data_bf16_k<0><<<grid, 256, 0, st>>>(dy, wp, dx, sh, Kdim,
+                                                  relu_y);
+}
+
+void launch_conv_bwd_data_bf16(const unsigned short* dy,
+                               const unsigned short* wp, unsigned short* dx,
+                               int Nb, int C, int H, int W, int Kout, int R,
+                               int S, int OH, int OW, int stride, int pad,
+                               void* s) {
+  launch_conv_bwd_data_bf16_relu(dy, wp, dx, nullptr, Nb, C, H, W, Kout, R,
+                                 S, OH, OW, stride, pad, s);
 }
 
 // variant 0: 64-wide tile depth-2; 1: 128-wide; 2: 64-wide depth-3 ring

@@ -108,7 +108,64 @@ __global__ void add_inplace_bf16_k(unsigned short* __restrict__ a,
     a[i] = f2bf_(bf2f_(a[i]) + bf2f_(b[i]));
 }
 
+// masked in-place join: a = (y>0) ? a+b : 0 — the residual gradient join
+// fused with the upstream relu mask (manual tape; bitwise-equal to add
+// followed by relu_bwd since masking commutes with the add)
+__global__ void add_relu_bwd_bf16_k(unsigned short* __restrict__ a,
+                                    const unsigned short* __restrict__ b,
+                                    const unsigned short* __restrict__ y,
+                                    long n) {
+  long i1 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n8 = n / 8;
+  for (long i = i1; i < n8; i += stride) {
+    ush8 va = ((const ush8*)a)[i];
+    ush8 vb = ((const ush8*)b)[i];
+    ush8 vy = ((const ush8*)y)[i];
+    ush8 out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = bf2f_(vy[e]) > 0.f
+                   ? f2bf_(bf2f_(va[e]) + bf2f_(vb[e]))
+                   : (unsigned short)0;
+    ((ush8*)a)[i] = out;
+  }
+  for (long i = n8 * 8 + i1; i < n; i += stride)
+    a[i] = bf2f_(y[i]) > 0.f ? f2bf_(bf2f_(a[i]) + bf2f_(b[i]))
+                             : (unsigned short)0;
+}
+
+__global__ void add_relu_bwd_f32_k(float* __restrict__ a,
+                                   const float* __restrict__ b,
+                                   const float* __restrict__ y, long n) {
+  long i4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long i = i4; i < n4; i += stride) {
+    float4 va = ((const float4*)a)[i];
+    float4 vb = ((const float4*)b)[i];
+    float4 vy = ((const float4*)y)[i];
+    va.x = vy.x > 0.f ? va.x + vb.x : 0.f;
+    va.y = vy.y > 0.f ? va.y + vb.y : 0.f;
+    va.z = vy.z > 0.f ? va.z + vb.z : 0.f;
+    va.w = vy.w > 0.f ? va.w + vb.w : 0.f;
+    ((float4*)a)[i] = va;
+  }
+  for (long i = n4 * 4 + i4; i < n; i += stride)
+    a[i] = y[i] > 0.f ? a[i] + b[i] : 0.f;
+}
+
 extern "C" {
+void launch_add_relu_bwd(float* a, const float* b, const float* y, long n,
+                         void* s) {
+  add_relu_bwd_f32_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(
+      a, b, y, n);
+}
+void launch_add_relu_bwd_bf16(unsigned short* a, const unsigned short* b,
+                              const unsigned short* y, long n, void* s) {
+  add_relu_bwd_bf16_k<<<grid_for(n / 8 + 1), kBlock, 0, (hipStream_t)s>>>(
+      a, b, y, n);
+}
 void launch_add_inplace(float* a, const float* b, long n, void* s) {
   add_inplace_k<<<grid_for(n / 4 + 1), kBlock, 0, (hipStream_t)s>>>(a, b, n);
 }
@@ -847,5 +904,50 @@ void launch_nhwc_unflatten_bf16(const unsigned short* in,
   dim3 g((C + 31) / 32, (HW + 31) / 32, B);
   batch_transpose_k<unsigned short><<<g, 256, 0, (hipStream_t)s>>>(in, out,
                                                                    C, HW);
+}
+}
+
+// masked gap backward (manual tape): dx = (y>0) ? dy[b,c]/HW : 0 — the
+// last residual block's add_relu backward folded into the gap gradient
+__global__ void gap_bwd_relu_k(const float* __restrict__ dy,
+                               const float* __restrict__ y,
+                               float* __restrict__ dx, long B, int HW,
+                               int C) {
+  long n = B * (long)HW * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / ((long)HW * C);
+    dx[i] = y[i] > 0.f ? dy[b * C + c] / HW : 0.f;
+  }
+}
+
+__global__ void gap_bwd_relu_bf16_k(const unsigned short* __restrict__ dy,
+                                    const unsigned short* __restrict__ y,
+                                    unsigned short* __restrict__ dx, long B,
+                                    int HW, int C) {
+  long n = B * (long)HW * C;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int c = i % C;
+    long b = i / ((long)HW * C);
+    dx[i] = bf2f_(y[i]) > 0.f ? f2bf_(bf2f_(dy[b * C + c]) / HW)
+                              : (unsigned short)0;
+  }
+}
+
+extern "C" {
+void launch_gap_bwd_relu(const float* dy, const float* y, float* dx, long B,
+                         int HW, int C, void* s) {
+  gap_bwd_relu_k<<<grid_for(B * (long)HW * C), kBlock, 0,
+                   (hipStream_t)s>>>(dy, y, dx, B, HW, C);
+}
+void launch_gap_bwd_relu_bf16(const unsigned short* dy,
+                              const unsigned short* y, unsigned short* dx,
+                              long B, int HW, int C, void* s) {
+  gap_bwd_relu_bf16_k<<<grid_for(B * (long)HW * C), kBlock, 0,
+                        (hipStream_t)s>>>(dy, y, dx, B, HW, C);
 }
 }

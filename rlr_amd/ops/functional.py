@@ -266,7 +266,8 @@ class _BatchNorm(torch.autograd.Function):
     def forward(ctx, x, w, b, running_mean, running_var, momentum, eps,
                 training):
         y, save_mean, save_rstd = ext().batchnorm_fwd(
-            x, w, b, running_mean, running_var, momentum, eps, training)
+            x, w, b, running_mean, running_var, momentum, eps, training,
+            False)
         ctx.save_for_backward(x, w, save_mean, save_rstd)
         return y
 
