@@ -306,7 +306,8 @@ def test_batchnorm_train_fwd_bwd(ext):
     rm = torch.zeros(c, device=DEV)
     rv = torch.ones(c, device=DEV)
     rm2, rv2 = rm.clone(), rv.clone()
-    y, sm, sr = ext.batchnorm_fwd(x, wt, b, rm, rv, 0.1, 1e-5, True)
+    y, sm, sr = ext.batchnorm_fwd(x, wt, b, rm, rv, 0.1, 1e-5, True,
+                                  False)
 
     xg = x.clone().requires_grad_(True)
     wg = wt.clone().requires_grad_(True)
@@ -332,7 +333,7 @@ def test_batchnorm_eval(ext):
     rm = torch.randn(c, device=DEV)
     rv = torch.rand(c, device=DEV) + 0.5
     y, _, _ = ext.batchnorm_fwd(x, wt, b, rm.clone(), rv.clone(), 0.1, 1e-5,
-                                False)
+                                False, False)
     ref = torch.nn.functional.batch_norm(x, rm, rv, wt, b, False, 0.1, 1e-5)
     t_allclose(y, ref, rtol=1e-4, atol=1e-4, msg='bn eval')
 
