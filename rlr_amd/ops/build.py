@@ -22,7 +22,7 @@ ARCH = os.environ.get('PYTORCH_ROCM_ARCH', 'gfx950')
 
 KERNEL_SOURCES = [
     'elementwise.hip', 'flatopt.hip', 'aggregation.hip', 'gemm_f32.hip',
-    'gemm_bf16.hip', 'conv_f32.hip', 'conv_bf16.hip', 'batchnorm.hip',
+    'gemm_bf16.hip', 'conv_f32.hip', 'conv_bf16.hip', 'conv_bwdw_tap.hip', 'batchnorm.hip',
     'poison.hip',
 ]
 BINDINGS = 'bindings.cpp'

@@ -1,0 +1,208 @@
+// Tap-accumulator conv backward-weight (bf16, 3x3 stride-1 pad-1,
+// W % 8 == 0 and W in {8,16,32}, H % rows_per_tile == 0, C % 32 == 0,
+// Kout % 32 == 0) — the ResNet18 hot shapes.
+//
+// The implicit-GEMM bwd-weight kernel re-reads x once per (r,s) tap and
+// dy once per crs-tile: ~600 MB of L2/L3 traffic per ResNet layer, ~16x
+// its MFMA floor (rocprofv3 r02 evidence).  This kernel streams BOTH
+// tensors ONCE per (ko,c) tile: a block owns a 32x32 (ko,c) tile and a
+// group of images; x rows pass through a sliding LDS ring holding THREE
+// horizontally pre-shifted copies of each row (so every tap's 8-element
+// fragment is an aligned ds_read_b128 — an unshifted layout would need
+// misaligned LDS reads), and all NINE taps accumulate simultaneously into
+// 9 MFMA accumulators per wave.  The zero halo baked into the ring edges
+// IS the pad-1 semantics.  Per-image-group partial tiles are written in
+// the final (ko,c,r,s) order and summed by a fixed-order combine
+// (deterministic, no atomics).
+//
+//   grid = (Kout/32, C/32, S);  S image-groups bound the slab memory for
+//   the deep 512-channel layers.
+#include "common.h"
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+
+template <int LOGW>
+__global__ __launch_bounds__(256)
+void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
+                          const unsigned short* __restrict__ x,
+                          float* __restrict__ partials, int Nb, int C,
+                          int H, int Kout, int n_per_block) {
+  constexpr int W = 1 << LOGW;
+  constexpr int RPT = 32 / W;         // rows per 32-pixel k-tile
+  constexpr int RING = RPT + 2;
+  constexpr int RS = W + 16;          // in-row stride (multiple of 8)
+  constexpr int KOT = 32, CT = 32;
+  // per-c row stride: multiple of 8 (16-B aligned fragment reads) with a
+  // dword spread that de-conflicts the 16-lane b128 read groups
+  constexpr int CSTRIDE = RING * RS + 8;
+
+  __shared__ __align__(16) unsigned short x_lds[3][CT][CSTRIDE];
+  __shared__ __align__(16) unsigned short dy_lds[KOT][32 + 8];
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 over (ko, c)
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int ko0 = blockIdx.x * KOT;
+  const int c0 = blockIdx.y * CT;
+  const int n_lo = blockIdx.z * n_per_block;
+  const int n_hi = min(Nb, n_lo + n_per_block);
+  const int tiles = H / RPT;
+
+  // zero once: stage writes never touch the halo positions, so the ring
+  // edges stay zero for the whole block (pad-1 for free)
+  for (int i = t; i < 3 * CT * CSTRIDE; i += 256)
+    ((unsigned short*)x_lds)[i] = 0;
+  __syncthreads();
+
+  f32x4 acc[9];
+#pragma unroll
+  for (int i = 0; i < 9; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  // per-lane fragment decomposition of the pixel chunk l4*8 (W%8==0:
+  // a chunk never crosses a row)
+  const int mrow_l = (l4 * 8) >> LOGW;
+  const int col_l = (l4 * 8) & (W - 1);
+
+  // stage one x row (row in [-1, H]; outside -> zeros) into ring `slot`,
+  // all three shifted copies: copy s stores x[row][col] at position
+  // col + 1 - s + 8, so a read at aligned col0 + 8 yields x[col0 + s - 1]
+  const int oct_per_row = W * CT / 8;
+  auto stage_row = [&](int n, int row, int slot) {
+    bool valid = (unsigned)row < (unsigned)H;
+    for (int o = t; o < oct_per_row; o += 256) {
+      int col = o >> 2;             // CT/8 = 4 octets per column
+      int coct = (o & 3) * 8;
+      bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (valid)
+        q = *(const bf16x8*)(x +
+                             ((((long)n * H + row) << LOGW) + col) * C +
+                             c0 + coct);
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        int pos = slot * RS + col + 1 - s + 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) x_lds[s][coct + e][pos] =
+            ((const unsigned short*)&q)[e];
+      }
+    }
+  };
+
+  auto stage_dy = [&](int n, int trow0) {
+    // 32 pixels x 4 ko-octets, transposed to [ko][pixel]
+    long pix0 = ((long)n * H + trow0) << LOGW;
+    for (int o = t; o < 128; o += 256) {
+      int pix = o >> 2;
+      int koct = (o & 3) * 8;
+      bf16x8 q = *(const bf16x8*)(dy + (pix0 + pix) * Kout + ko0 + koct);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        dy_lds[koct + e][pix] = ((const unsigned short*)&q)[e];
+    }
+  };
+
+  for (int n = n_lo; n < n_hi; ++n) {
+    // warm the ring: rows -1 .. RPT (slot(row) = (row+1) % RING)
+#pragma unroll
+    for (int r = -1; r <= RPT; ++r) stage_row(n, r, (r + 1) % RING);
+    for (int tile = 0; tile < tiles; ++tile) {
+      int trow0 = tile * RPT;
+      stage_dy(n, trow0);
+      __syncthreads();
+      bf16x8 a = *(const bf16x8*)&dy_lds[wr * 16 + l15][l4 * 8];
+      const int c_lane = wc * 16 + l15;
+#pragma unroll
+      for (int r = 0; r < 3; ++r) {
+        int row = trow0 + mrow_l + r - 1;
+        int slot = (row + 1) % RING;
+        const unsigned short* base =
+            &x_lds[0][c_lane][slot * RS + col_l + 8];
+#pragma unroll
+        for (int s = 0; s < 3; ++s) {
+          bf16x8 b = *(const bf16x8*)(base + (long)s * CT * CSTRIDE);
+          acc[r * 3 + s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[r * 3 + s], 0, 0, 0);
+        }
+      }
+      __syncthreads();  // drain reads before the ring advances
+      if (tile + 1 < tiles) {
+        // rows for the next tile: trow0+RPT+1 .. trow0+2*RPT
+#pragma unroll
+        for (int r2 = 1; r2 <= RPT; ++r2) {
+          int row = trow0 + RPT + r2;
+          stage_row(n, row, (row + 1) % RING);
+        }
+      }
+    }
+    __syncthreads();  // image boundary: ring refill starts clean
+  }
+
+  // write this block's (ko,c) tile partials in final (ko,c,r,s) order
+  long zbase = (long)blockIdx.z * Kout * C * 9;
+#pragma unroll
+  for (int rs = 0; rs < 9; ++rs)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int ko = ko0 + wr * 16 + l4 * 4 + e;
+      int c = c0 + wc * 16 + l15;
+      partials[zbase + ((long)ko * C + c) * 9 + rs] = acc[rs][e];
+    }
+}
+
+// fixed-order combine: dw[(ko*C + c)*9 + rs] = sum over image groups
+__global__ void bwdw_tap_combine_k(const float* __restrict__ partials,
+                                   float* __restrict__ dw, long n_out,
+                                   int S) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += stride) {
+    float a = 0.f;
+    for (int z = 0; z < S; ++z) a += partials[(long)z * n_out + i];
+    dw[i] = a;
+  }
+}
+
+extern "C" {
+
+// 1 if the tap kernel covers this shape
+int conv_bwdw_tap_ok(int C, int H, int W, int Kout, int R, int S,
+                     int stride, int pad) {
+  if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
+  if ((C % 32) || (Kout % 32)) return 0;
+  if (W != 8 && W != 16 && W != 32) return 0;
+  int rpt = 32 / W;
+  return (H % rpt) == 0;
+}
+
+// number of image-group slabs (bounds ws to ~64 MB of fp32 partials)
+int conv_bwdw_tap_slabs(int Nb, int C, int Kout) {
+  long kc9 = (long)Kout * C * 9 * 4;
+  long s = (64L << 20) / kc9;
+  if (s < 1) s = 1;
+  if (s > Nb) s = Nb;
+  return (int)s;
+}
+
+void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
+                               const unsigned short* x, float* dw,
+                               float* ws, int Nb, int C, int H, int W,
+                               int Kout, void* st) {
+  hipStream_t s = (hipStream_t)st;
+  int S = conv_bwdw_tap_slabs(Nb, C, Kout);
+  int G = (Nb + S - 1) / S;
+  S = (Nb + G - 1) / G;
+  dim3 grid(Kout / 32, C / 32, S);
+  if (W == 32)
+    conv_bwdw_tap_bf16_k<5><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
+                                                 G);
+  else if (W == 16)
+    conv_bwdw_tap_bf16_k<4><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
+                                                 G);
+  else
+    conv_bwdw_tap_bf16_k<3><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
+                                                 G);
+  long n_out = (long)Kout * C * 9;
+  bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws, dw, n_out, S);
+}
+}
