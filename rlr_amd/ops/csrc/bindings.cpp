@@ -892,7 +892,8 @@ static void bf16_dw(const torch::Tensor& dy, const torch::Tensor& x,
                     int stride, int pad, void* st) {
   if (conv_bwdw_tap_ok(C, H, W, Kout, R, S, stride, pad)) {
     int SL = conv_bwdw_tap_slabs(Nb, C, Kout);
-    auto ws = torch::empty({(long)SL * Kout * C * 9},
+    // +16 slabs: scratch for the two-stage combine's chunk sums
+    auto ws = torch::empty({(long)(SL + 16) * Kout * C * 9},
                            w.options().dtype(torch::kFloat));
     launch_conv_bwdw_tap_bf16((const unsigned short*)dy.data_ptr(),
                               (const unsigned short*)x.data_ptr(),

@@ -30,9 +30,7 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
                           int H, int Kout, int n_per_block) {
   constexpr int W = 1 << LOGW;
   constexpr int RPT = 32 / W;         // rows per 32-pixel k-tile
-  // ring holds TWO tiles of rows (+halo): tile t computes while t+1's dy
-  // and t+2's rows stage — ONE barrier per tile instead of three
-  constexpr int RING = 2 * RPT + 2;
+  constexpr int RING = RPT + 2;
   constexpr int RS = W + 16;          // in-row stride (multiple of 8)
   constexpr int KOT = 32, CT = 32;
   // per-c row stride: multiple of 8 (16-B aligned fragment reads) with a
@@ -40,7 +38,7 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   constexpr int CSTRIDE = RING * RS + 8;
 
   __shared__ __align__(16) unsigned short x_lds[3][CT][CSTRIDE];
-  __shared__ __align__(16) unsigned short dy_lds[2][KOT][32 + 8];
+  __shared__ __align__(16) unsigned short dy_lds[KOT][32 + 8];
 
   const int t = threadIdx.x;
   const int wave = t >> 6, lane = t & 63;
@@ -91,7 +89,7 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
     }
   };
 
-  auto stage_dy = [&](int n, int trow0, int buf) {
+  auto stage_dy = [&](int n, int trow0) {
     // 32 pixels x 4 ko-octets, transposed to [ko][pixel]
     long pix0 = ((long)n * H + trow0) << LOGW;
     for (int o = t; o < 128; o += 256) {
@@ -100,21 +98,19 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
       bf16x8 q = *(const bf16x8*)(dy + (pix0 + pix) * Kout + ko0 + koct);
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        dy_lds[buf][koct + e][pix] = ((const unsigned short*)&q)[e];
+        dy_lds[koct + e][pix] = ((const unsigned short*)&q)[e];
     }
   };
 
   for (int n = n_lo; n < n_hi; ++n) {
-    // warm: rows for tiles 0 and 1 (-1 .. 2*RPT) + tile 0's dy
+    // warm the ring: rows -1 .. RPT (slot(row) = (row+1) % RING)
 #pragma unroll
-    for (int r = -1; r <= 2 * RPT; ++r)
-      if (r <= H) stage_row(n, r, (r + 1) % RING);
-    stage_dy(n, 0, 0);
-    __syncthreads();
+    for (int r = -1; r <= RPT; ++r) stage_row(n, r, (r + 1) % RING);
     for (int tile = 0; tile < tiles; ++tile) {
       int trow0 = tile * RPT;
-      bf16x8 a =
-          *(const bf16x8*)&dy_lds[tile & 1][wr * 16 + l15][l4 * 8];
+      stage_dy(n, trow0);
+      __syncthreads();
+      bf16x8 a = *(const bf16x8*)&dy_lds[wr * 16 + l15][l4 * 8];
       const int c_lane = wc * 16 + l15;
 #pragma unroll
       for (int r = 0; r < 3; ++r) {
@@ -129,18 +125,17 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
               a, b, acc[r * 3 + s], 0, 0, 0);
         }
       }
-      // stage tile t+1's dy and tile t+2's rows while t computes; the
-      // new rows only reuse slots of rows tile t no longer reads
-      if (tile + 1 < tiles) stage_dy(n, trow0 + RPT, (tile + 1) & 1);
-      if (tile + 2 < tiles) {
+      __syncthreads();  // drain reads before the ring advances
+      if (tile + 1 < tiles) {
+        // rows for the next tile: trow0+RPT+1 .. trow0+2*RPT
 #pragma unroll
         for (int r2 = 1; r2 <= RPT; ++r2) {
           int row = trow0 + RPT + r2;
           stage_row(n, row, (row + 1) % RING);
         }
       }
-      __syncthreads();
     }
+    __syncthreads();  // image boundary: ring refill starts clean
   }
 
   // write this block's (ko,c) tile partials in final (ko,c,r,s) order
@@ -156,40 +151,32 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
 }
 
 // fixed-order combine: dw[(ko*C + c)*9 + rs] = sum over image groups.
-// Thread-per-output with 4-deep ILP when slabs are few; wave-per-output
-// (lane-strided slabs + shuffle tree) when the slab count is deep — the
-// first form is parallelism-starved at S=256 (35.8 us measured for a
-// 37 MB reduction).
+// Two stages when the slab count is deep: a thread-per-output z-loop at
+// S=256 is parallelism-starved (36 k threads, one load in flight each —
+// 35.8 us measured for a 37 MB reduction); stage 1 gives every
+// (output, 16-slab chunk) its own thread (coalesced, 16x the threads),
+// stage 2 folds the chunks.
 __global__ void bwdw_tap_combine_k(const float* __restrict__ partials,
                                    float* __restrict__ dw, long n_out,
-                                   int S) {
+                                   int S, int zstride) {
   long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
-       i += stride) {
+  long total = n_out * ((S + zstride - 1) / zstride);
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < total;
+       v += stride) {
+    long i = v % n_out;
+    int chunk = (int)(v / n_out);
+    int z0 = chunk * zstride, z1 = min(S, z0 + zstride);
     float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-    int z = 0;
-    for (; z + 3 < S; z += 4) {
+    int z = z0;
+    for (; z + 3 < z1; z += 4) {
       a0 += partials[(long)z * n_out + i];
       a1 += partials[(long)(z + 1) * n_out + i];
       a2 += partials[(long)(z + 2) * n_out + i];
       a3 += partials[(long)(z + 3) * n_out + i];
     }
-    for (; z < S; ++z) a0 += partials[(long)z * n_out + i];
-    dw[i] = (a0 + a1) + (a2 + a3);
+    for (; z < z1; ++z) a0 += partials[(long)z * n_out + i];
+    dw[(long)chunk * n_out + i] = (a0 + a1) + (a2 + a3);
   }
-}
-
-__global__ void bwdw_tap_combine_wave_k(const float* __restrict__ partials,
-                                        float* __restrict__ dw, long n_out,
-                                        int S) {
-  long i = blockIdx.x * (long)(blockDim.x / kWave) + threadIdx.x / kWave;
-  int lane = threadIdx.x % kWave;
-  if (i >= n_out) return;
-  float a = 0.f;
-  for (int z = lane; z < S; z += kWave) a += partials[(long)z * n_out + i];
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) a += __shfl_down(a, off, kWave);
-  if (lane == 0) dw[i] = a;
 }
 
 extern "C" {
@@ -232,12 +219,18 @@ void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
     conv_bwdw_tap_bf16_k<3><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
                                                  G);
   long n_out = (long)Kout * C * 9;
-  if (S >= 32) {
-    int wpb = kBlock / kWave;
-    bwdw_tap_combine_wave_k<<<(n_out + wpb - 1) / wpb, kBlock, 0, s>>>(
-        ws, dw, n_out, S);
+  if (S > 16) {
+    // stage 1 writes its chunk sums PAST the S slabs (ws is allocated
+    // with 16 extra slabs), then stage 2 folds the chunks
+    int chunks = (S + 15) / 16;
+    float* ws2 = ws + (long)S * n_out;
+    bwdw_tap_combine_k<<<grid_for(n_out * chunks), kBlock, 0, s>>>(
+        ws, ws2, n_out, S, 16);
+    bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws2, dw, n_out,
+                                                          chunks, chunks);
   } else {
-    bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws, dw, n_out, S);
+    bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws, dw, n_out, S,
+                                                          S);
   }
 }
 }
