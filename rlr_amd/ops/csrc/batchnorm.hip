@@ -54,10 +54,17 @@ __global__ void bn_stats1_k(const T* __restrict__ x,
 // partials.  One block covers 16 channels x 16 chunk slices (32-iteration
 // strided loops), slices combined in LDS in fixed order (deterministic).
 // A single-block variant at 512 chunks measured 87 us — this is ~3 us.
+// The fwd finalize (mean/rstd + running stats) is folded in behind
+// `count > 0` — one launch fewer per BN forward (20/step on ResNet18).
 __global__ void bn_stats2_k(const float* __restrict__ partials,
                             float* __restrict__ sums, int C, int chunks,
                             float* __restrict__ copy0,
-                            float* __restrict__ copy1) {
+                            float* __restrict__ copy1, long count,
+                            float momentum, float eps,
+                            float* __restrict__ save_mean,
+                            float* __restrict__ save_rstd,
+                            float* __restrict__ running_mean,
+                            float* __restrict__ running_var) {
   constexpr int CB = 16, SL = 16;  // CB*SL == blockDim.x == 256
   __shared__ float l_s[256], l_ss[256];
   int lc = threadIdx.x % CB;
@@ -84,6 +91,20 @@ __global__ void bn_stats2_k(const float* __restrict__ partials,
     // D2D copies after the fact (12k extra launches per bench, profiles/)
     if (copy0) copy0[c] = s;
     if (copy1) copy1[c] = ss;
+    if (count > 0) {  // fused fwd finalize
+      float mean = s / count;
+      float var = ss / count - mean * mean;
+      var = fmaxf(var, 0.f);
+      save_mean[c] = mean;
+      save_rstd[c] = rsqrtf(var + eps);
+      if (running_mean) {
+        float unbiased = count > 1 ? var * count / (count - 1) : var;
+        running_mean[c] =
+            (1.f - momentum) * running_mean[c] + momentum * mean;
+        running_var[c] =
+            (1.f - momentum) * running_var[c] + momentum * unbiased;
+      }
+    }
   }
 }
 
@@ -246,11 +267,9 @@ static void bn_fwd_impl(const T* x, const float* w, const float* b,
     int chunks = bn_chunks(M);
     dim3 g1((chunks + sub_per - 1) / sub_per, (C + C_blk - 1) / C_blk);
     bn_stats1_k<T><<<g1, 256, 0, st>>>(x, partials, M, C, C_blk);
-    bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, sums, C, chunks,
-                                               nullptr, nullptr);
-    bn_finalize_k<<<(C + 255) / 256, 256, 0, st>>>(
-        sums, save_mean, save_rstd, running_mean, running_var, C, M,
-        momentum, eps);
+    bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(
+        partials, sums, C, chunks, nullptr, nullptr, M, momentum, eps,
+        save_mean, save_rstd, running_mean, running_var);
   } else {
     bn_eval_stats_k<<<(C + 255) / 256, 256, 0, st>>>(
         running_mean, running_var, save_mean, save_rstd, C, eps);
@@ -275,7 +294,8 @@ static void bn_bwd_impl(const T* x, const T* dy, const float* w,
   bn_bwd_stats1_k<T><<<g1, 256, 0, st>>>(x, dy, save_mean, save_rstd,
                                          partials, M, C, C_blk);
   bn_stats2_k<<<(C + 15) / 16, 256, 0, st>>>(partials, stats, C, chunks,
-                                             db, dw);
+                                             db, dw, 0, 0.f, 0.f, nullptr,
+                                             nullptr, nullptr, nullptr);
   long n = M * C;
   bn_bwd_dx_k<T><<<bn_grid(n, C), kBlock, 0, st>>>(
       x, dy, w, save_mean, save_rstd, stats, dx, n, C,

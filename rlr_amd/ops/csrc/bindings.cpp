@@ -160,6 +160,11 @@ int conv_bwdw_tap_slabs(int, int, int);
 void launch_conv_bwdw_tap_bf16(const unsigned short*, const unsigned short*,
                                float*, float*, int, int, int, int, int,
                                void*);
+int conv_tap_fwd_ok(int, int, int, int, int, int, int, int);
+void launch_conv_tap_fwd_bf16(const unsigned short*, const unsigned short*,
+                              const float*, unsigned short*,
+                              const unsigned short*, int, int, int, int,
+                              int, int, int, void*);
 void launch_wperm_rsc_ko_bf16(const float*, unsigned short*, int, int, int,
                               void*);
 void launch_wperm_rsko_c_bf16(const float*, unsigned short*, int, int, int,
@@ -860,12 +865,20 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
                                (unsigned short*)wt.data_ptr(), Kout, C,
                                R * S, stream_of(x));
       auto y = empty_cl({Nb, Kout, OH, OW}, x.options());
-      launch_conv_fwd_bf16((const unsigned short*)x.data_ptr(),
-                           (const unsigned short*)wt.data_ptr(),
-                           b ? b->data_ptr<float>() : nullptr,
-                           (unsigned short*)y.data_ptr(), Nb, C, H, W, Kout,
-                           R, S, OH, OW, (int)stride, (int)pad,
-                           relu ? 1 : 0, stream_of(x));
+      if (conv_tap_fwd_ok(C, H, W, Kout, R, S, (int)stride, (int)pad))
+        launch_conv_tap_fwd_bf16((const unsigned short*)x.data_ptr(),
+                                 (const unsigned short*)wt.data_ptr(),
+                                 b ? b->data_ptr<float>() : nullptr,
+                                 (unsigned short*)y.data_ptr(), nullptr,
+                                 Nb, C, H, W, Kout, relu ? 1 : 0, 0,
+                                 stream_of(x));
+      else
+        launch_conv_fwd_bf16((const unsigned short*)x.data_ptr(),
+                             (const unsigned short*)wt.data_ptr(),
+                             b ? b->data_ptr<float>() : nullptr,
+                             (unsigned short*)y.data_ptr(), Nb, C, H, W,
+                             Kout, R, S, OH, OW, (int)stride, (int)pad,
+                             relu ? 1 : 0, stream_of(x));
       return y;
     }
     // first-layer fallback (C=1,3): fp32 kernels, bf16 in/out casts
@@ -949,11 +962,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
                                (unsigned short*)wp.data_ptr(), Kout, C,
                                R * S, st);
       dxb = empty_cl({Nb, C, H, W}, x.options());
-      launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
-                                (const unsigned short*)wp.data_ptr(),
-                                (unsigned short*)dxb.data_ptr(), Nb, C, H,
-                                W, Kout, R, S, OH, OW, (int)stride,
-                                (int)pad, st);
+      if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+        // bwd-data == the same correlation over dy with flipped taps
+        launch_conv_tap_fwd_bf16((const unsigned short*)dy.data_ptr(),
+                                 (const unsigned short*)wp.data_ptr(),
+                                 nullptr, (unsigned short*)dxb.data_ptr(),
+                                 nullptr, Nb, Kout, H, W, C, 0, 1, st);
+      else
+        launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
+                                  (const unsigned short*)wp.data_ptr(),
+                                  (unsigned short*)dxb.data_ptr(), Nb, C,
+                                  H, W, Kout, R, S, OH, OW, (int)stride,
+                                  (int)pad, st);
     }
     auto dw = torch::empty_like(w);
     bf16_dw(dy, x, dw, w, Nb, C, H, W, Kout, R, S, (int)stride, (int)pad,
@@ -1110,11 +1130,17 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
       ryt = cl(*relu_y, "conv_bwd.relu_y");
       ry = (const unsigned short*)ryt.data_ptr();
     }
-    launch_conv_bwd_data_bf16_relu((const unsigned short*)dy.data_ptr(),
-                                   (const unsigned short*)wp.data_ptr(),
-                                   (unsigned short*)dxb.data_ptr(), ry, Nb,
-                                   C, H, W, Kout, R, S, OH, OW, (int)stride,
-                                   (int)pad, st);
+    if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+      launch_conv_tap_fwd_bf16((const unsigned short*)dy.data_ptr(),
+                               (const unsigned short*)wp.data_ptr(),
+                               nullptr, (unsigned short*)dxb.data_ptr(),
+                               ry, Nb, Kout, H, W, C, 0, 1, st);
+    else
+      launch_conv_bwd_data_bf16_relu(
+          (const unsigned short*)dy.data_ptr(),
+          (const unsigned short*)wp.data_ptr(),
+          (unsigned short*)dxb.data_ptr(), ry, Nb, C, H, W, Kout, R, S, OH,
+          OW, (int)stride, (int)pad, st);
   }
   bf16_dw(dy, x, dw_out, w, Nb, C, H, W, Kout, R, S, (int)stride,
           (int)pad, st);

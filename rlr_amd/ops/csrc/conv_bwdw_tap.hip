@@ -234,3 +234,165 @@ void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
   }
 }
 }
+
+// ---------------------------------------------------------------------
+// Tap-resident bf16 conv FORWARD / stride-1 BACKWARD-DATA (3x3 s1 p1,
+// W in {8,16,32}, Cin % 32 == 0, Cout % 32 == 0).
+//
+// The implicit-GEMM fwd/bwd-data kernels re-gather their input once per
+// (r,s) tap (~9x traffic through L2/L3).  Here the contraction runs along
+// the CHANNEL axis, so the input tile lives in LDS in its natural NHWC
+// layout (c contiguous = the MFMA k fragment; the shifted taps merely
+// select a different PIXEL row of the fragment — no alignment tricks
+// needed), staged ONCE per (supertile, c-chunk).  A zero halo column/row
+// ring provides pad-1 for free.  One kernel serves both directions:
+// backward-data is the same correlation with the tap index flipped
+// (w[2-r][2-s]) and the weight matrix already stored as [(r,s,ko)][C].
+//
+//   grid = (Cout/32, Nb); block streams its image's 128-pixel supertiles,
+//   looping c-chunks of 32 with 9-tap MFMA accumulation in registers.
+// ---------------------------------------------------------------------
+
+template <int LOGW>
+__global__ __launch_bounds__(256)
+void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
+                         const unsigned short* __restrict__ wt,
+                         const float* __restrict__ bias,
+                         unsigned short* __restrict__ y,
+                         const unsigned short* __restrict__ relu_y,
+                         int Nb, int Cin, int H, int Cout, int relu,
+                         int flip) {
+  constexpr int W = 1 << LOGW;
+  constexpr int ST = 128;             // pixels per supertile
+  constexpr int RPS = ST / W;         // rows per supertile
+  constexpr int XROWS = RPS + 2;      // + halo
+  constexpr int CP = 36;              // c-dim stride (pad: bank spread)
+  constexpr int COT = 32, CCH = 32;   // out-channel tile, c-chunk
+
+  __shared__ __align__(16) unsigned short x_lds[XROWS * (W + 2)][CP];
+  __shared__ __align__(16) unsigned short w_lds[9][COT][CP];
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 over (pixel-half, ko-half)
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int ko0 = blockIdx.x * COT;
+  const int n = blockIdx.y;
+  const int P = H * W;
+  const int n_super = (P + ST - 1) / ST;
+
+  for (int st = 0; st < n_super; ++st) {
+    const int p0 = st * ST;
+    const int row0 = p0 >> LOGW;          // first image row of this tile
+    f32x4 acc[ST / 32];
+#pragma unroll
+    for (int i = 0; i < ST / 32; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int c0 = 0; c0 < Cin; c0 += CCH) {
+      // ---- stage w chunk: [(rs*Cin + c)][Cout] -> [rs][ko][c] ----
+      // (flip re-indexes rs for backward-data)
+      for (int o = t; o < 9 * CCH * COT / 8; o += 256) {
+        int rsc = o >> 2;                // (rs, c) pair; COT/8 = 4 octets
+        int rs = rsc >> 5;               // CCH = 32
+        int c = rsc & 31;
+        int koct = (o & 3) * 8;
+        bf16x8 q = *(const bf16x8*)(
+            wt + ((long)rs * Cin + c0 + c) * Cout + ko0 + koct);
+        int rsd = flip ? 8 - rs : rs;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          w_lds[rsd][koct + e][c] = ((const unsigned short*)&q)[e];
+      }
+      // ---- stage x rows row0-1 .. row0+RPS (halo-zeroed) ----
+      // layout [xrow * (W+2) + col + 1][c]
+      for (int i = t; i < XROWS * (W + 2); i += 256) {
+        int xr = i / (W + 2);
+        int col = i - xr * (W + 2) - 1;     // -1 .. W (halo edges -> 0)
+        int row = row0 - 1 + xr;
+        bool ok = (unsigned)row < (unsigned)H && (unsigned)col < (unsigned)W;
+        const unsigned short* src =
+            xin + ((((long)n * H + (ok ? row : 0)) << LOGW) +
+                   (ok ? col : 0)) * Cin + c0;
+#pragma unroll
+        for (int oc = 0; oc < 4; ++oc) {
+          bf16x8 qq = {0, 0, 0, 0, 0, 0, 0, 0};
+          if (ok) qq = *(const bf16x8*)(src + oc * 8);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            x_lds[i][oc * 8 + e] = ((const unsigned short*)&qq)[e];
+        }
+      }
+      __syncthreads();
+
+      // ---- 9-tap MFMA over the supertile's 32-pixel tiles ----
+#pragma unroll
+      for (int tt = 0; tt < ST / 32; ++tt) {
+        int p = tt * 32 + wr * 16 + l15;   // pixel within supertile
+        int prow = (p >> LOGW) + 1;        // +1: halo row offset
+        int pcol = (p & (W - 1)) + 1;      // +1: halo col offset
+        bool pval = p0 + p < P;
+        int base = pval ? (prow * (W + 2) + pcol) : (1 * (W + 2) + 1);
+#pragma unroll
+        for (int r = 0; r < 3; ++r) {
+#pragma unroll
+          for (int s = 0; s < 3; ++s) {
+            bf16x8 a = *(const bf16x8*)&x_lds[base + (r - 1) * (W + 2) +
+                                              (s - 1)][l4 * 8];
+            bf16x8 b = *(const bf16x8*)&w_lds[r * 3 + s][wc * 16 + l15]
+                                             [l4 * 8];
+            acc[tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
+                                                              acc[tt], 0,
+                                                              0, 0);
+          }
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: y[p][ko], bias/relu or relu_y mask ----
+#pragma unroll
+    for (int tt = 0; tt < ST / 32; ++tt) {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        int p = p0 + tt * 32 + wr * 16 + l4 * 4 + e;
+        int ko = ko0 + wc * 16 + l15;
+        if (p >= P) continue;
+        long oidx = ((long)n * P + p) * Cout + ko;
+        float v = acc[tt][e];
+        if (bias) v += bias[ko];
+        if (relu) v = fmaxf(v, 0.f);
+        if (relu_y && bf2f_(relu_y[oidx]) <= 0.f) v = 0.f;
+        y[oidx] = f2bf_(v);
+      }
+    }
+  }
+}
+
+extern "C" {
+
+int conv_tap_fwd_ok(int Cin, int H, int W, int Cout, int R, int S,
+                    int stride, int pad) {
+  if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
+  if ((Cin % 32) || (Cout % 32)) return 0;
+  return (W == 8 || W == 16 || W == 32) && H == W;
+}
+
+void launch_conv_tap_fwd_bf16(const unsigned short* x,
+                              const unsigned short* wt, const float* bias,
+                              unsigned short* y,
+                              const unsigned short* relu_y, int Nb,
+                              int Cin, int H, int W, int Cout, int relu,
+                              int flip, void* st) {
+  hipStream_t s = (hipStream_t)st;
+  dim3 grid(Cout / 32, Nb);
+  if (W == 32)
+    conv_tap_fwd_bf16_k<5><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
+                                                Cin, H, Cout, relu, flip);
+  else if (W == 16)
+    conv_tap_fwd_bf16_k<4><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
+                                                Cin, H, Cout, relu, flip);
+  else
+    conv_tap_fwd_bf16_k<3><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
+                                                Cin, H, Cout, relu, flip);
+}
+}
