@@ -253,7 +253,7 @@ void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
 //   looping c-chunks of 32 with 9-tap MFMA accumulation in registers.
 // ---------------------------------------------------------------------
 
-template <int LOGW>
+template <int LOGW, int SPLIT, int G>
 __global__ __launch_bounds__(256)
 void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
                          const unsigned short* __restrict__ wt,
@@ -263,37 +263,42 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
                          int Nb, int Cin, int H, int Cout, int relu,
                          int flip) {
   constexpr int W = 1 << LOGW;
-  constexpr int ST = 128;             // pixels per supertile
-  constexpr int RPS = ST / W;         // rows per supertile
-  constexpr int XROWS = RPS + 2;      // + halo
-  constexpr int CP = 36;              // c-dim stride (pad: bank spread)
-  constexpr int COT = 32, CCH = 32;   // out-channel tile, c-chunk
+  constexpr int ST = LOGW == 3 ? 64 : 128;  // pixels per supertile
+  constexpr int NT = ST / 32;               // 32-pixel tiles per supertile
+  constexpr int RPS = ST / W;               // rows per supertile
+  constexpr int XROWS = RPS + 2;            // + halo
+  constexpr int CP = 36;                    // c stride (bank spread)
+  constexpr int COT = 32, CCH = 32;
 
-  __shared__ __align__(16) unsigned short x_lds[XROWS * (W + 2)][CP];
+  // G images share one block (small deep-layer images would otherwise
+  // re-stage the w chunk once per image); SPLIT spreads a big image's
+  // supertiles over gridDim.z for occupancy.
+  __shared__ __align__(16) unsigned short x_lds[G][XROWS * (W + 2)][CP];
   __shared__ __align__(16) unsigned short w_lds[9][COT][CP];
 
   const int t = threadIdx.x;
   const int wave = t >> 6, lane = t & 63;
-  const int wr = wave >> 1, wc = wave & 1;  // 2x2 over (pixel-half, ko-half)
+  const int wr = wave >> 1, wc = wave & 1;
   const int l15 = lane & 15, l4 = lane >> 4;
   const int ko0 = blockIdx.x * COT;
-  const int n = blockIdx.y;
+  const int n_lo = blockIdx.y * G;
   const int P = H * W;
   const int n_super = (P + ST - 1) / ST;
 
-  for (int st = 0; st < n_super; ++st) {
+  for (int st = blockIdx.z; st < n_super; st += SPLIT) {
     const int p0 = st * ST;
-    const int row0 = p0 >> LOGW;          // first image row of this tile
-    f32x4 acc[ST / 32];
+    const int row0 = p0 >> LOGW;
+    f32x4 acc[G][NT];
 #pragma unroll
-    for (int i = 0; i < ST / 32; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int i = 0; i < NT; ++i) acc[g][i] = {0.f, 0.f, 0.f, 0.f};
 
     for (int c0 = 0; c0 < Cin; c0 += CCH) {
-      // ---- stage w chunk: [(rs*Cin + c)][Cout] -> [rs][ko][c] ----
-      // (flip re-indexes rs for backward-data)
+      // ---- w chunk: [(rs*Cin + c)][Cout] -> [rs][ko][c] (flip: bwd) ----
       for (int o = t; o < 9 * CCH * COT / 8; o += 256) {
-        int rsc = o >> 2;                // (rs, c) pair; COT/8 = 4 octets
-        int rs = rsc >> 5;               // CCH = 32
+        int rsc = o >> 2;
+        int rs = rsc >> 5;
         int c = rsc & 31;
         int koct = (o & 3) * 8;
         bf16x8 q = *(const bf16x8*)(
@@ -303,67 +308,75 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
         for (int e = 0; e < 8; ++e)
           w_lds[rsd][koct + e][c] = ((const unsigned short*)&q)[e];
       }
-      // ---- stage x rows row0-1 .. row0+RPS (halo-zeroed) ----
-      // layout [xrow * (W+2) + col + 1][c]
-      for (int i = t; i < XROWS * (W + 2); i += 256) {
-        int xr = i / (W + 2);
-        int col = i - xr * (W + 2) - 1;     // -1 .. W (halo edges -> 0)
-        int row = row0 - 1 + xr;
-        bool ok = (unsigned)row < (unsigned)H && (unsigned)col < (unsigned)W;
-        const unsigned short* src =
-            xin + ((((long)n * H + (ok ? row : 0)) << LOGW) +
-                   (ok ? col : 0)) * Cin + c0;
+      // ---- x slices: rows row0-1..row0+RPS per image, halo zeroed ----
 #pragma unroll
-        for (int oc = 0; oc < 4; ++oc) {
-          bf16x8 qq = {0, 0, 0, 0, 0, 0, 0, 0};
-          if (ok) qq = *(const bf16x8*)(src + oc * 8);
+      for (int g = 0; g < G; ++g) {
+        int n = n_lo + g;
+        bool nval = n < Nb;
+        for (int i = t; i < XROWS * (W + 2); i += 256) {
+          int xr = i / (W + 2);
+          int col = i - xr * (W + 2) - 1;
+          int row = row0 - 1 + xr;
+          bool ok = nval && (unsigned)row < (unsigned)H &&
+                    (unsigned)col < (unsigned)W;
+          const unsigned short* src =
+              xin + ((((long)(ok ? n : 0) * H + (ok ? row : 0)) << LOGW) +
+                     (ok ? col : 0)) * Cin + c0;
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            x_lds[i][oc * 8 + e] = ((const unsigned short*)&qq)[e];
-        }
-      }
-      __syncthreads();
-
-      // ---- 9-tap MFMA over the supertile's 32-pixel tiles ----
+          for (int oc = 0; oc < 4; ++oc) {
+            bf16x8 qq = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (ok) qq = *(const bf16x8*)(src + oc * 8);
 #pragma unroll
-      for (int tt = 0; tt < ST / 32; ++tt) {
-        int p = tt * 32 + wr * 16 + l15;   // pixel within supertile
-        int prow = (p >> LOGW) + 1;        // +1: halo row offset
-        int pcol = (p & (W - 1)) + 1;      // +1: halo col offset
-        bool pval = p0 + p < P;
-        int base = pval ? (prow * (W + 2) + pcol) : (1 * (W + 2) + 1);
-#pragma unroll
-        for (int r = 0; r < 3; ++r) {
-#pragma unroll
-          for (int s = 0; s < 3; ++s) {
-            bf16x8 a = *(const bf16x8*)&x_lds[base + (r - 1) * (W + 2) +
-                                              (s - 1)][l4 * 8];
-            bf16x8 b = *(const bf16x8*)&w_lds[r * 3 + s][wc * 16 + l15]
-                                             [l4 * 8];
-            acc[tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
-                                                              acc[tt], 0,
-                                                              0, 0);
+            for (int e = 0; e < 8; ++e)
+              x_lds[g][i][oc * 8 + e] = ((const unsigned short*)&qq)[e];
           }
         }
       }
       __syncthreads();
+
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int tt = 0; tt < NT; ++tt) {
+          int p = tt * 32 + wr * 16 + l15;
+          int prow = (p >> LOGW) + 1;
+          int pcol = (p & (W - 1)) + 1;
+          bool pval = p0 + p < P;
+          int base = pval ? (prow * (W + 2) + pcol) : (1 * (W + 2) + 1);
+#pragma unroll
+          for (int r = 0; r < 3; ++r)
+#pragma unroll
+            for (int s2 = 0; s2 < 3; ++s2) {
+              bf16x8 a = *(const bf16x8*)&x_lds[g][base +
+                                                (r - 1) * (W + 2) +
+                                                (s2 - 1)][l4 * 8];
+              bf16x8 b =
+                  *(const bf16x8*)&w_lds[r * 3 + s2][wc * 16 + l15][l4 * 8];
+              acc[g][tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, b, acc[g][tt], 0, 0, 0);
+            }
+        }
+      __syncthreads();
     }
 
-    // ---- epilogue: y[p][ko], bias/relu or relu_y mask ----
 #pragma unroll
-    for (int tt = 0; tt < ST / 32; ++tt) {
+    for (int g = 0; g < G; ++g) {
+      int n = n_lo + g;
+      if (n >= Nb) continue;
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        int p = p0 + tt * 32 + wr * 16 + l4 * 4 + e;
-        int ko = ko0 + wc * 16 + l15;
-        if (p >= P) continue;
-        long oidx = ((long)n * P + p) * Cout + ko;
-        float v = acc[tt][e];
-        if (bias) v += bias[ko];
-        if (relu) v = fmaxf(v, 0.f);
-        if (relu_y && bf2f_(relu_y[oidx]) <= 0.f) v = 0.f;
-        y[oidx] = f2bf_(v);
-      }
+      for (int tt = 0; tt < NT; ++tt)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          int p = p0 + tt * 32 + wr * 16 + l4 * 4 + e;
+          int ko = ko0 + wc * 16 + l15;
+          if (p >= P) continue;
+          long oidx = ((long)n * P + p) * Cout + ko;
+          float v = acc[g][tt][e];
+          if (bias) v += bias[ko];
+          if (relu) v = fmaxf(v, 0.f);
+          if (relu_y && bf2f_(relu_y[oidx]) <= 0.f) v = 0.f;
+          y[oidx] = f2bf_(v);
+        }
     }
   }
 }
@@ -384,15 +397,18 @@ void launch_conv_tap_fwd_bf16(const unsigned short* x,
                               int Cin, int H, int W, int Cout, int relu,
                               int flip, void* st) {
   hipStream_t s = (hipStream_t)st;
-  dim3 grid(Cout / 32, Nb);
-  if (W == 32)
-    conv_tap_fwd_bf16_k<5><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
-                                                Cin, H, Cout, relu, flip);
-  else if (W == 16)
-    conv_tap_fwd_bf16_k<4><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
-                                                Cin, H, Cout, relu, flip);
-  else
-    conv_tap_fwd_bf16_k<3><<<grid, 256, 0, s>>>(x, wt, bias, y, relu_y, Nb,
-                                                Cin, H, Cout, relu, flip);
+  if (W == 32) {  // 8 supertiles/image: split 4-way for occupancy
+    dim3 grid(Cout / 32, Nb, 4);
+    conv_tap_fwd_bf16_k<5, 4, 1><<<grid, 256, 0, s>>>(
+        x, wt, bias, y, relu_y, Nb, Cin, H, Cout, relu, flip);
+  } else if (W == 16) {  // 2 supertiles/image: split 2-way
+    dim3 grid(Cout / 32, Nb, 2);
+    conv_tap_fwd_bf16_k<4, 2, 1><<<grid, 256, 0, s>>>(
+        x, wt, bias, y, relu_y, Nb, Cin, H, Cout, relu, flip);
+  } else {  // 64-pixel images: 4 images share a block's w staging
+    dim3 grid(Cout / 32, (Nb + 3) / 4, 1);
+    conv_tap_fwd_bf16_k<3, 1, 4><<<grid, 256, 0, s>>>(
+        x, wt, bias, y, relu_y, Nb, Cin, H, Cout, relu, flip);
+  }
 }
 }
