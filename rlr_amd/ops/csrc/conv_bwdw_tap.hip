@@ -412,157 +412,3 @@ void launch_conv_tap_fwd_bf16(const unsigned short* x,
   }
 }
 }
-
-// ---------------------------------------------------------------------
-// fp32 tap-resident stride-1 backward-data (3x3, pad 0 or 1): same
-// channel-axis contraction as the bf16 tap fwd, on v_mfma_f32_16x16x4.
-// Covers the reference-CNN dx shapes (fmnist conv2 26x26/KO64, cifar
-// conv2 15x15/KO128, conv3 6x6/KO256) where the implicit-GEMM kernel
-// re-reads dy once per tap.  W is a template value (not pow2-limited:
-// pixel->row decomposition is hoisted per supertile).
-// ---------------------------------------------------------------------
-
-template <int W>
-__global__ __launch_bounds__(256)
-void conv_tap_bwdd_f32_k(const float* __restrict__ dyin,
-                         const float* __restrict__ wp,  // [(rs*KO+ko)][C]
-                         float* __restrict__ dx,
-                         const float* __restrict__ relu_y, int Nb, int KO,
-                         int H, int C) {
-  // dx (H x W) = full correlation of dy ((H-2) x (W-2)) with the flipped
-  // taps: tap (r,s) of dx pixel (ih,iw) reads dy[ih+r-2][iw+s-2] and
-  // w[2-r][2-s] — a pad-2 gather; the 2-wide zero halo in LDS provides it
-  constexpr int ST = 128;
-  constexpr int NT = ST / 32;
-  constexpr int RPS = 127 / W + 2;    // dx rows a supertile can span
-  constexpr int XROWS = RPS + 2;      // dy rows [row0-2, row0+RPS-1]
-  constexpr int XW = W + 2;           // dy cols [-2, W-1] at idx col+2
-  constexpr int CP = 33;              // +1 pad: b32 bank spread
-  constexpr int COT = 32, CCH = 32;
-
-  __shared__ __align__(16) float x_lds[XROWS * XW][CP];
-  __shared__ __align__(16) float w_lds[9][COT][CP];
-
-  const int t = threadIdx.x;
-  const int wave = t >> 6, lane = t & 63;
-  const int wr = wave >> 1, wc = wave & 1;
-  const int l15 = lane & 15, l4 = lane >> 4;
-  const int c0 = blockIdx.x * COT;    // output channels (dx)
-  const int n = blockIdx.y;
-  const int P = H * W;
-  const int OH = H - 2, OW = W - 2;   // dy dims
-  const int n_super = (P + ST - 1) / ST;
-
-  for (int st = blockIdx.z; st < n_super; st += gridDim.z) {
-    const int p0 = st * ST;
-    const int row0 = p0 / W;
-    f32x4 acc[NT];
-#pragma unroll
-    for (int i = 0; i < NT; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
-
-    // hoisted per-lane pixel positions (c-chunk invariant)
-    int pbase[NT];
-#pragma unroll
-    for (int tt = 0; tt < NT; ++tt) {
-      int p = tt * 32 + wr * 16 + l15;
-      int pg = p0 + (p0 + p < P ? p : 0);
-      int prow = pg / W - row0;
-      int pcol = pg - (pg / W) * W;
-      pbase[tt] = prow * XW + pcol;   // tap (r,s) adds r*XW + s
-    }
-
-    for (int k0 = 0; k0 < KO; k0 += CCH) {
-      // w chunk: [(rs*KO + ko)][C] -> [8-rs][c][ko]  (taps flipped)
-      for (int o = t; o < 9 * CCH * COT / 8; o += 256) {
-        int rsk = o >> 2;             // (rs, ko) pair
-        int rs = rsk >> 5;
-        int ko = rsk & 31;
-        int coct = (o & 3) * 8;
-        const float* src = wp + ((long)rs * KO + k0 + ko) * C + c0 + coct;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          w_lds[8 - rs][coct + e][ko] = src[e];
-      }
-      // dy rows row0-2 .. row0+RPS-1 at xr = row-row0+2; col c at c+2
-      for (int i = t; i < XROWS * XW; i += 256) {
-        int xr = i / XW;
-        int col = i - xr * XW - 2;    // -2 .. W-1
-        int row = row0 - 2 + xr;
-        bool ok = (unsigned)row < (unsigned)OH &&
-                  (unsigned)col < (unsigned)OW;
-        const float* src =
-            dyin + (((long)n * OH + (ok ? row : 0)) * OW +
-                    (ok ? col : 0)) * KO + k0;
-#pragma unroll
-        for (int oc = 0; oc < 8; ++oc) {
-          float4 q = {0.f, 0.f, 0.f, 0.f};
-          if (ok) q = *(const float4*)(src + oc * 4);
-          x_lds[i][oc * 4 + 0] = q.x;
-          x_lds[i][oc * 4 + 1] = q.y;
-          x_lds[i][oc * 4 + 2] = q.z;
-          x_lds[i][oc * 4 + 3] = q.w;
-        }
-      }
-      __syncthreads();
-
-#pragma unroll
-      for (int tt = 0; tt < NT; ++tt) {
-#pragma unroll
-        for (int r = 0; r < 3; ++r)
-#pragma unroll
-          for (int s2 = 0; s2 < 3; ++s2) {
-            const float* ap = &x_lds[pbase[tt] + r * XW + s2][l4];
-            const float* bp = &w_lds[r * 3 + s2][wc * 16 + l15][l4];
-#pragma unroll
-            for (int kk = 0; kk < 8; ++kk)
-              acc[tt] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                  ap[kk * 4], bp[kk * 4], acc[tt], 0, 0, 0);
-          }
-      }
-      __syncthreads();
-    }
-
-#pragma unroll
-    for (int tt = 0; tt < NT; ++tt)
-#pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        int p = p0 + tt * 32 + wr * 16 + l4 * 4 + e;
-        int c = c0 + wc * 16 + l15;
-        if (p >= P) continue;
-        long oidx = ((long)n * P + p) * C + c;
-        float v = acc[tt][e];
-        if (relu_y && relu_y[oidx] <= 0.f) v = 0.f;
-        dx[oidx] = v;
-      }
-  }
-}
-
-extern "C" {
-
-int conv_tap_bwdd_f32_ok(int C, int H, int W, int KO, int R, int S,
-                         int stride, int pad) {
-  if (R != 3 || S != 3 || stride != 1 || pad != 0) return 0;
-  if ((C % 32) || (KO % 32)) return 0;
-  return (W == 26 && H == 26) || (W == 15 && H == 15) ||
-         (W == 6 && H == 6);
-}
-
-void launch_conv_tap_bwdd_f32(const float* dy, const float* wp, float* dx,
-                              const float* relu_y, int Nb, int KO, int H,
-                              int W, int C, void* st) {
-  hipStream_t s = (hipStream_t)st;
-  if (W == 26) {
-    dim3 grid(C / 32, Nb, 3);   // 6 supertiles/image
-    conv_tap_bwdd_f32_k<26><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y, Nb,
-                                                 KO, H, C);
-  } else if (W == 15) {
-    dim3 grid(C / 32, Nb, 2);
-    conv_tap_bwdd_f32_k<15><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y, Nb,
-                                                 KO, H, C);
-  } else {
-    dim3 grid(C / 32, Nb, 1);
-    conv_tap_bwdd_f32_k<6><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y, Nb, KO,
-                                                H, C);
-  }
-}
-}

@@ -1247,11 +1247,6 @@ __global__ void wperm_rsko_c_k(const float* __restrict__ w,
 extern "C" void launch_splitk_reduce(const float* ws, float* C,
                                      const float* bias, int M, int N,
                                      int ldc, int SK, int relu, void* s);
-extern "C" int conv_tap_bwdd_f32_ok(int, int, int, int, int, int, int,
-                                    int);
-extern "C" void launch_conv_tap_bwdd_f32(const float*, const float*,
-                                         float*, const float*, int, int,
-                                         int, int, int, void*);
 
 extern "C" {
 void launch_conv_fwd(const float* x, const float* wt, const float* bias,
@@ -1311,12 +1306,6 @@ void launch_conv_bwd_data_relu(const float* dy, const float* wp,
   long M = (long)Nb * H * W;
   dim3 grid((M + 127) / 128, (C + BN - 1) / BN, 1);
   hipStream_t st = (hipStream_t)s;
-  if (conv_tap_bwdd_f32_ok(C, H, W, Kout, R, S, stride, pad)) {
-    // single-pass dy streaming (conv_bwdw_tap.hip): the implicit-GEMM
-    // form re-reads dy once per tap
-    launch_conv_tap_bwdd_f32(dy, wp, dx, relu_y, Nb, Kout, H, W, C, s);
-    return;
-  }
   bool v4 = (Kout % 32) == 0;
   if (C <= 32) {
     dim3 g32((M + 127) / 128, (C + 31) / 32, 1);
