@@ -197,82 +197,76 @@ class Aggregation:
     # -------------------------------------------- Fisher diagnostics
 
     def comp_diag_fisher(self, model_params, poisoned_eval, adv=True):
-        """Diagonal FIM over the poisoned val data (reference
-        aggregation.py:102-129; device handling fixed — the reference
-        instantiates the probe model on CPU even for GPU loaders,
-        SURVEY.md §7 quirks).  `poisoned_eval` is (X, Y) tensors.
-        Mirrors the reference's quirk of gathering LOGITS (not log-probs)
-        for the objective (aggregation.py:123)."""
+        """Diagonal Fisher over the poisoned validation tensors — the
+        per-parameter squared-gradient of the summed target logit
+        (reference aggregation.py:102-129 semantics, tensor-resident and
+        on-device; the reference's probe model never leaves the CPU,
+        SURVEY.md §7 quirks).  adv=True scores against the TRUE labels,
+        adv=False against the attacker's base class.  The reference
+        gathers raw LOGITS (its log_softmax result is discarded,
+        aggregation.py:123) — behavior kept."""
         from . import models as M
         from .flatmodel import FlatParamModel
         X, Y = poisoned_eval
-        model = M.get_model(self.args.data,
-                            getattr(self.args, 'model', None))
-        fm = FlatParamModel(model, X.device)
-        fm.load_vector(model_params.float())
-        fm.eval()
-        precision = torch.zeros_like(fm.flat_params)
-        bs = self.args.bs
-        n_total = X.shape[0]
-        for lo in range(0, n_total, bs):
-            inputs = X[lo:lo + bs]
-            labels = Y[lo:lo + bs].view(-1, 1)
-            if not adv:
-                labels = torch.full_like(labels, self.args.base_class)
-            fm.zero_grad()
-            outputs = fm(inputs)
-            F.log_softmax(outputs, dim=1)  # parity: computed, unused (ref quirk)
-            target_logits = outputs.gather(1, labels)
-            target_logits.sum().backward()
-            precision += (fm.flat_grads ** 2) / n_total
-        return precision.detach()
+        probe = FlatParamModel(
+            M.get_model(self.args.data, getattr(self.args, 'model', None)),
+            X.device)
+        probe.load_vector(model_params.float())
+        probe.eval()
+        fisher = torch.zeros_like(probe.flat_params)
+        n = X.shape[0]
+        for lo in range(0, n, self.args.bs):
+            xb = X[lo:lo + self.args.bs]
+            yb = (Y[lo:lo + self.args.bs] if adv
+                  else torch.full_like(Y[lo:lo + self.args.bs],
+                                       self.args.base_class))
+            probe.zero_grad()
+            logits = probe(xb)
+            F.log_softmax(logits, dim=1)  # parity: computed, unused
+            logits.gather(1, yb.view(-1, 1)).sum().backward()
+            fisher += probe.flat_grads.square() / n
+        return fisher.detach()
 
     def plot_sign_agreement(self, robustLR, cur_global_params,
                             new_global_params, cur_round):
-        """Sign-agreement diagnostics (reference aggregation.py:132-191):
-        intersect top-Fisher coords with RLR-maximized/minimized coords and
-        log 9 L2 scalars."""
-        import numpy as np
-        update = (new_global_params - cur_global_params)
+        """Sign-agreement diagnostics (reference aggregation.py:132-191
+        scalar names/semantics): how much of this round's movement landed
+        on coordinates the attacker's Fisher ranks as important but the
+        honest Fisher does not — split by whether the RLR vote pushed the
+        coordinate forward (+server_lr) or flipped it (-server_lr)."""
+        update = new_global_params - cur_global_params
+        n_par = update.numel()
 
-        fisher_adv = self.comp_diag_fisher(cur_global_params,
-                                           self.poisoned_val_loader)
-        fisher_hon = self.comp_diag_fisher(cur_global_params,
-                                           self.poisoned_val_loader, adv=False)
-        _, adv_idxs = fisher_adv.sort()
-        _, hon_idxs = fisher_hon.sort()
-        n_idxs = self.args.top_frac
-        adv_top = adv_idxs[-n_idxs:].cpu().numpy()
-        hon_top = hon_idxs[-n_idxs:].cpu().numpy()
+        def top_mask(fisher):
+            m = torch.zeros(n_par, dtype=torch.bool, device=update.device)
+            m[fisher.topk(self.args.top_frac).indices] = True
+            return m
 
-        min_idxs = (robustLR == -self.server_lr).nonzero().cpu().numpy()
-        max_idxs = (robustLR == self.server_lr).nonzero().cpu().numpy()
+        adv_top = top_mask(self.comp_diag_fisher(cur_global_params,
+                                                 self.poisoned_val_loader))
+        hon_top = top_mask(self.comp_diag_fisher(cur_global_params,
+                                                 self.poisoned_val_loader,
+                                                 adv=False))
+        fwd = robustLR.view(-1) == self.server_lr    # vote kept direction
+        flipped = robustLR.view(-1) == -self.server_lr
 
-        max_adv = np.intersect1d(adv_top, max_idxs)
-        max_hon = np.intersect1d(hon_top, max_idxs)
-        min_adv = np.intersect1d(adv_top, min_idxs)
-        min_hon = np.intersect1d(hon_top, min_idxs)
+        def l2_on(mask):
+            return update[mask].norm().item()
 
-        max_adv_only = np.setdiff1d(max_adv, max_hon)
-        max_hon_only = np.setdiff1d(max_hon, max_adv)
-        min_adv_only = np.setdiff1d(min_adv, min_hon)
-        min_hon_only = np.setdiff1d(min_hon, min_adv)
-
-        l2 = lambda idxs: torch.norm(update[idxs]).item()
-        max_adv_l2, max_hon_l2 = l2(max_adv_only), l2(max_hon_only)
-        min_adv_l2, min_hon_l2 = l2(min_adv_only), l2(min_hon_only)
-
+        scalars = {
+            'Sign/Adv_Maxim_L2': l2_on(fwd & adv_top & ~hon_top),
+            'Sign/Hon_Maxim_L2': l2_on(fwd & hon_top & ~adv_top),
+            'Sign/Adv_Minim_L2': l2_on(flipped & adv_top & ~hon_top),
+            'Sign/Hon_Minim_L2': l2_on(flipped & hon_top & ~adv_top),
+        }
+        scalars['Sign/Adv_Net_L2'] = (scalars['Sign/Adv_Maxim_L2']
+                                      - scalars['Sign/Adv_Minim_L2'])
+        scalars['Sign/Hon_Net_L2'] = (scalars['Sign/Hon_Maxim_L2']
+                                      - scalars['Sign/Hon_Minim_L2'])
+        self.cum_net_mov += (scalars['Sign/Hon_Net_L2']
+                             - scalars['Sign/Adv_Net_L2'])
+        scalars['Sign/Model_Net_L2_Cumulative'] = self.cum_net_mov
         if self.writer:
-            self.writer.add_scalar('Sign/Hon_Maxim_L2', max_hon_l2, cur_round)
-            self.writer.add_scalar('Sign/Adv_Maxim_L2', max_adv_l2, cur_round)
-            self.writer.add_scalar('Sign/Adv_Minim_L2', min_adv_l2, cur_round)
-            self.writer.add_scalar('Sign/Hon_Minim_L2', min_hon_l2, cur_round)
-        net_adv = max_adv_l2 - min_adv_l2
-        net_hon = max_hon_l2 - min_hon_l2
-        if self.writer:
-            self.writer.add_scalar('Sign/Adv_Net_L2', net_adv, cur_round)
-            self.writer.add_scalar('Sign/Hon_Net_L2', net_hon, cur_round)
-        self.cum_net_mov += (net_hon - net_adv)
-        if self.writer:
-            self.writer.add_scalar('Sign/Model_Net_L2_Cumulative',
-                                   self.cum_net_mov, cur_round)
+            for name, v in scalars.items():
+                self.writer.add_scalar(name, v, cur_round)
+        return scalars
