@@ -13,12 +13,14 @@
 // pass that bn_norm_k covers in 8 us.
 #include "common.h"
 
-constexpr int kBnChunksMax = 512;  // partials allocation bound
-// row chunks scale with M: big activations want ~512-way stage-1
-// parallelism; small deep-layer tensors (M=4k) waste stage-2 iterations
-// on empty chunks.  Multiple of 16 so stage-1 slice blocks tile exactly.
+constexpr int kBnChunksMax = 2048;  // partials allocation bound
+// row chunks scale with M: the big stem/L1 activations want deep stage-1
+// parallelism (at 512 chunks each thread walked 512 serial rows — the
+// 23.8 us bwd-stats1 was ~2.2x its read floor); small deep-layer tensors
+// waste stage-2 iterations on empty chunks.  Multiple of 16 so stage-1
+// slice blocks tile exactly.
 static __host__ __device__ inline int bn_chunks(long M) {
-  long c = M / 32;
+  long c = M / 16;
   if (c > kBnChunksMax) c = kBnChunksMax;
   if (c < 64) c = 64;
   return (int)(c & ~15L);
