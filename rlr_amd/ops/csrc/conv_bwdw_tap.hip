@@ -30,7 +30,12 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
                           int H, int Kout, int n_per_block) {
   constexpr int W = 1 << LOGW;
   constexpr int RPT = 32 / W;         // rows per 32-pixel k-tile
-  constexpr int RING = RPT + 2;
+  // PIPE: a two-tile ring + double-buffered dy lets tile t+1's staging
+  // run beside tile t's MFMA with ONE barrier per tile.  At W=8 the
+  // bigger ring costs a residency level (28->48 KB LDS) for nothing
+  // (2 tiles/image), so the small-ring two-barrier form stays there.
+  constexpr bool PIPE = LOGW != 3;
+  constexpr int RING = PIPE ? 2 * RPT + 2 : RPT + 2;
   constexpr int RS = W + 16;          // in-row stride (multiple of 8)
   constexpr int KOT = 32, CT = 32;
   // per-c row stride: multiple of 8 (16-B aligned fragment reads) with a
@@ -38,7 +43,7 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   constexpr int CSTRIDE = RING * RS + 8;
 
   __shared__ __align__(16) unsigned short x_lds[3][CT][CSTRIDE];
-  __shared__ __align__(16) unsigned short dy_lds[KOT][32 + 8];
+  __shared__ __align__(16) unsigned short dy_lds[PIPE ? 2 : 1][KOT][32 + 8];
 
   const int t = threadIdx.x;
   const int wave = t >> 6, lane = t & 63;
@@ -89,7 +94,7 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
     }
   };
 
-  auto stage_dy = [&](int n, int trow0) {
+  auto stage_dy = [&](int n, int trow0, int buf) {
     // 32 pixels x 4 ko-octets, transposed to [ko][pixel]
     long pix0 = ((long)n * H + trow0) << LOGW;
     for (int o = t; o < 128; o += 256) {
@@ -98,44 +103,70 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
       bf16x8 q = *(const bf16x8*)(dy + (pix0 + pix) * Kout + ko0 + koct);
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        dy_lds[koct + e][pix] = ((const unsigned short*)&q)[e];
+        dy_lds[buf][koct + e][pix] = ((const unsigned short*)&q)[e];
+    }
+  };
+
+  auto mfma_tile = [&](int trow0, int buf) {
+    bf16x8 a = *(const bf16x8*)&dy_lds[buf][wr * 16 + l15][l4 * 8];
+    const int c_lane = wc * 16 + l15;
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      int row = trow0 + mrow_l + r - 1;
+      int slot = (row + 1) % RING;
+      const unsigned short* base = &x_lds[0][c_lane][slot * RS + col_l + 8];
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        bf16x8 b = *(const bf16x8*)(base + (long)s * CT * CSTRIDE);
+        acc[r * 3 + s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, acc[r * 3 + s], 0, 0, 0);
+      }
     }
   };
 
   for (int n = n_lo; n < n_hi; ++n) {
-    // warm the ring: rows -1 .. RPT (slot(row) = (row+1) % RING)
+    if (PIPE) {
+      // warm: rows for tiles 0 AND 1, plus tile 0's dy
 #pragma unroll
-    for (int r = -1; r <= RPT; ++r) stage_row(n, r, (r + 1) % RING);
-    for (int tile = 0; tile < tiles; ++tile) {
-      int trow0 = tile * RPT;
-      stage_dy(n, trow0);
+      for (int r = -1; r <= 2 * RPT; ++r)
+        if (r <= H) stage_row(n, r, (r + 1) % RING);
+      stage_dy(n, 0, 0);
       __syncthreads();
-      bf16x8 a = *(const bf16x8*)&dy_lds[wr * 16 + l15][l4 * 8];
-      const int c_lane = wc * 16 + l15;
+      for (int tile = 0; tile < tiles; ++tile) {
+        int trow0 = tile * RPT;
+        mfma_tile(trow0, tile & 1);
+        // stage tile t+1's dy and rows WHILE t computes: the new rows'
+        // ring slots belong to rows < t*RPT-1, which tile t never reads
+        if (tile + 1 < tiles) {
+          stage_dy(n, trow0 + RPT, (tile + 1) & 1);
 #pragma unroll
-      for (int r = 0; r < 3; ++r) {
-        int row = trow0 + mrow_l + r - 1;
-        int slot = (row + 1) % RING;
-        const unsigned short* base =
-            &x_lds[0][c_lane][slot * RS + col_l + 8];
+          for (int r2 = 1; r2 <= RPT; ++r2) {
+            int row = trow0 + RPT + r2;
+            if (row <= H) stage_row(n, row, (row + 1) % RING);
+          }
+        }
+        __syncthreads();
+      }
+    } else {
+      // small-image form: ring holds one tile (+halo); two barriers
 #pragma unroll
-        for (int s = 0; s < 3; ++s) {
-          bf16x8 b = *(const bf16x8*)(base + (long)s * CT * CSTRIDE);
-          acc[r * 3 + s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a, b, acc[r * 3 + s], 0, 0, 0);
+      for (int r = -1; r <= RPT; ++r) stage_row(n, r, (r + 1) % RING);
+      for (int tile = 0; tile < tiles; ++tile) {
+        int trow0 = tile * RPT;
+        stage_dy(n, trow0, 0);
+        __syncthreads();
+        mfma_tile(trow0, 0);
+        __syncthreads();  // drain reads before the ring advances
+        if (tile + 1 < tiles) {
+#pragma unroll
+          for (int r2 = 1; r2 <= RPT; ++r2) {
+            int row = trow0 + RPT + r2;
+            stage_row(n, row, (row + 1) % RING);
+          }
         }
       }
-      __syncthreads();  // drain reads before the ring advances
-      if (tile + 1 < tiles) {
-        // rows for the next tile: trow0+RPT+1 .. trow0+2*RPT
-#pragma unroll
-        for (int r2 = 1; r2 <= RPT; ++r2) {
-          int row = trow0 + RPT + r2;
-          stage_row(n, row, (row + 1) % RING);
-        }
-      }
+      __syncthreads();  // image boundary: ring refill starts clean
     }
-    __syncthreads();  // image boundary: ring refill starts clean
   }
 
   // write this block's (ko,c) tile partials in final (ko,c,r,s) order
