@@ -31,10 +31,11 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   constexpr int W = 1 << LOGW;
   constexpr int RPT = 32 / W;         // rows per 32-pixel k-tile
   // PIPE: a two-tile ring + double-buffered dy lets tile t+1's staging
-  // run beside tile t's MFMA with ONE barrier per tile.  At W=8 the
-  // bigger ring costs a residency level (28->48 KB LDS) for nothing
-  // (2 tiles/image), so the small-ring two-barrier form stays there.
-  constexpr bool PIPE = LOGW != 3;
+  // run beside tile t's MFMA with ONE barrier per tile.  Measured LOSS at
+  // every W (W=32: 79 -> 101 us; the doubled ring costs a residency level
+  // and the spare waves were already covering the barriers) — kept as a
+  // compile-time branch with the small-ring two-barrier form as default.
+  constexpr bool PIPE = false;
   constexpr int RING = PIPE ? 2 * RPT + 2 : RPT + 2;
   constexpr int RS = W + 16;          // in-row stride (multiple of 8)
   constexpr int KOT = 32, CT = 32;
