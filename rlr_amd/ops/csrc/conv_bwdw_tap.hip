@@ -301,10 +301,17 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
   constexpr int XROWS = RPS + 2;            // + halo
   constexpr int CP = 36;                    // c stride (bank spread)
   constexpr int COT = 32, CCH = 32;
+  constexpr int TOT_W = 9 * CCH * COT / 8;  // w octets per chunk
+  constexpr int PXR = XROWS * (W + 2) * 4;  // x octets per image slice
+  constexpr int TOT_X = G * PXR;
+  constexpr int NW = (TOT_W + 255) / 256;
+  constexpr int NX = (TOT_X + 255) / 256;
 
   // G images share one block (small deep-layer images would otherwise
   // re-stage the w chunk once per image); SPLIT spreads a big image's
-  // supertiles over gridDim.z for occupancy.
+  // supertiles over gridDim.z.  The next chunk's w/x GLOBAL loads are
+  // prefetched into registers DURING the MFMA phase (PMC r02: without
+  // this the kernel sat 57-83% in SQ_WAIT_ANY on the staging loads).
   __shared__ __align__(16) unsigned short x_lds[G][XROWS * (W + 2)][CP];
   __shared__ __align__(16) unsigned short w_lds[9][COT][CP];
 
@@ -317,7 +324,78 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
   const int P = H * W;
   const int n_super = (P + ST - 1) / ST;
 
-  for (int st = blockIdx.z; st < n_super; st += SPLIT) {
+  bf16x8 pw[NW], px[NX];
+
+  auto prefetch = [&](int c0, int row0p) {
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      int o = t + k * 256;
+      if (o < TOT_W) {
+        int rsc = o >> 2;
+        int rs = rsc >> 5;
+        int c = rsc & 31;
+        int koct = (o & 3) * 8;
+        pw[k] = *(const bf16x8*)(wt + ((long)rs * Cin + c0 + c) * Cout +
+                                 ko0 + koct);
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < NX; ++k) {
+      int o = t + k * 256;
+      bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (o < TOT_X) {
+        int g = o / PXR;
+        int rem = o - g * PXR;
+        int i = rem >> 2;
+        int oc = rem & 3;
+        int xr = i / (W + 2);
+        int col = i - xr * (W + 2) - 1;
+        int row = row0p - 1 + xr;
+        int n = n_lo + g;
+        bool ok = n < Nb && (unsigned)row < (unsigned)H &&
+                  (unsigned)col < (unsigned)W;
+        if (ok)
+          q = *(const bf16x8*)(xin +
+                               ((((long)n * H + row) << LOGW) + col) * Cin +
+                               c0 + oc * 8);
+      }
+      px[k] = q;
+    }
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      int o = t + k * 256;
+      if (o < TOT_W) {
+        int rsc = o >> 2;
+        int rs = rsc >> 5;
+        int c = rsc & 31;
+        int koct = (o & 3) * 8;
+        int rsd = flip ? 8 - rs : rs;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          w_lds[rsd][koct + e][c] = ((const unsigned short*)&pw[k])[e];
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < NX; ++k) {
+      int o = t + k * 256;
+      if (o < TOT_X) {
+        int g = o / PXR;
+        int rem = o - g * PXR;
+        int i = rem >> 2;
+        int oc = rem & 3;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          x_lds[g][i][oc * 8 + e] = ((const unsigned short*)&px[k])[e];
+      }
+    }
+  };
+
+  const int st0 = blockIdx.z;
+  if (st0 < n_super) prefetch(0, (st0 * ST) >> LOGW);
+  for (int st = st0; st < n_super; st += SPLIT) {
     const int p0 = st * ST;
     const int row0 = p0 >> LOGW;
     f32x4 acc[G][NT];
@@ -327,44 +405,14 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
       for (int i = 0; i < NT; ++i) acc[g][i] = {0.f, 0.f, 0.f, 0.f};
 
     for (int c0 = 0; c0 < Cin; c0 += CCH) {
-      // ---- w chunk: [(rs*Cin + c)][Cout] -> [rs][ko][c] (flip: bwd) ----
-      for (int o = t; o < 9 * CCH * COT / 8; o += 256) {
-        int rsc = o >> 2;
-        int rs = rsc >> 5;
-        int c = rsc & 31;
-        int koct = (o & 3) * 8;
-        bf16x8 q = *(const bf16x8*)(
-            wt + ((long)rs * Cin + c0 + c) * Cout + ko0 + koct);
-        int rsd = flip ? 8 - rs : rs;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          w_lds[rsd][koct + e][c] = ((const unsigned short*)&q)[e];
-      }
-      // ---- x slices: rows row0-1..row0+RPS per image, halo zeroed ----
-#pragma unroll
-      for (int g = 0; g < G; ++g) {
-        int n = n_lo + g;
-        bool nval = n < Nb;
-        for (int i = t; i < XROWS * (W + 2); i += 256) {
-          int xr = i / (W + 2);
-          int col = i - xr * (W + 2) - 1;
-          int row = row0 - 1 + xr;
-          bool ok = nval && (unsigned)row < (unsigned)H &&
-                    (unsigned)col < (unsigned)W;
-          const unsigned short* src =
-              xin + ((((long)(ok ? n : 0) * H + (ok ? row : 0)) << LOGW) +
-                     (ok ? col : 0)) * Cin + c0;
-#pragma unroll
-          for (int oc = 0; oc < 4; ++oc) {
-            bf16x8 qq = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (ok) qq = *(const bf16x8*)(src + oc * 8);
-#pragma unroll
-            for (int e = 0; e < 8; ++e)
-              x_lds[g][i][oc * 8 + e] = ((const unsigned short*)&qq)[e];
-          }
-        }
-      }
+      write_lds();
       __syncthreads();
+      // next chunk's (or next supertile's) global loads fly under the
+      // MFMA phase below
+      if (c0 + CCH < Cin)
+        prefetch(c0 + CCH, row0);
+      else if (st + SPLIT < n_super)
+        prefetch(0, ((st + SPLIT) * ST) >> LOGW);
 
 #pragma unroll
       for (int g = 0; g < G; ++g)
@@ -437,9 +485,9 @@ void launch_conv_tap_fwd_bf16(const unsigned short* x,
     dim3 grid(Cout / 32, Nb, 2);
     conv_tap_fwd_bf16_k<4, 2, 1><<<grid, 256, 0, s>>>(
         x, wt, bias, y, relu_y, Nb, Cin, H, Cout, relu, flip);
-  } else {  // 64-pixel images: 4 images share a block's w staging
-    dim3 grid(Cout / 32, (Nb + 3) / 4, 1);
-    conv_tap_fwd_bf16_k<3, 1, 4><<<grid, 256, 0, s>>>(
+  } else {  // 64-pixel images: 2 images share a block's w staging
+    dim3 grid(Cout / 32, (Nb + 1) / 2, 1);
+    conv_tap_fwd_bf16_k<3, 1, 2><<<grid, 256, 0, s>>>(
         x, wt, bias, y, relu_y, Nb, Cin, H, Cout, relu, flip);
   }
 }
