@@ -299,7 +299,8 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
   constexpr int NT = ST / 32;               // 32-pixel tiles per supertile
   constexpr int RPS = ST / W;               // rows per supertile
   constexpr int XROWS = RPS + 2;            // + halo
-  constexpr int CP = 36;                    // c stride (bank spread)
+  constexpr int CP = 40;                    // c stride: multiple of 8
+  // (16-B-aligned b128 stores) whose dword form (20) de-conflicts banks
   constexpr int COT = 32, CCH = 32;
   constexpr int TOT_W = 9 * CCH * COT / 8;  // w octets per chunk
   constexpr int PXR = XROWS * (W + 2) * 4;  // x octets per image slice
@@ -331,12 +332,20 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
     for (int k = 0; k < NW; ++k) {
       int o = t + k * 256;
       if (o < TOT_W) {
-        int rsc = o >> 2;
-        int rs = rsc >> 5;
-        int c = rsc & 31;
-        int koct = (o & 3) * 8;
-        pw[k] = *(const bf16x8*)(wt + ((long)rs * Cin + c0 + c) * Cout +
-                                 ko0 + koct);
+        // task = (rs, ko, c-octet): 8 narrow c-strided loads so the LDS
+        // store below is ONE aligned b128 (the 8x b16-store form made the
+        // write phase the pole: 72 ds_write_b16 per thread per chunk)
+        int rs = o / (COT * 4);
+        int rem = o - rs * (COT * 4);
+        int ko = rem >> 2;
+        int coct = (rem & 3) * 8;
+        const unsigned short* src =
+            wt + ((long)rs * Cin + c0 + coct) * Cout + ko0 + ko;
+        bf16x8 q;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          ((unsigned short*)&q)[e] = src[(long)e * Cout];
+        pw[k] = q;
       }
     }
 #pragma unroll
@@ -368,14 +377,12 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
     for (int k = 0; k < NW; ++k) {
       int o = t + k * 256;
       if (o < TOT_W) {
-        int rsc = o >> 2;
-        int rs = rsc >> 5;
-        int c = rsc & 31;
-        int koct = (o & 3) * 8;
+        int rs = o / (COT * 4);
+        int rem = o - rs * (COT * 4);
+        int ko = rem >> 2;
+        int coct = (rem & 3) * 8;
         int rsd = flip ? 8 - rs : rs;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          w_lds[rsd][koct + e][c] = ((const unsigned short*)&pw[k])[e];
+        *(bf16x8*)&w_lds[rsd][ko][coct] = pw[k];
       }
     }
 #pragma unroll
@@ -386,9 +393,7 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
         int rem = o - g * PXR;
         int i = rem >> 2;
         int oc = rem & 3;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          x_lds[g][i][oc * 8 + e] = ((const unsigned short*)&px[k])[e];
+        *(bf16x8*)&x_lds[g][i][oc * 8] = px[k];
       }
     }
   };
