@@ -332,20 +332,12 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
     for (int k = 0; k < NW; ++k) {
       int o = t + k * 256;
       if (o < TOT_W) {
-        // task = (rs, ko, c-octet): 8 narrow c-strided loads so the LDS
-        // store below is ONE aligned b128 (the 8x b16-store form made the
-        // write phase the pole: 72 ds_write_b16 per thread per chunk)
-        int rs = o / (COT * 4);
-        int rem = o - rs * (COT * 4);
-        int ko = rem >> 2;
-        int coct = (rem & 3) * 8;
-        const unsigned short* src =
-            wt + ((long)rs * Cin + c0 + coct) * Cout + ko0 + ko;
-        bf16x8 q;
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          ((unsigned short*)&q)[e] = src[(long)e * Cout];
-        pw[k] = q;
+        int rsc = o >> 2;
+        int rs = rsc >> 5;
+        int c = rsc & 31;
+        int koct = (o & 3) * 8;
+        pw[k] = *(const bf16x8*)(wt + ((long)rs * Cin + c0 + c) * Cout +
+                                 ko0 + koct);
       }
     }
 #pragma unroll
@@ -377,12 +369,14 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
     for (int k = 0; k < NW; ++k) {
       int o = t + k * 256;
       if (o < TOT_W) {
-        int rs = o / (COT * 4);
-        int rem = o - rs * (COT * 4);
-        int ko = rem >> 2;
-        int coct = (rem & 3) * 8;
+        int rsc = o >> 2;
+        int rs = rsc >> 5;
+        int c = rsc & 31;
+        int koct = (o & 3) * 8;
         int rsd = flip ? 8 - rs : rs;
-        *(bf16x8*)&w_lds[rsd][ko][coct] = pw[k];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          w_lds[rsd][koct + e][c] = ((const unsigned short*)&pw[k])[e];
       }
     }
 #pragma unroll
