@@ -299,8 +299,7 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
   constexpr int NT = ST / 32;               // 32-pixel tiles per supertile
   constexpr int RPS = ST / W;               // rows per supertile
   constexpr int XROWS = RPS + 2;            // + halo
-  constexpr int CP = 40;                    // c stride: multiple of 8
-  // (16-B-aligned b128 stores) whose dword form (20) de-conflicts banks
+  constexpr int CP = 36;                    // c stride (bank spread)
   constexpr int COT = 32, CCH = 32;
   constexpr int TOT_W = 9 * CCH * COT / 8;  // w octets per chunk
   constexpr int PXR = XROWS * (W + 2) * 4;  // x octets per image slice
@@ -387,7 +386,11 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
         int rem = o - g * PXR;
         int i = rem >> 2;
         int oc = rem & 3;
-        *(bf16x8*)&x_lds[g][i][oc * 8] = px[k];
+        // scalar stores: a CP=40 layout enabling b128 stores measured 7%
+        // WORSE (the wider row hurt the read-side more than the stores)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          x_lds[g][i][oc * 8 + e] = ((const unsigned short*)&px[k])[e];
       }
     }
   };
