@@ -168,7 +168,14 @@ class Aggregation:
                           else buffer_deltas[0].device)
         if isinstance(buffer_deltas, list):
             buffer_deltas = torch.stack(buffer_deltas)
-        mean = (buffer_deltas * (w / w.sum()).unsqueeze(1)).sum(dim=0)
+        if _gpu(buffer_deltas):
+            # same weighted-mean HIP kernel as the parameter path (the
+            # buffer matrix is tiny — K x ~10k — so the fp64 round-trip
+            # is free and keeps one aggregation code path)
+            mean = ext().agg_avg(buffer_deltas.double().contiguous(), w)
+        else:
+            mean = (buffer_deltas.double()
+                    * (w / w.sum()).unsqueeze(1)).sum(dim=0)
         global_model.flat_buffers.add_(mean.to(global_model.flat_buffers.dtype))
 
     # ------------------------------------------------- server-side extras
