@@ -700,7 +700,8 @@ torch::Tensor gemm(torch::Tensor A, torch::Tensor B,
   torch::Tensor ws;
   float* wsp = nullptr;
   if (SK > 1) {
-    ws = torch::empty({(long)SK * M * N}, A.options());
+    ws = torch::empty({((long)SK + (SK > 16 ? (SK + 15) / 16 : 0)) * M *
+                       N}, A.options());
     wsp = ws.data_ptr<float>();
   }
   launch_gemm_f32(A.data_ptr<float>(), B.data_ptr<float>(),
@@ -721,7 +722,8 @@ static torch::Tensor gemm_layout(const torch::Tensor& A,
   torch::Tensor ws;
   float* wsp = nullptr;
   if (SK > 1) {
-    ws = torch::empty({(long)SK * M * N}, A.options());
+    ws = torch::empty({((long)SK + (SK > 16 ? (SK + 15) / 16 : 0)) * M *
+                       N}, A.options());
     wsp = ws.data_ptr<float>();
   }
   launch_gemm_f32(A.data_ptr<float>(), B.data_ptr<float>(),
@@ -744,7 +746,8 @@ torch::Tensor gemm_bf16(torch::Tensor A, torch::Tensor B,
   torch::Tensor ws;
   float* wsp = nullptr;
   if (SK > 1) {
-    ws = torch::empty({(long)SK * M * N}, A.options().dtype(torch::kFloat));
+    ws = torch::empty({((long)SK + (SK > 16 ? (SK + 15) / 16 : 0)) * M *
+                       N}, A.options().dtype(torch::kFloat));
     wsp = ws.data_ptr<float>();
   }
   torch::Tensor C;
@@ -827,7 +830,8 @@ torch::Tensor linear_bwd_into(torch::Tensor x, torch::Tensor w,
     torch::Tensor ws;
     float* wsp = nullptr;
     if (SK > 1) {
-      ws = torch::empty({(long)SK * out * in}, x.options());
+      ws = torch::empty({((long)SK + (SK > 16 ? (SK + 15) / 16 : 0)) *
+                         out * in}, x.options());
       wsp = ws.data_ptr<float>();
     }
     launch_gemm_f32(dy.data_ptr<float>(), x.data_ptr<float>(),
@@ -920,8 +924,9 @@ static void bf16_dw(const torch::Tensor& dy, const torch::Tensor& x,
   int OH = dy.size(2), OW = dy.size(3);
   long Kd = (long)Nb * OH * OW;
   int SK = conv_bwd_weight_bf16_splitk(Kout, Ncrs, Kd);
-  auto ws = torch::empty({((long)SK + 1) * Kout * Ncrs},
-                         w.options().dtype(torch::kFloat));
+  auto ws = torch::empty(
+      {((long)SK + 1 + (SK > 16 ? (SK + 15) / 16 : 0)) * Kout * Ncrs},
+      w.options().dtype(torch::kFloat));
   launch_conv_bwd_weight_bf16((const unsigned short*)dy.data_ptr(),
                               (const unsigned short*)x.data_ptr(),
                               dw_target.data_ptr<float>(),
@@ -1008,7 +1013,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
   auto dw = torch::empty_like(w);
   // ws: SK slabs + one rsc-ordered temp; the small-Ncrs direct path uses
   // 512 chunk partials instead
-  long ws_mult = (Ncrs <= 32 && Kout <= 64) ? 2049 : (long)SK + 1;
+  long ws_mult = (Ncrs <= 32 && Kout <= 64)
+                     ? 2049
+                     : (long)SK + 1 + (SK > 16 ? (SK + 15) / 16 : 0);
   auto ws = torch::empty({ws_mult * Kout * Ncrs}, w.options());
   launch_conv_bwd_weight(dy.data_ptr<float>(), x.data_ptr<float>(),
                          dw.data_ptr<float>(), ws.data_ptr<float>(), SK, Nb,
@@ -1070,7 +1077,9 @@ torch::Tensor conv2d_bwd_into(torch::Tensor x, torch::Tensor w,
   int Ncrs = C * R * S;
   long Kdim = (long)Nb * OH * OW;
   int SK = conv_bwd_weight_splitk(Kout, Ncrs, Kdim);
-  long ws_mult = (Ncrs <= 32 && Kout <= 64) ? 2049 : (long)SK + 1;
+  long ws_mult = (Ncrs <= 32 && Kout <= 64)
+                     ? 2049
+                     : (long)SK + 1 + (SK > 16 ? (SK + 15) / 16 : 0);
   auto ws = torch::empty({ws_mult * Kout * Ncrs}, w.options());
   launch_conv_bwd_weight(dy.data_ptr<float>(), x.data_ptr<float>(),
                          dw_out.data_ptr<float>(), ws.data_ptr<float>(), SK,

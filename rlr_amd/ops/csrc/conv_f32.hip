@@ -1396,10 +1396,22 @@ void launch_conv_bwd_weight(const float* dy, const float* x, float* dw,
   else
     conv_bwd_weight_k<false, false><<<grid, 256, 0, st>>>(
         dy, x, target, sh, Ncrs, k_per_chunk, SK == 1);
-  // one fused combine+permute pass (for SK==1 the kernel degenerates to
-  // the permute of the directly-written rsc tensor)
-  splitk_reduce_dwperm_k<<<grid_for((long)Kout * Ncrs), kBlock, 0, st>>>(
-      SK == 1 ? rsc : slabs, dw, Kout, C, R * S, SK);
+  // fused combine+permute (for SK==1 it degenerates to the permute of
+  // the directly-written rsc tensor); deep SK first folds 16-slab chunks
+  // with a parallel stage-1 into the extra workspace slabs
+  long n_out = (long)Kout * Ncrs;
+  if (SK > 16) {
+    extern void launch_splitk_partial(const float*, float*, long, int,
+                                      int, void*);
+    int chunks = (SK + 15) / 16;
+    float* ws2 = slabs + (long)SK * n_out + n_out;  // after slabs + rsc
+    launch_splitk_partial(slabs, ws2, n_out, SK, 16, s);
+    splitk_reduce_dwperm_k<<<grid_for(n_out), kBlock, 0, st>>>(
+        ws2, dw, Kout, C, R * S, chunks);
+  } else {
+    splitk_reduce_dwperm_k<<<grid_for(n_out), kBlock, 0, st>>>(
+        SK == 1 ? rsc : slabs, dw, Kout, C, R * S, SK);
+  }
 }
 
 int conv_db_chunks(long M, int Kout) {
@@ -1431,8 +1443,19 @@ void launch_conv_db(const float* dy, float* db, float* partials, int Nb,
 
 void launch_splitk_reduce_dwperm(const float* ws, float* dw, int Kout,
                                  int C, int RS, int SK, void* s) {
-  splitk_reduce_dwperm_k<<<grid_for((long)Kout * C * RS), kBlock, 0,
-                           (hipStream_t)s>>>(ws, dw, Kout, C, RS, SK);
+  long n_out = (long)Kout * C * RS;
+  if (SK > 16) {  // caller provides the stage-1 slabs after slabs + rsc
+    extern void launch_splitk_partial(const float*, float*, long, int, int,
+                                      void*);
+    int chunks = (SK + 15) / 16;
+    float* ws2 = const_cast<float*>(ws) + (long)(SK + 1) * n_out;
+    launch_splitk_partial(ws, ws2, n_out, SK, 16, s);
+    splitk_reduce_dwperm_k<<<grid_for(n_out), kBlock, 0, (hipStream_t)s>>>(
+        ws2, dw, Kout, C, RS, chunks);
+    return;
+  }
+  splitk_reduce_dwperm_k<<<grid_for(n_out), kBlock, 0, (hipStream_t)s>>>(
+      ws, dw, Kout, C, RS, SK);
 }
 
 void launch_dwperm_rsc_crs(const float* in, float* out, int Kout, int C,

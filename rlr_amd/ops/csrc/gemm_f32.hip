@@ -447,11 +447,46 @@ void launch_gemm_f32(const float* A, const float* B, float* C,
   }
 }
 
+// stage-1 partial for deep split-K: each (output, 16-slab chunk) gets its
+// own thread — the single-stage thread-per-output form is parallelism-
+// starved when n_out is small and SK deep (e.g. 64 slabs x 32k outputs)
+__global__ void splitk_partial_k(const float* __restrict__ ws,
+                                 float* __restrict__ out, long n_out,
+                                 int S, int zstride) {
+  long stride = (long)gridDim.x * blockDim.x;
+  long total = n_out * ((S + zstride - 1) / zstride);
+  for (long v = (long)blockIdx.x * blockDim.x + threadIdx.x; v < total;
+       v += stride) {
+    long i = v % n_out;
+    int chunk = (int)(v / n_out);
+    int z0 = chunk * zstride, z1 = min(S, z0 + zstride);
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int z = z0;
+    for (; z + 3 < z1; z += 4) {
+      a0 += ws[(long)z * n_out + i];
+      a1 += ws[(long)(z + 1) * n_out + i];
+      a2 += ws[(long)(z + 2) * n_out + i];
+      a3 += ws[(long)(z + 3) * n_out + i];
+    }
+    for (; z < z1; ++z) a0 += ws[(long)z * n_out + i];
+    out[(long)chunk * n_out + i] = (a0 + a1) + (a2 + a3);
+  }
+}
+
+// callers allocating a split-K workspace with SK > 16 must provide
+// ceil(SK/16) EXTRA slabs after the SK partials (stage-1 scratch)
 void launch_splitk_reduce(const float* ws, float* C, const float* bias,
                           int M, int N, int ldc, int SK, int relu, void* s) {
   long n_out = (long)M * N;
-  // wave path only when outputs are few (its z-strided lane loads are
-  // uncoalesced; for many outputs the thread-per-output ILP loop wins)
+  if (SK > 16 && n_out <= 262144) {
+    int chunks = (SK + 15) / 16;
+    float* ws2 = const_cast<float*>(ws) + (long)SK * n_out;
+    splitk_partial_k<<<grid_for(n_out * chunks), kBlock, 0,
+                       (hipStream_t)s>>>(ws, ws2, n_out, SK, 16);
+    splitk_reduce_k<<<grid_for(n_out), kBlock, 0, (hipStream_t)s>>>(
+        ws2, C, bias, M, N, ldc, chunks, relu);
+    return;
+  }
   if (n_out <= 8192 && SK >= 16) {
     int wpb = kBlock / kWave;
     splitk_reduce_wave_k<<<(n_out + wpb - 1) / wpb, kBlock, 0,
@@ -463,6 +498,13 @@ void launch_splitk_reduce(const float* ws, float* C, const float* bias,
   }
 }
 
+
+void launch_splitk_partial(const float* ws, float* out, long n_out,
+                           int S, int zstride, void* s) {
+  int chunks = (S + zstride - 1) / zstride;
+  splitk_partial_k<<<grid_for(n_out * chunks), kBlock, 0,
+                     (hipStream_t)s>>>(ws, out, n_out, S, zstride);
+}
 
 void launch_colsum(const float* dY, float* db, int M, int N, void* s) {
   int wpb = kBlock / kWave;
