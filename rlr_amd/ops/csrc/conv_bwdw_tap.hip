@@ -213,19 +213,12 @@ __global__ void bwdw_tap_combine_k(const float* __restrict__ partials,
 
 extern "C" {
 
-// 1 if the tap kernel covers this shape (RLR_TAP_BWDW_MIN_W overrides
-// the minimum image width for A/B runs)
+// 1 if the tap kernel covers this shape
 int conv_bwdw_tap_ok(int C, int H, int W, int Kout, int R, int S,
                      int stride, int pad) {
   if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
   if ((C % 32) || (Kout % 32)) return 0;
   if (W != 8 && W != 16 && W != 32) return 0;
-  static int min_w = -1;
-  if (min_w < 0) {
-    const char* e = getenv("RLR_TAP_BWDW_MIN_W");
-    min_w = e ? atoi(e) : 8;
-  }
-  if (W < min_w) return 0;
   int rpt = 32 / W;
   return (H % rpt) == 0;
 }
@@ -423,29 +416,24 @@ void conv_tap_fwd_bf16_k(const unsigned short* __restrict__ xin,
       else if (st + SPLIT < n_super)
         prefetch(0, ((st + SPLIT) * ST) >> LOGW);
 
-      // one base pointer per (g, tt), all 9 tap reads at COMPILE-TIME
-      // immediate offsets from the (r=0,s=0) corner (same for the w
-      // fragments via the rs index) — the per-tap address recomputation
-      // measured 10.8 VALU/MFMA on W=32 (PMC r02)
-      const unsigned short* bp0 = &w_lds[0][wc * 16 + l15][l4 * 8];
 #pragma unroll
       for (int g = 0; g < G; ++g)
 #pragma unroll
         for (int tt = 0; tt < NT; ++tt) {
           int p = tt * 32 + wr * 16 + l15;
-          int prow = p >> LOGW;            // corner: (prow, pcol) + halo-1
-          int pcol = p & (W - 1);
+          int prow = (p >> LOGW) + 1;
+          int pcol = (p & (W - 1)) + 1;
           bool pval = p0 + p < P;
-          const unsigned short* ap0 =
-              &x_lds[g][pval ? (prow * (W + 2) + pcol) : 0][l4 * 8];
+          int base = pval ? (prow * (W + 2) + pcol) : (1 * (W + 2) + 1);
 #pragma unroll
           for (int r = 0; r < 3; ++r)
 #pragma unroll
             for (int s2 = 0; s2 < 3; ++s2) {
-              bf16x8 a = *(const bf16x8*)(ap0 +
-                                          (r * (W + 2) + s2) * CP);
-              bf16x8 b = *(const bf16x8*)(bp0 +
-                                          (r * 3 + s2) * COT * CP);
+              bf16x8 a = *(const bf16x8*)&x_lds[g][base +
+                                                (r - 1) * (W + 2) +
+                                                (s2 - 1)][l4 * 8];
+              bf16x8 b =
+                  *(const bf16x8*)&w_lds[r * 3 + s2][wc * 16 + l15][l4 * 8];
               acc[g][tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                   a, b, acc[g][tt], 0, 0, 0);
             }
