@@ -165,6 +165,11 @@ void launch_conv_tap_fwd_bf16(const unsigned short*, const unsigned short*,
                               const float*, unsigned short*,
                               const unsigned short*, int, int, int, int,
                               int, int, int, void*);
+int conv_tap_bwdd_s2_ok(int, int, int, int, int, int, int, int);
+void launch_conv_tap_bwdd_s2_bf16(const unsigned short*,
+                                  const unsigned short*, unsigned short*,
+                                  const unsigned short*, int, int, int,
+                                  int, int, void*);
 void launch_wperm_rsc_ko_bf16(const float*, unsigned short*, int, int, int,
                               void*);
 void launch_wperm_rsko_c_bf16(const float*, unsigned short*, int, int, int,
@@ -973,6 +978,13 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
                                  (const unsigned short*)wp.data_ptr(),
                                  nullptr, (unsigned short*)dxb.data_ptr(),
                                  nullptr, Nb, Kout, H, W, C, 0, 1, st);
+      else if (conv_tap_bwdd_s2_ok(C, H, W, Kout, R, S, (int)stride,
+                                   (int)pad))
+        launch_conv_tap_bwdd_s2_bf16(
+            (const unsigned short*)dy.data_ptr(),
+            (const unsigned short*)wp.data_ptr(),
+            (unsigned short*)dxb.data_ptr(), nullptr, Nb, Kout, H, W, C,
+            st);
       else
         launch_conv_bwd_data_bf16((const unsigned short*)dy.data_ptr(),
                                   (const unsigned short*)wp.data_ptr(),
@@ -1144,6 +1156,12 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
                                (const unsigned short*)wp.data_ptr(),
                                nullptr, (unsigned short*)dxb.data_ptr(),
                                ry, Nb, Kout, H, W, C, 0, 1, st);
+    else if (conv_tap_bwdd_s2_ok(C, H, W, Kout, R, S, (int)stride,
+                                 (int)pad))
+      launch_conv_tap_bwdd_s2_bf16((const unsigned short*)dy.data_ptr(),
+                                   (const unsigned short*)wp.data_ptr(),
+                                   (unsigned short*)dxb.data_ptr(), ry, Nb,
+                                   Kout, H, W, C, st);
     else
       launch_conv_bwd_data_bf16_relu(
           (const unsigned short*)dy.data_ptr(),

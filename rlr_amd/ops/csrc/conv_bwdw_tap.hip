@@ -494,3 +494,164 @@ void launch_conv_tap_fwd_bf16(const unsigned short* x,
   }
 }
 }
+
+// ---------------------------------------------------------------------
+// Tap-resident bf16 STRIDE-2 backward-data (3x3 s2 p1, dx W in {8,16,32},
+// KO % 32 == 0, C % 32 == 0): dx[ih][iw] = sum over taps (r,s) of
+// dy[(ih+1-r)/2][(iw+1-s)/2] * w[r][s] where the division must be exact
+// (parity): lanes whose pixel has the wrong parity for a tap read a
+// dedicated zero LDS row, so every tap is still one MFMA.  dy is staged
+// once per (supertile, ko-chunk) — the implicit-GEMM form re-gathers it
+// per tap with stride/parity checks in the inner loop.
+// ---------------------------------------------------------------------
+
+template <int LOGW>
+__global__ __launch_bounds__(256)
+void conv_tap_bwdd_s2_bf16_k(const unsigned short* __restrict__ dy,
+                             const unsigned short* __restrict__ wp,
+                             unsigned short* __restrict__ dx,
+                             const unsigned short* __restrict__ relu_y,
+                             int Nb, int KO, int H, int C) {
+  constexpr int W = 1 << LOGW;
+  constexpr int OW = W / 2;
+  constexpr int ST = LOGW == 3 ? 64 : 128;
+  constexpr int NT = ST / 32;
+  constexpr int RPS = ST / W;             // dx rows per supertile
+  constexpr int DYR = RPS / 2 + 2;        // dy rows staged (+halo)
+  constexpr int DYW = OW + 1;             // + one zero col for ow == OW
+  constexpr int CP = 36;
+  constexpr int COT = 32, KCH = 32;
+  constexpr int ZROW = DYR * DYW;         // always-zero position
+
+  __shared__ __align__(16) unsigned short dy_lds[DYR * DYW + 1][CP];
+  __shared__ __align__(16) unsigned short w_lds[9][COT][CP];
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int c0 = blockIdx.x * COT;        // dx channels
+  const int n = blockIdx.y;
+  const int P = H * W;
+  const int OH = H / 2;
+  const int n_super = (P + ST - 1) / ST;
+
+  for (int st = blockIdx.z; st < n_super; st += gridDim.z) {
+    const int p0 = st * ST;
+    const int r0 = p0 >> LOGW;            // first dx row
+    const int oh_base = (r0 - 1) >> 1;    // first staged dy row (may be -1)
+    f32x4 acc[NT];
+#pragma unroll
+    for (int i = 0; i < NT; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < KO; k0 += KCH) {
+      // w chunk: [(rs*KO + ko)][C] -> [rs][c][ko]
+      for (int o = t; o < 9 * KCH * COT / 8; o += 256) {
+        int rsk = o >> 2;                 // (rs, ko) pair; 4 c-octets
+        int rs = rsk >> 5;
+        int ko = rsk & 31;
+        int coct = (o & 3) * 8;
+        const unsigned short* src =
+            wp + ((long)rs * KO + k0 + ko) * C + c0 + coct;
+        // transposed write: w_lds[rs][c][ko]
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          w_lds[rs][coct + e][ko] = src[e];
+      }
+      // dy rows oh_base .. oh_base+DYR-1, col ow in [0, OW) (+1 zero col)
+      for (int i = t; i < DYR * DYW; i += 256) {
+        int lr = i / DYW;
+        int ow = i - lr * DYW;
+        int oh = oh_base + lr;
+        bool ok = (unsigned)oh < (unsigned)OH && ow < OW;
+        const unsigned short* src =
+            dy + (((long)n * OH + (ok ? oh : 0)) * OW + (ok ? ow : 0)) *
+                     KO + k0;
+#pragma unroll
+        for (int oc = 0; oc < 4; ++oc) {
+          bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+          if (ok) q = *(const bf16x8*)(src + oc * 8);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            dy_lds[i][oc * 8 + e] = ((const unsigned short*)&q)[e];
+        }
+      }
+      if (t < CP)  // the zero row
+#pragma unroll
+        for (int e = 0; e < 1; ++e) dy_lds[ZROW][t] = 0;
+      __syncthreads();
+
+#pragma unroll
+      for (int tt = 0; tt < NT; ++tt) {
+        int p = tt * 32 + wr * 16 + l15;
+        int ih = r0 + (p >> LOGW);
+        int iw = p & (W - 1);
+        bool pval = p0 + p < P;
+#pragma unroll
+        for (int r = 0; r < 3; ++r) {
+          int nh = ih + 1 - r;
+          int lr = (nh >> 1) - oh_base;
+          bool okh = pval && !(nh & 1);
+#pragma unroll
+          for (int s2 = 0; s2 < 3; ++s2) {
+            int nw = iw + 1 - s2;
+            int ow = nw >> 1;
+            bool ok = okh && !(nw & 1) && nw >= 0;
+            int pos = ok ? lr * DYW + ow : ZROW;
+            bf16x8 a = *(const bf16x8*)&dy_lds[pos][l4 * 8];
+            bf16x8 b = *(const bf16x8*)&w_lds[r * 3 + s2][wc * 16 + l15]
+                                             [l4 * 8];
+            acc[tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
+                                                              acc[tt], 0,
+                                                              0, 0);
+          }
+        }
+      }
+      __syncthreads();
+    }
+
+#pragma unroll
+    for (int tt = 0; tt < NT; ++tt)
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        int p = p0 + tt * 32 + wr * 16 + l4 * 4 + e;
+        int c = c0 + wc * 16 + l15;
+        if (p >= P) continue;
+        long oidx = ((long)n * P + p) * C + c;
+        float v = acc[tt][e];
+        if (relu_y && bf2f_(relu_y[oidx]) <= 0.f) v = 0.f;
+        dx[oidx] = f2bf_(v);
+      }
+  }
+}
+
+extern "C" {
+
+int conv_tap_bwdd_s2_ok(int C, int H, int W, int KO, int R, int S,
+                        int stride, int pad) {
+  if (R != 3 || S != 3 || stride != 2 || pad != 1) return 0;
+  if ((C % 32) || (KO % 32)) return 0;
+  return (W == 8 || W == 16 || W == 32) && H == W;
+}
+
+void launch_conv_tap_bwdd_s2_bf16(const unsigned short* dy,
+                                  const unsigned short* wp,
+                                  unsigned short* dx,
+                                  const unsigned short* relu_y, int Nb,
+                                  int KO, int H, int W, int C, void* st) {
+  hipStream_t s = (hipStream_t)st;
+  if (W == 32) {
+    dim3 grid(C / 32, Nb, 4);
+    conv_tap_bwdd_s2_bf16_k<5><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y,
+                                                    Nb, KO, H, C);
+  } else if (W == 16) {
+    dim3 grid(C / 32, Nb, 2);
+    conv_tap_bwdd_s2_bf16_k<4><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y,
+                                                    Nb, KO, H, C);
+  } else {
+    dim3 grid(C / 32, Nb, 1);
+    conv_tap_bwdd_s2_bf16_k<3><<<grid, 256, 0, s>>>(dy, wp, dx, relu_y,
+                                                    Nb, KO, H, C);
+  }
+}
+}
