@@ -156,7 +156,11 @@ void launch_conv_bwd_weight_bf16(const unsigned short*,
                                  int, int, int, void*);
 // conv_bwdw_tap.hip
 int conv_bwdw_tap_ok(int, int, int, int, int, int, int, int);
+int conv_bwdw_tap_s2_ok(int, int, int, int, int, int, int, int);
 int conv_bwdw_tap_slabs(int, int, int);
+void launch_conv_bwdw_tap_s2_bf16(const unsigned short*,
+                                  const unsigned short*, float*, float*,
+                                  int, int, int, int, int, void*);
 void launch_conv_bwdw_tap_bf16(const unsigned short*, const unsigned short*,
                                float*, float*, int, int, int, int, int,
                                void*);
@@ -921,6 +925,17 @@ static void bf16_dw(const torch::Tensor& dy, const torch::Tensor& x,
                               (const unsigned short*)x.data_ptr(),
                               dw_target.data_ptr<float>(),
                               ws.data_ptr<float>(), Nb, C, H, W, Kout, st);
+    return;
+  }
+  if (conv_bwdw_tap_s2_ok(C, H, W, Kout, R, S, stride, pad)) {
+    int SL = conv_bwdw_tap_slabs(Nb, C, Kout);
+    auto ws = torch::empty({(long)(SL + 16) * Kout * C * 9},
+                           w.options().dtype(torch::kFloat));
+    launch_conv_bwdw_tap_s2_bf16((const unsigned short*)dy.data_ptr(),
+                                 (const unsigned short*)x.data_ptr(),
+                                 dw_target.data_ptr<float>(),
+                                 ws.data_ptr<float>(), Nb, C, H, W, Kout,
+                                 st);
     return;
   }
   int Ncrs = C * R * S;

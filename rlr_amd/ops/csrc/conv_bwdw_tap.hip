@@ -655,3 +655,205 @@ void launch_conv_tap_bwdd_s2_bf16(const unsigned short* dy,
   }
 }
 }
+
+// ---------------------------------------------------------------------
+// Tap-accumulator bf16 STRIDE-2 backward-weight (3x3 s2 p1, dy OW in
+// {8,16}, C % 32 == 0, Kout % 32 == 0): dw[ko][c][r][s] = sum over dy
+// pixels of dy[oh][ow][ko] * x[2oh+r-1][2ow+s-1][c].  The x columns a
+// tap touches are parity-split: col 2ow (s=1) lives in an even-column
+// copy, cols 2ow -/+ 1 (s=0/2) in two shifted odd-column copies — each
+// OW wide, so every tap fragment is an aligned contiguous b128 read.
+// Rows likewise: x row 2oh in an even-row record, 2oh+1 in an odd-row
+// record (tap r=0 at oh reads the odd record of oh-1).  dy and x stream
+// ONCE per (ko,c) tile like the stride-1 tap kernel.
+// ---------------------------------------------------------------------
+
+template <int LOGOW>
+__global__ __launch_bounds__(256)
+void conv_bwdw_tap_s2_bf16_k(const unsigned short* __restrict__ dy,
+                             const unsigned short* __restrict__ x,
+                             float* __restrict__ partials, int Nb, int C,
+                             int OH, int Kout, int n_per_block) {
+  constexpr int OW = 1 << LOGOW;
+  constexpr int RPT = 32 / OW;        // dy rows per 32-pixel k-tile
+  constexpr int RING = RPT + 2;       // oh-keyed ring (needs oh-1)
+  constexpr int RS = OW + 16;         // in-row stride (multiple of 8)
+  constexpr int KOT = 32, CT = 32;
+  constexpr int CSTRIDE = RING * RS + 8;
+  // copies: 0 = even cols shifted for s=0 (store x[2k+1] at k+1... see
+  // stage_rows), 1 = even-col (s=1), 2 = odd-col (s=2); records: even
+  // x-row (r=1) and odd x-row (r=0 at oh-1 / r=2 at oh)
+  __shared__ __align__(16) unsigned short xe_lds[3][CT][CSTRIDE];  // even
+  __shared__ __align__(16) unsigned short xo_lds[3][CT][CSTRIDE];  // odd
+  __shared__ __align__(16) unsigned short dy_lds[KOT][32 + 8];
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int ko0 = blockIdx.x * KOT;
+  const int c0 = blockIdx.y * CT;
+  const int n_lo = blockIdx.z * n_per_block;
+  const int n_hi = min(Nb, n_lo + n_per_block);
+  const int tiles = OH / RPT;
+  const int H = OH * 2, W = OW * 2;
+
+  for (int i = t; i < 3 * CT * CSTRIDE; i += 256) {
+    ((unsigned short*)xe_lds)[i] = 0;
+    ((unsigned short*)xo_lds)[i] = 0;
+  }
+  __syncthreads();
+
+  f32x4 acc[9];
+#pragma unroll
+  for (int i = 0; i < 9; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int mrow_l = (l4 * 8) >> LOGOW;
+  const int col_l = (l4 * 8) & (OW - 1);
+
+  // stage both x rows of dy-row `oh` (x rows 2oh and 2oh+1) into ring
+  // slot (oh+1)%RING.  Copy layout: a read at aligned position col0+8 of
+  // copy s yields x[2*col0 + s - 1].
+  const int oct_per = OW * CT / 8;    // octets per (row, col-parity)
+  auto stage_rows = [&](int n, int oh, int slot) {
+    // tasks: 2 x-rows (2oh, 2oh+1) x 2 column parities x OW/8*4 octets.
+    // Copy contract: a read at position p+8 of copy cpy (in the record
+    // matching the row parity) yields x[2p + cpy - 1]:
+    //   even cols x[2k]   -> copy 1 at p = k
+    //   odd  cols x[2k+1] -> copy 0 at p = k+1  AND  copy 2 at p = k
+    for (int o = t; o < 4 * oct_per; o += 256) {
+      int rowpar = o / (2 * oct_per);
+      int rem = o - rowpar * (2 * oct_per);
+      int colpar = rem >= oct_per;
+      int oo = colpar ? rem - oct_per : rem;
+      int k = oo >> 2;
+      int coct = (oo & 3) * 8;
+      int xrow = 2 * oh + rowpar;
+      int xcol = 2 * k + colpar;
+      bool ok = (unsigned)xrow < (unsigned)H;
+      bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ok)
+        q = *(const bf16x8*)(x + (((long)n * H + xrow) * W + xcol) * C +
+                             c0 + coct);
+      unsigned short* base = rowpar ? &xo_lds[0][0][0] : &xe_lds[0][0][0];
+#pragma unroll
+      for (int w2 = 0; w2 < 2; ++w2) {
+        int cpy, p;
+        if (colpar == 0) {
+          if (w2 == 1) continue;
+          cpy = 1; p = k;
+        } else {
+          cpy = w2 == 0 ? 0 : 2;
+          p = w2 == 0 ? k + 1 : k;
+        }
+        unsigned short* dst = base + (long)cpy * CT * CSTRIDE;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          dst[(long)(coct + e) * CSTRIDE + slot * RS + p + 8] =
+              ((const unsigned short*)&q)[e];
+      }
+    }
+  };
+
+  auto stage_dy = [&](int n, int trow0) {
+    long pix0 = ((long)n * OH + trow0) << LOGOW;
+    for (int o = t; o < 128; o += 256) {
+      int pix = o >> 2;
+      int koct = (o & 3) * 8;
+      bf16x8 q = *(const bf16x8*)(dy + (pix0 + pix) * Kout + ko0 + koct);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        dy_lds[koct + e][pix] = ((const unsigned short*)&q)[e];
+    }
+  };
+
+  for (int n = n_lo; n < n_hi; ++n) {
+    // warm: dy rows -1 (odd record only matters) .. RPT
+#pragma unroll
+    for (int r = -1; r <= RPT; ++r) stage_rows(n, r, (r + 1) % RING);
+    for (int tile = 0; tile < tiles; ++tile) {
+      int trow0 = tile * RPT;
+      stage_dy(n, trow0);
+      __syncthreads();
+      bf16x8 a = *(const bf16x8*)&dy_lds[wr * 16 + l15][l4 * 8];
+      const int c_lane = wc * 16 + l15;
+#pragma unroll
+      for (int r = 0; r < 3; ++r) {
+        // r=1: even record of oh; r=0: odd record of oh-1; r=2: odd of oh
+        int oh = trow0 + mrow_l + (r == 0 ? -1 : 0);
+        int slot = (oh + 1) % RING;
+        const unsigned short* rec =
+            (r == 1 ? &xe_lds[0][c_lane][0] : &xo_lds[0][c_lane][0]) +
+            slot * RS + col_l + 8;
+#pragma unroll
+        for (int s2 = 0; s2 < 3; ++s2) {
+          bf16x8 b = *(const bf16x8*)(rec + (long)s2 * CT * CSTRIDE);
+          acc[r * 3 + s2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[r * 3 + s2], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+      if (tile + 1 < tiles) {
+#pragma unroll
+        for (int r2 = 1; r2 <= RPT; ++r2) {
+          int oh = trow0 + RPT + r2;
+          stage_rows(n, oh, (oh + 1) % RING);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  long zbase = (long)blockIdx.z * Kout * C * 9;
+#pragma unroll
+  for (int rs = 0; rs < 9; ++rs)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int ko = ko0 + wr * 16 + l4 * 4 + e;
+      int c = c0 + wc * 16 + l15;
+      partials[zbase + ((long)ko * C + c) * 9 + rs] = acc[rs][e];
+    }
+}
+
+extern "C" {
+
+int conv_bwdw_tap_s2_ok(int C, int H, int W, int Kout, int R, int S,
+                        int stride, int pad) {
+  if (R != 3 || S != 3 || stride != 2 || pad != 1) return 0;
+  if ((C % 32) || (Kout % 32)) return 0;
+  int OW = W / 2;
+  if (OW != 8 && OW != 16) return 0;
+  int rpt = 32 / OW;
+  return H == W && (H / 2) % rpt == 0;
+}
+
+void launch_conv_bwdw_tap_s2_bf16(const unsigned short* dy,
+                                  const unsigned short* x, float* dw,
+                                  float* ws, int Nb, int C, int H, int W,
+                                  int Kout, void* st) {
+  hipStream_t s = (hipStream_t)st;
+  int S = conv_bwdw_tap_slabs(Nb, C, Kout);
+  int G = (Nb + S - 1) / S;
+  S = (Nb + G - 1) / G;
+  dim3 grid(Kout / 32, C / 32, S);
+  int OH = H / 2;
+  if (W / 2 == 16)
+    conv_bwdw_tap_s2_bf16_k<4><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, OH,
+                                                    Kout, G);
+  else
+    conv_bwdw_tap_s2_bf16_k<3><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, OH,
+                                                    Kout, G);
+  long n_out = (long)Kout * C * 9;
+  if (S > 16) {
+    int chunks = (S + 15) / 16;
+    float* ws2 = ws + (long)S * n_out;
+    bwdw_tap_combine_k<<<grid_for(n_out * chunks), kBlock, 0, s>>>(
+        ws, ws2, n_out, S, 16);
+    bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws2, dw, n_out,
+                                                          chunks, chunks);
+  } else {
+    bwdw_tap_combine_k<<<grid_for(n_out), kBlock, 0, s>>>(ws, dw, n_out, S,
+                                                          S);
+  }
+}
+}
