@@ -170,6 +170,11 @@ void launch_conv_tap_fwd_bf16(const unsigned short*, const unsigned short*,
                               const unsigned short*, int, int, int, int,
                               int, int, int, void*);
 int conv_tap_bwdd_s2_ok(int, int, int, int, int, int, int, int);
+int conv_tap_fwd_w4_ok(int, int, int, int, int, int, int, int);
+void launch_conv_tap_fwd_w4_bf16(const unsigned short*,
+                                 const unsigned short*, const float*,
+                                 unsigned short*, const unsigned short*,
+                                 int, int, int, int, int, void*);
 void launch_conv_tap_bwdd_s2_bf16(const unsigned short*,
                                   const unsigned short*, unsigned short*,
                                   const unsigned short*, int, int, int,
@@ -878,7 +883,14 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
                                (unsigned short*)wt.data_ptr(), Kout, C,
                                R * S, stream_of(x));
       auto y = empty_cl({Nb, Kout, OH, OW}, x.options());
-      if (conv_tap_fwd_ok(C, H, W, Kout, R, S, (int)stride, (int)pad))
+      if (conv_tap_fwd_w4_ok(C, H, W, Kout, R, S, (int)stride, (int)pad))
+        launch_conv_tap_fwd_w4_bf16(
+            (const unsigned short*)x.data_ptr(),
+            (const unsigned short*)wt.data_ptr(),
+            b ? b->data_ptr<float>() : nullptr,
+            (unsigned short*)y.data_ptr(), nullptr, Nb, C, Kout,
+            relu ? 1 : 0, 0, stream_of(x));
+      else if (conv_tap_fwd_ok(C, H, W, Kout, R, S, (int)stride, (int)pad))
         launch_conv_tap_fwd_bf16((const unsigned short*)x.data_ptr(),
                                  (const unsigned short*)wt.data_ptr(),
                                  b ? b->data_ptr<float>() : nullptr,
@@ -987,7 +999,13 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> conv2d_bwd(
                                (unsigned short*)wp.data_ptr(), Kout, C,
                                R * S, st);
       dxb = empty_cl({Nb, C, H, W}, x.options());
-      if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+      if (conv_tap_fwd_w4_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+        launch_conv_tap_fwd_w4_bf16(
+            (const unsigned short*)dy.data_ptr(),
+            (const unsigned short*)wp.data_ptr(), nullptr,
+            (unsigned short*)dxb.data_ptr(), nullptr, Nb, Kout, C, 0, 1,
+            st);
+      else if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
         // bwd-data == the same correlation over dy with flipped taps
         launch_conv_tap_fwd_bf16((const unsigned short*)dy.data_ptr(),
                                  (const unsigned short*)wp.data_ptr(),
@@ -1166,7 +1184,12 @@ torch::Tensor conv2d_bwd_wdx_into(torch::Tensor x, torch::Tensor w,
       ryt = cl(*relu_y, "conv_bwd.relu_y");
       ry = (const unsigned short*)ryt.data_ptr();
     }
-    if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+    if (conv_tap_fwd_w4_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
+      launch_conv_tap_fwd_w4_bf16((const unsigned short*)dy.data_ptr(),
+                                  (const unsigned short*)wp.data_ptr(),
+                                  nullptr, (unsigned short*)dxb.data_ptr(),
+                                  ry, Nb, Kout, C, 0, 1, st);
+    else if (conv_tap_fwd_ok(Kout, H, W, C, R, S, (int)stride, (int)pad))
       launch_conv_tap_fwd_bf16((const unsigned short*)dy.data_ptr(),
                                (const unsigned short*)wp.data_ptr(),
                                nullptr, (unsigned short*)dxb.data_ptr(),

@@ -857,3 +857,137 @@ void launch_conv_bwdw_tap_s2_bf16(const unsigned short* dy,
   }
 }
 }
+
+// ---------------------------------------------------------------------
+// W=4 tap-resident bf16 fwd / s1 bwd-data: 16-pixel images are PACKED
+// two-per-32-pixel MFMA tile (the result-row dimension indexes a virtual
+// pixel space across GI images whose LDS slices are contiguous).  Same
+// channel-axis contraction and flip convention as conv_tap_fwd_bf16_k.
+// ---------------------------------------------------------------------
+
+template <int GI>  // images per block, multiple of 2
+__global__ __launch_bounds__(256)
+void conv_tap_fwd_w4_bf16_k(const unsigned short* __restrict__ xin,
+                            const unsigned short* __restrict__ wt,
+                            const float* __restrict__ bias,
+                            unsigned short* __restrict__ y,
+                            const unsigned short* __restrict__ relu_y,
+                            int Nb, int Cin, int Cout, int relu,
+                            int flip) {
+  constexpr int W = 4, H = 4, P = 16;
+  constexpr int XW = W + 2, XROWS = H + 2;   // whole image + halo
+  constexpr int NPOS = XROWS * XW;           // 36
+  constexpr int NT = GI / 2;                 // 32-pixel tiles per block
+  constexpr int CP = 36;
+  constexpr int COT = 32, CCH = 32;
+
+  __shared__ __align__(16) unsigned short x_lds[GI * NPOS][CP];
+  __shared__ __align__(16) unsigned short w_lds[9][COT][CP];
+
+  const int t = threadIdx.x;
+  const int wave = t >> 6, lane = t & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int ko0 = blockIdx.x * COT;
+  const int n_lo = blockIdx.y * GI;
+
+  f32x4 acc[NT];
+#pragma unroll
+  for (int i = 0; i < NT; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int c0 = 0; c0 < Cin; c0 += CCH) {
+    for (int o = t; o < 9 * CCH * COT / 8; o += 256) {
+      int rsc = o >> 2;
+      int rs = rsc >> 5;
+      int c = rsc & 31;
+      int koct = (o & 3) * 8;
+      bf16x8 q = *(const bf16x8*)(
+          wt + ((long)rs * Cin + c0 + c) * Cout + ko0 + koct);
+      int rsd = flip ? 8 - rs : rs;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        w_lds[rsd][koct + e][c] = ((const unsigned short*)&q)[e];
+    }
+    for (int i = t; i < GI * NPOS; i += 256) {
+      int g = i / NPOS;
+      int pos = i - g * NPOS;
+      int xr = pos / XW;
+      int col = pos - xr * XW - 1;
+      int row = xr - 1;
+      int n = n_lo + g;
+      bool ok = n < Nb && (unsigned)row < (unsigned)H &&
+                (unsigned)col < (unsigned)W;
+      const unsigned short* src =
+          xin + (((long)(ok ? n : 0) * H + (ok ? row : 0)) * W +
+                 (ok ? col : 0)) * Cin + c0;
+#pragma unroll
+      for (int oc = 0; oc < 4; ++oc) {
+        bf16x8 q = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (ok) q = *(const bf16x8*)(src + oc * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          x_lds[i][oc * 8 + e] = ((const unsigned short*)&q)[e];
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int tt = 0; tt < NT; ++tt) {
+      int p = wr * 16 + l15;              // virtual pixel in [0, 32)
+      int g = tt * 2 + (p >> 4);
+      int pix = p & 15;
+      int base = g * NPOS + ((pix >> 2) + 1) * XW + (pix & 3) + 1;
+#pragma unroll
+      for (int r = 0; r < 3; ++r)
+#pragma unroll
+        for (int s2 = 0; s2 < 3; ++s2) {
+          bf16x8 a = *(const bf16x8*)&x_lds[base + (r - 1) * XW +
+                                            (s2 - 1)][l4 * 8];
+          bf16x8 b =
+              *(const bf16x8*)&w_lds[r * 3 + s2][wc * 16 + l15][l4 * 8];
+          acc[tt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[tt],
+                                                            0, 0, 0);
+        }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int tt = 0; tt < NT; ++tt)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int p = wr * 16 + l4 * 4 + e;
+      int g = tt * 2 + (p >> 4);
+      int pix = p & 15;
+      int n = n_lo + g;
+      int ko = ko0 + wc * 16 + l15;
+      if (n >= Nb) continue;
+      long oidx = ((long)n * P + pix) * Cout + ko;
+      float v = acc[tt][e];
+      if (bias) v += bias[ko];
+      if (relu) v = fmaxf(v, 0.f);
+      if (relu_y && bf2f_(relu_y[oidx]) <= 0.f) v = 0.f;
+      y[oidx] = f2bf_(v);
+    }
+}
+
+extern "C" {
+
+int conv_tap_fwd_w4_ok(int Cin, int H, int W, int Cout, int R, int S,
+                       int stride, int pad) {
+  if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
+  if ((Cin % 32) || (Cout % 32)) return 0;
+  return W == 4 && H == 4;
+}
+
+void launch_conv_tap_fwd_w4_bf16(const unsigned short* x,
+                                 const unsigned short* wt,
+                                 const float* bias, unsigned short* y,
+                                 const unsigned short* relu_y, int Nb,
+                                 int Cin, int Cout, int relu, int flip,
+                                 void* st) {
+  dim3 grid(Cout / 32, (Nb + 3) / 4);
+  conv_tap_fwd_w4_bf16_k<4><<<grid, 256, 0, (hipStream_t)st>>>(
+      x, wt, bias, y, relu_y, Nb, Cin, Cout, relu, flip);
+}
+}
