@@ -119,26 +119,7 @@ __global__ void bn_stats2_k(const float* __restrict__ partials,
   }
 }
 
-// ---- finalize mean/rstd + running stats (torch: running_var unbiased) ----
-__global__ void bn_finalize_k(const float* __restrict__ sums,
-                              float* __restrict__ save_mean,
-                              float* __restrict__ save_rstd,
-                              float* __restrict__ running_mean,
-                              float* __restrict__ running_var, int C,
-                              long count, float momentum, float eps) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mean = sums[c] / count;
-  float var = sums[C + c] / count - mean * mean;
-  var = fmaxf(var, 0.f);
-  save_mean[c] = mean;
-  save_rstd[c] = rsqrtf(var + eps);
-  if (running_mean) {
-    float unbiased = count > 1 ? var * count / (count - 1) : var;
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
-  }
-}
+// (bn_finalize_k was folded into bn_stats2_k's count>0 branch)
 
 __global__ void bn_eval_stats_k(const float* __restrict__ running_mean,
                                 const float* __restrict__ running_var,

@@ -951,8 +951,6 @@ static void bf16_dw(const torch::Tensor& dy, const torch::Tensor& x,
     return;
   }
   int Ncrs = C * R * S;
-  long Kdim = (long)Nb * H * W;  // overwritten below with OH*OW by caller
-  (void)Kdim;
   int OH = dy.size(2), OW = dy.size(3);
   long Kd = (long)Nb * OH * OW;
   int SK = conv_bwd_weight_bf16_splitk(Kout, Ncrs, Kd);

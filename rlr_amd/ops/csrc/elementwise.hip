@@ -842,40 +842,8 @@ __global__ void batch_transpose_k(const T* __restrict__ in,
   }
 }
 
-template <typename T>
-__global__ void nhwc_flatten_k(const T* __restrict__ in, T* __restrict__ out,
-                               long B, int C, int H, int W) {
-  long n = B * C * (long)H * W;
-  long stride = (long)gridDim.x * blockDim.x;
-  int HW = H * W;
-  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < n;
-       o += stride) {
-    int w = o % W;
-    int h = (o / W) % H;
-    int c = (o / HW) % C;
-    long b = o / ((long)HW * C);
-    out[o] = in[((b * H + h) * (long)W + w) * C + c];
-  }
-}
-
-template <typename T>
-__global__ void nhwc_unflatten_k(const T* __restrict__ in,  // CHW flat
-                                 T* __restrict__ out,       // NHWC storage
-                                 long B, int C, int H, int W) {
-  long n = B * C * (long)H * W;
-  long stride = (long)gridDim.x * blockDim.x;
-  int HW = H * W;
-  for (long o = (long)blockIdx.x * blockDim.x + threadIdx.x; o < n;
-       o += stride) {
-    // o indexes the NHWC output (coalesced writes)
-    int c = o % C;
-    long rest = o / C;
-    int w = rest % W;
-    int h = (rest / W) % H;
-    long b = rest / ((long)W * H);
-    out[o] = in[((b * C + c) * (long)H + h) * W + w];
-  }
-}
+// (the scalar flatten/unflatten kernels were replaced by the
+// LDS-tiled batch_transpose_k above)
 
 extern "C" {
 void launch_nhwc_flatten(const float* in, float* out, long B, int C, int H,
