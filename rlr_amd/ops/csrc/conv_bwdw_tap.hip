@@ -1,6 +1,15 @@
-// Tap-accumulator conv backward-weight (bf16, 3x3 stride-1 pad-1,
-// W % 8 == 0 and W in {8,16,32}, H % rows_per_tile == 0, C % 32 == 0,
-// Kout % 32 == 0) — the ResNet18 hot shapes.
+// Tap-resident bf16 convolution kernels for the ResNet18 3x3 shapes —
+// the family that replaced the implicit-GEMM forms wherever the input
+// re-read per (r,s) tap (up to ~9x L2/L3 traffic) was the bound:
+//   conv_bwdw_tap_bf16_k      3x3 s1 p1 bwd-weight, W in {8,16,32}
+//   conv_tap_fwd_bf16_k       3x3 s1 p1 fwd AND bwd-data (tap-flipped)
+//   conv_tap_fwd_w4_bf16_k    the 4x4-image fwd/bwd-data (packed pairs)
+//   conv_tap_bwdd_s2_bf16_k   3x3 s2 p1 bwd-data (parity-masked rows)
+//   conv_bwdw_tap_s2_bf16_k   3x3 s2 p1 bwd-weight (parity-split copies)
+// plus the shared fixed-order partial combines.  Everything deterministic:
+// no atomics, image-group partials summed in index order.
+//
+// First member (bwd-weight s1) in detail:
 //
 // The implicit-GEMM bwd-weight kernel re-reads x once per (r,s) tap and
 // dy once per crs-tile: ~600 MB of L2/L3 traffic per ResNet layer, ~16x
