@@ -100,3 +100,30 @@ def test_cnn_bf16_learns():
         assert h['val_acc'][-1] > 0.5, h['val_acc']
     finally:
         D.DEFAULT_SIZES['fmnist'] = old
+
+
+def test_tap_kernels_run_to_run_deterministic():
+    """The tap-resident bf16 conv kernels (s1/s2, all width classes) use
+    fixed-order partial combines — two identical calls must be bitwise
+    equal (no atomics anywhere on the training path)."""
+    import torch
+    from rlr_amd.ops import ext
+    E = ext()
+    torch.manual_seed(11)
+    for (c, hw, ko, stride, pad) in [(64, 32, 64, 1, 1),   # tap s1 W=32
+                                     (256, 8, 256, 1, 1),  # tap s1 W=8
+                                     (512, 4, 512, 1, 1),  # W=4 packed
+                                     (64, 32, 128, 2, 1),  # tap s2
+                                     (256, 8, 512, 2, 1)]: # s2 OW=4 bwd-d
+        x = (torch.randn(32, c, hw, hw, device='cuda:0')
+             .to(torch.bfloat16)
+             .contiguous(memory_format=torch.channels_last))
+        w = torch.randn(ko, c, 3, 3, device='cuda:0') * 0.05
+        y1 = E.conv2d_fwd(x, w, None, stride, pad, False)
+        y2 = E.conv2d_fwd(x, w, None, stride, pad, False)
+        assert torch.equal(y1, y2), (c, hw, ko, stride)
+        dy = torch.randn_like(y1)
+        dx1, dw1, _ = E.conv2d_bwd(x, w, dy, stride, pad, False, True)
+        dx2, dw2, _ = E.conv2d_bwd(x, w, dy, stride, pad, False, True)
+        assert torch.equal(dw1, dw2), (c, hw, ko, stride)
+        assert torch.equal(dx1, dx2), (c, hw, ko, stride)
