@@ -831,7 +831,9 @@ int conv_bwdw_tap_s2_ok(int C, int H, int W, int Kout, int R, int S,
   if (R != 3 || S != 3 || stride != 2 || pad != 1) return 0;
   if ((C % 32) || (Kout % 32)) return 0;
   int OW = W / 2;
-  if (OW != 8 && OW != 16) return 0;
+  // OW=16 only: at OW=8 the parity-split staging outweighs the traffic
+  // win (tap 159 us vs ~130 implicit-GEMM, profiles/r02_final_resnet.md)
+  if (OW != 16) return 0;
   int rpt = 32 / OW;
   return H == W && (H / 2) % rpt == 0;
 }
@@ -995,8 +997,10 @@ void launch_conv_tap_fwd_w4_bf16(const unsigned short* x,
                                  const unsigned short* relu_y, int Nb,
                                  int Cin, int Cout, int relu, int flip,
                                  void* st) {
-  dim3 grid(Cout / 32, (Nb + 3) / 4);
-  conv_tap_fwd_w4_bf16_k<4><<<grid, 256, 0, (hipStream_t)st>>>(
+  // 8 images per block: at GI=4 the 16 Cin-chunk barrier pairs dominated
+  // the 18-MFMA compute phases (76 us, profiles/r02_final_resnet.md)
+  dim3 grid(Cout / 32, (Nb + 7) / 8);
+  conv_tap_fwd_w4_bf16_k<8><<<grid, 256, 0, (hipStream_t)st>>>(
       x, wt, bias, y, relu_y, Nb, Cin, Cout, relu, flip);
 }
 }
