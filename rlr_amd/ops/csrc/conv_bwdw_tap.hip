@@ -997,10 +997,10 @@ void launch_conv_tap_fwd_w4_bf16(const unsigned short* x,
                                  const unsigned short* relu_y, int Nb,
                                  int Cin, int Cout, int relu, int flip,
                                  void* st) {
-  // 8 images per block: at GI=4 the 16 Cin-chunk barrier pairs dominated
-  // the 18-MFMA compute phases (76 us, profiles/r02_final_resnet.md)
-  dim3 grid(Cout / 32, (Nb + 7) / 8);
-  conv_tap_fwd_w4_bf16_k<8><<<grid, 256, 0, (hipStream_t)st>>>(
+  // GI=8 A/B was a wash (77.2 vs 76.4 us) — stay at 4 images/block for
+  // the wider grid
+  dim3 grid(Cout / 32, (Nb + 3) / 4);
+  conv_tap_fwd_w4_bf16_k<4><<<grid, 256, 0, (hipStream_t)st>>>(
       x, wt, bias, y, relu_y, Nb, Cin, Cout, relu, flip);
 }
 }
