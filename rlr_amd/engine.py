@@ -2,16 +2,19 @@
 
 One FL step of the tiny reference CNNs is ~35-40 short kernels; eagerly
 launched that is host-launch-bound (~3.5 us per launch — MI355X_MICROARCH
-'graph-replay-floor').  The TrainEngine captures the whole step —
-grad-zero, forward, CE loss, backward, fused clip+SGD, optional PGD
-projection — into one hipGraph per batch size and replays it (~10-16 us
-host cost).  Per step only three cheap ops run outside the graph: two
-index_select gathers into the static input buffers and the dropout-state
-advance (the philox seed/offset pair lives in DEVICE memory precisely so
-replays draw fresh masks — ops/functional.DropoutCtx).
+'graph-replay-floor').  The TrainEngine captures the whole step — the
+model's hand-rolled manual tape (fwd + fused backward with grads written
+straight into the flat views; models/*.manual_step) or the autograd
+fallback, then the fused clip+SGD and optional PGD projection — into one
+hipGraph per batch size and replays it (~10-16 us host cost).  Per step
+only three cheap ops run outside the graph: two index_select gathers into
+the static input buffers and the dropout-state advance (the philox
+seed/offset pair lives in DEVICE memory precisely so replays draw fresh
+masks — ops/functional.DropoutCtx).
 
 The captured kernel stream is identical to the eager GPU path, so results
-are bitwise equal with graphs on or off (asserted in test_e2e_gpu)."""
+are bitwise equal with graphs on or off (asserted in test_e2e_gpu), and
+the manual tape is bitwise-equal to autograd (test_manual_tape_gpu)."""
 
 import torch
 
