@@ -985,10 +985,16 @@ void conv_small_fwd_k(const float* __restrict__ x,
       }
     }
     float* yo = y + m * sh.Kout;
+    // float4 stores: the scalar form issued KO_T stores per pixel and the
+    // kernel sat 4.5x over its write floor
 #pragma unroll
-    for (int k = 0; k < KO_T; ++k) {
-      float v = relu ? fmaxf(acc[k], 0.f) : acc[k];
-      yo[k] = v;
+    for (int k = 0; k < KO_T; k += 4) {
+      float4 q = {acc[k], acc[k + 1], acc[k + 2], acc[k + 3]};
+      if (relu) {
+        q.x = fmaxf(q.x, 0.f); q.y = fmaxf(q.y, 0.f);
+        q.z = fmaxf(q.z, 0.f); q.w = fmaxf(q.w, 0.f);
+      }
+      *(float4*)(yo + k) = q;
     }
   }
 }
