@@ -31,7 +31,7 @@
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 
-template <int LOGW>
+template <int LOGW, int KOT = 32>
 __global__ __launch_bounds__(256)
 void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
                           const unsigned short* __restrict__ x,
@@ -47,7 +47,10 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   constexpr bool PIPE = false;
   constexpr int RING = PIPE ? 2 * RPT + 2 : RPT + 2;
   constexpr int RS = W + 16;          // in-row stride (multiple of 8)
-  constexpr int KOT = 32, CT = 32;
+  constexpr int CT = 32;
+  constexpr int MI = KOT / 32;        // wave ko-fragments (KOT=64: 2
+                                      // MFMA per tap per wave — halves
+                                      // the x-staging cost per MFMA)
   // per-c row stride: multiple of 8 (16-B aligned fragment reads) with a
   // dword spread that de-conflicts the 16-lane b128 read groups
   constexpr int CSTRIDE = RING * RS + 8;
@@ -71,9 +74,11 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
     ((unsigned short*)x_lds)[i] = 0;
   __syncthreads();
 
-  f32x4 acc[9];
+  f32x4 acc[9][MI];
 #pragma unroll
-  for (int i = 0; i < 9; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < 9; ++i)
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi) acc[i][mi] = {0.f, 0.f, 0.f, 0.f};
 
   // per-lane fragment decomposition of the pixel chunk l4*8 (W%8==0:
   // a chunk never crosses a row)
@@ -105,11 +110,11 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   };
 
   auto stage_dy = [&](int n, int trow0, int buf) {
-    // 32 pixels x 4 ko-octets, transposed to [ko][pixel]
+    // 32 pixels x KOT/8 ko-octets, transposed to [ko][pixel]
     long pix0 = ((long)n * H + trow0) << LOGW;
-    for (int o = t; o < 128; o += 256) {
-      int pix = o >> 2;
-      int koct = (o & 3) * 8;
+    for (int o = t; o < 32 * KOT / 8; o += 256) {
+      int pix = o / (KOT / 8);
+      int koct = (o - pix * (KOT / 8)) * 8;
       bf16x8 q = *(const bf16x8*)(dy + (pix0 + pix) * Kout + ko0 + koct);
 #pragma unroll
       for (int e = 0; e < 8; ++e)
@@ -118,7 +123,11 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
   };
 
   auto mfma_tile = [&](int trow0, int buf) {
-    bf16x8 a = *(const bf16x8*)&dy_lds[buf][wr * 16 + l15][l4 * 8];
+    bf16x8 a[MI];
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+      a[mi] = *(const bf16x8*)&dy_lds[buf][wr * (16 * MI) + mi * 16 + l15]
+                                    [l4 * 8];
     const int c_lane = wc * 16 + l15;
 #pragma unroll
     for (int r = 0; r < 3; ++r) {
@@ -128,8 +137,10 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
 #pragma unroll
       for (int s = 0; s < 3; ++s) {
         bf16x8 b = *(const bf16x8*)(base + (long)s * CT * CSTRIDE);
-        acc[r * 3 + s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, b, acc[r * 3 + s], 0, 0, 0);
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+          acc[r * 3 + s][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mi], b, acc[r * 3 + s][mi], 0, 0, 0);
       }
     }
   };
@@ -184,11 +195,13 @@ void conv_bwdw_tap_bf16_k(const unsigned short* __restrict__ dy,
 #pragma unroll
   for (int rs = 0; rs < 9; ++rs)
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      int ko = ko0 + wr * 16 + l4 * 4 + e;
-      int c = c0 + wc * 16 + l15;
-      partials[zbase + ((long)ko * C + c) * 9 + rs] = acc[rs][e];
-    }
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        int ko = ko0 + wr * (16 * MI) + mi * 16 + l4 * 4 + e;
+        int c = c0 + wc * 16 + l15;
+        partials[zbase + ((long)ko * C + c) * 9 + rs] = acc[rs][mi][e];
+      }
 }
 
 // fixed-order combine: dw[(ko*C + c)*9 + rs] = sum over image groups.
@@ -228,6 +241,7 @@ int conv_bwdw_tap_ok(int C, int H, int W, int Kout, int R, int S,
   if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
   if ((C % 32) || (Kout % 32)) return 0;
   if (W != 8 && W != 16 && W != 32) return 0;
+  if (W == 8 && (Kout % 64)) return 0;  // the W=8 form tiles ko by 64
   int rpt = 32 / W;
   return (H % rpt) == 0;
 }
@@ -249,16 +263,21 @@ void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
   int S = conv_bwdw_tap_slabs(Nb, C, Kout);
   int G = (Nb + S - 1) / S;
   S = (Nb + G - 1) / G;
-  dim3 grid(Kout / 32, C / 32, S);
-  if (W == 32)
+  if (W == 32) {
+    dim3 grid(Kout / 32, C / 32, S);
     conv_bwdw_tap_bf16_k<5><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
                                                  G);
-  else if (W == 16)
+  } else if (W == 16) {
+    dim3 grid(Kout / 32, C / 32, S);
     conv_bwdw_tap_bf16_k<4><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
                                                  G);
-  else
-    conv_bwdw_tap_bf16_k<3><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
-                                                 G);
+  } else {
+    // W=8 was staging-bound at KOT=32 (134 us): 64-wide ko tiles halve
+    // the x-staging cost per MFMA
+    dim3 grid(Kout / 64, C / 32, S);
+    conv_bwdw_tap_bf16_k<3, 64><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H,
+                                                     Kout, G);
+  }
   long n_out = (long)Kout * C * 9;
   if (S > 16) {
     // stage 1 writes its chunk sums PAST the S slabs (ws is allocated
