@@ -241,7 +241,6 @@ int conv_bwdw_tap_ok(int C, int H, int W, int Kout, int R, int S,
   if (R != 3 || S != 3 || stride != 1 || pad != 1) return 0;
   if ((C % 32) || (Kout % 32)) return 0;
   if (W != 8 && W != 16 && W != 32) return 0;
-  if (W == 8 && (Kout % 64)) return 0;  // the W=8 form tiles ko by 64
   int rpt = 32 / W;
   return (H % rpt) == 0;
 }
@@ -272,11 +271,11 @@ void launch_conv_bwdw_tap_bf16(const unsigned short* dy,
     conv_bwdw_tap_bf16_k<4><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
                                                  G);
   } else {
-    // W=8 was staging-bound at KOT=32 (134 us): 64-wide ko tiles halve
-    // the x-staging cost per MFMA
-    dim3 grid(Kout / 64, C / 32, S);
-    conv_bwdw_tap_bf16_k<3, 64><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H,
-                                                     Kout, G);
+    // KOT=64 A/B at W=8: staging per MFMA halves but the grid halves too
+    // — 142 vs 134 us, starvation wins.  Stays at 32-wide ko tiles.
+    dim3 grid(Kout / 32, C / 32, S);
+    conv_bwdw_tap_bf16_k<3><<<grid, 256, 0, s>>>(dy, x, ws, Nb, C, H, Kout,
+                                                 G);
   }
   long n_out = (long)Kout * C * 9;
   if (S > 16) {
