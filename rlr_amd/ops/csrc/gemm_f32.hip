@@ -379,14 +379,23 @@ __global__ void colsum_k(const float* __restrict__ dY,
 
 extern "C" {
 
-// Heuristic split-K: fill the chip (256 CUs) when the tile grid is small.
+// Heuristic split-K: fill the chip (256 CUs) when the tile grid is
+// small.  The block target is env-tunable for sweeps
+// (RLR_GEMM_SK_TARGET); the sweep at the fc1 shapes read 5.43 / 5.35 /
+// 5.23 rounds/s at targets 256 / 512 / 1024 — deeper splits pay more in
+// slab+reduce traffic than the overlapped prologue returns.
 int gemm_f32_splitk(int M, int N, int K) {
+  static long target = -1;
+  if (target < 0) {
+    const char* e = getenv("RLR_GEMM_SK_TARGET");
+    target = e ? atol(e) : 256;
+  }
   long tiles = ((M + BM - 1) / BM) * (long)((N + BN - 1) / BN);
   if (tiles >= 192 || K <= 2 * BK) return 1;
-  long want = (256 + tiles - 1) / tiles;
+  long want = (target + tiles - 1) / tiles;
   long max_chunks = (K + BK - 1) / BK;
   long sk = want < max_chunks ? want : max_chunks;
-  return (int)(sk < 1 ? 1 : (sk > 128 ? 128 : sk));
+  return (int)(sk < 1 ? 1 : (sk > 192 ? 192 : sk));
 }
 
 // ws: null unless SK>1, then SK*M*N floats.
